@@ -1,0 +1,350 @@
+#!/usr/bin/env python3
+"""bench.py — TSBS-devops scan benchmark for the cnosdb_gs MI355X engine.
+
+Measures BASELINE.json's metric ("decoded+filtered values/sec & HBM GB/s,
+TSBS devops scan") on BASELINE.json configs[1] — the north-star workload:
+
+  10,000 series x 1,000,000 points per GPU; timestamps at 1 s spacing
+  (DeltaTs/RLE pages), f64 gauge values = 0.1-quantized random walk in
+  [0,100] (TSBS cpu-usage-like, Gorilla ~7.6 bits/value); pages of
+  125,000 rows (8 per series); time-range filter ts in [25%,75%] of the
+  range (50% selectivity); fused per-5-minute max/sum/count buckets.
+
+One STEP = one full pass of the hot path over the resident page set:
+decode ts + f64 -> closed-interval time filter -> compacted row output
++ per-bucket aggregates.  Inputs (raw TSM pages) are resident in HBM
+before the timed region.  Multi-GPU: series sharded by series_id % N
+(weak scaling — each rank holds its own 10k-series shard), one RCCL
+all-reduce of the per-bucket sum/count (+max) per step, exactly the one
+collective the path needs.
+
+`value` = field values decoded+filtered per second, whole-job across all
+ranks.  A "value" is one point of the measured f64 field column (1e7*N
+series-points per step... see config).  Roofline + cpu_baseline objects
+per the driver contract; the roofline covers the dominant kernel
+(k_seq_f64, the Gorilla page decoder).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+NS = 1_000_000_000
+T0 = 1_700_000_000_000_000_000
+BUCKET_NS = 300 * NS  # 5 minutes
+
+
+def build_workload(nseries, npts, page_rows, unique, seed=231):
+    """Generate encoded TSM pages. `unique` distinct value series are
+    generated and replicated across series (device copies are distinct, so
+    HBM traffic is real); ts pages are identical across series (TSBS
+    devops: all hosts share the epoch grid)."""
+    import cnosdb_amd as gs
+    rng = np.random.default_rng(seed)
+    npages = npts // page_rows
+    ts_pages = []
+    for p in range(npages):
+        ts = T0 + (np.arange(page_rows, dtype=np.int64) + p * page_rows) * NS
+        ts_pages.append(gs.page_of(ts, gs.CT_TIME))
+
+    # unique value patterns, encoded with the OMP batch encoder in chunks
+    lib = gs.PageLib().lib
+    import ctypes
+    lib.gs_encode_f64_pages_omp.restype = ctypes.c_int32
+    cap = page_rows * 12 + 64
+    chunk = max(1, min(unique, 512 // npages or 1))
+    val_pages = []  # [unique][npages] full page bytes
+    u = 0
+    while u < unique:
+        cu = min(chunk, unique - u)
+        vals = np.empty((cu, npts))
+        for k in range(cu):
+            vals[k] = np.round(
+                np.clip(np.cumsum(rng.normal(0, 0.5, npts)) + 50, 0, 100), 1)
+        flat = np.ascontiguousarray(vals.reshape(-1))
+        total_pages = cu * npages
+        enc = np.zeros(total_pages * cap, dtype=np.uint8)
+        lens = np.zeros(total_pages, dtype=np.int64)
+        st = lib.gs_encode_f64_pages_omp(
+            flat.ctypes.data_as(ctypes.c_void_p), page_rows, total_pages,
+            enc.ctypes.data_as(ctypes.c_void_p), cap,
+            lens.ctypes.data_as(ctypes.c_void_p), os.cpu_count() or 8)
+        assert st == 0, st
+        for k in range(cu):
+            row = []
+            for p in range(npages):
+                i = k * npages + p
+                data = enc[i * cap:i * cap + lens[i]].tobytes()
+                row.append(gs.build_page(data, page_rows))
+            val_pages.append(row)
+        u += cu
+
+    raw_bytes = nseries * sum(len(b) for b in val_pages[0]) + \
+        nseries * sum(len(b) for b in ts_pages)
+    # column groups: one per (series, page) — each page pair is one time slab
+    pattern = rng.integers(0, unique, nseries)
+    groups = []
+    for s in range(nseries):
+        vp = val_pages[pattern[s]]
+        for p in range(npages):
+            groups.append((s, [(ts_pages[p], gs.CT_TIME), (vp[p], gs.CT_F64)]))
+    return groups, raw_bytes, npages
+
+
+def cpu_baseline_leg(nseries_sample, npts, page_rows, lo, hi, seed=231):
+    """Oracle (C restatement, OpenMP) timed on host cores over a bounded
+    sample of the same workload. kind="port" (the reference is Rust and
+    cannot be compiled here)."""
+    import ctypes
+    from oracle import pyoracle as orc
+    import cnosdb_amd as gs
+
+    rng = np.random.default_rng(seed)
+    npages = npts // page_rows
+    # build sample pages (encoded data buffers + bitsets)
+    ts_datas, val_datas = [], []
+    for p in range(npages):
+        ts = T0 + (np.arange(page_rows, dtype=np.int64) + p * page_rows) * NS
+        ts_datas.append(gs.encode_ts(ts))
+    for s in range(nseries_sample):
+        v = np.round(np.clip(np.cumsum(rng.normal(0, 0.5, npts)) + 50, 0, 100), 1)
+        for p in range(npages):
+            val_datas.append(gs.encode_f64(v[p * page_rows:(p + 1) * page_rows]))
+
+    o = orc.Oracle().lib
+
+    class PD(ctypes.Structure):
+        _fields_ = [("data", ctypes.c_void_p), ("data_len", ctypes.c_uint64),
+                    ("bitset", ctypes.c_void_p), ("nrows", ctypes.c_int64),
+                    ("out_off", ctypes.c_uint64), ("ctype", ctypes.c_uint8)]
+
+    nb = (page_rows + 7) // 8
+    bitset = np.full(nb, 0xFF, dtype=np.uint8)
+    total_pages = nseries_sample * npages
+    descs_v = (PD * total_pages)()
+    keep = []
+    for i, d in enumerate(val_datas):
+        a = np.frombuffer(d, dtype=np.uint8)
+        keep.append(a)
+        descs_v[i] = PD(a.ctypes.data, a.size, bitset.ctypes.data, page_rows,
+                        i * page_rows, 1)
+    descs_t = (PD * total_pages)()
+    for s in range(nseries_sample):
+        for p in range(npages):
+            i = s * npages + p
+            a = np.frombuffer(ts_datas[p], dtype=np.uint8)
+            keep.append(a)
+            descs_t[i] = PD(a.ctypes.data, a.size, bitset.ctypes.data,
+                            page_rows, i * page_rows, 0)
+    rows = total_pages * page_rows
+    out_ts = np.zeros(rows, dtype=np.int64)
+    out_v = np.zeros(rows, dtype=np.float64)
+    cores = os.cpu_count() or 8
+    t = time.perf_counter()
+    st = o.orc_decode_pages_omp(descs_t, total_pages,
+                                out_ts.ctypes.data_as(ctypes.c_void_p), cores)
+    assert st == 0
+    st = o.orc_decode_pages_omp(descs_v, total_pages,
+                                out_v.ctypes.data_as(ctypes.c_void_p), cores)
+    assert st == 0
+    # per-series time filter + compaction (numpy leg of the CPU path)
+    sel_parts = []
+    ots = out_ts.reshape(nseries_sample, npts)
+    ov = out_v.reshape(nseries_sample, npts)
+    for s in range(nseries_sample):
+        a = np.searchsorted(ots[s], lo, side="left")
+        b = np.searchsorted(ots[s], hi, side="right")
+        sel_parts.append(ov[s][a:b])
+    np.concatenate(sel_parts)
+    dt = time.perf_counter() - t
+    values = nseries_sample * npts
+    return {"value": values / dt, "unit": "values/s", "cores": cores,
+            "kind": "port",
+            "sample": f"{nseries_sample} series x {npts} pts "
+                      f"(decode ts+f64, filter, compact; oracle C+OpenMP)"}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--series", type=int, default=10000)
+    ap.add_argument("--npts", type=int, default=1_000_000)
+    ap.add_argument("--page-rows", type=int, default=125_000)
+    ap.add_argument("--sub-batches", type=int, default=4)
+    ap.add_argument("--unique", type=int, default=256)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+    import cnosdb_amd as gs
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+    device = torch.device(f"cuda:{local_rank}")
+    torch.cuda.set_device(device)
+
+    nseries, npts, page_rows = args.series, args.npts, args.page_rows
+    assert npts % page_rows == 0
+    assert nseries % args.sub_batches == 0
+    lo = T0 + int(0.25 * npts) * NS
+    hi = T0 + int(0.75 * npts) * NS - 1  # closed interval, 50% of rows
+    nbuckets = int(npts * NS // BUCKET_NS) + 1
+
+    # ---- build + upload (untimed setup) ----
+    t_setup = time.perf_counter()
+    groups, raw_bytes, npages = build_workload(
+        nseries, npts, page_rows, min(args.unique, nseries),
+        seed=231 + rank)
+    eng = gs.Engine(local_rank)
+    per_sb = nseries // args.sub_batches
+    gpp = npages  # groups per series
+    sets = []
+    for sb in range(args.sub_batches):
+        part = groups[sb * per_sb * gpp:(sb + 1) * per_sb * gpp]
+        sets.append(eng.upload(part, validate_crc=False))
+    sb_rows = sets[0].rows
+    d_ts = torch.zeros(sb_rows, dtype=torch.int64, device=device)
+    d_val = torch.zeros(sb_rows, dtype=torch.float64, device=device)
+    d_ots = torch.zeros(sb_rows, dtype=torch.int64, device=device)
+    d_oval = torch.zeros(sb_rows, dtype=torch.float64, device=device)
+    d_max = torch.full((nbuckets,), -np.inf, dtype=torch.float64, device=device)
+    d_sum = torch.zeros(nbuckets, dtype=torch.float64, device=device)
+    d_cnt = torch.zeros(nbuckets, dtype=torch.int64, device=device)
+    agg = dict(bucket_ns=BUCKET_NS, t0=T0, n_buckets=nbuckets,
+               d_max=d_max, d_sum=d_sum, d_count=d_cnt)
+    setup_s = time.perf_counter() - t_setup
+
+    def step():
+        out_rows = 0
+        phase_ms = np.zeros(5)
+        for st_ in sets:
+            r = eng.scan(st_, d_ts, d_val, time_range=(lo, hi),
+                         d_out_ts=d_ots, d_out_val=d_oval, agg=agg)
+            out_rows += r.out_rows
+            phase_ms += [r.ms_decode_ts, r.ms_decode_val, r.ms_filter,
+                         r.ms_compact, r.ms_agg]
+        if dist is not None:
+            dist.all_reduce(d_sum)
+            dist.all_reduce(d_cnt)
+            dist.all_reduce(d_max, op=dist.ReduceOp.MAX)
+        return out_rows, phase_ms
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        out_rows, _ = step()
+    if dist is not None:
+        dist.barrier()
+    torch.cuda.synchronize()
+
+    # ---- timed steps ----
+    t0 = time.perf_counter()
+    phases = np.zeros(5)
+    for _ in range(args.steps):
+        out_rows, pm = step()
+        phases += pm
+    if dist is not None:
+        dist.barrier()
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+
+    # MAX over ranks
+    if dist is not None:
+        t_t = torch.tensor([dt], device=device)
+        dist.all_reduce(t_t, op=dist.ReduceOp.MAX)
+        dt = float(t_t.item())
+
+    values_per_step = nseries * npts  # field values decoded+filtered per rank
+    total_values = values_per_step * world * args.steps
+    value = total_values / dt
+    ms_per_step = dt / args.steps * 1000
+
+    # roofline: dominant kernel = k_seq_f64 (Gorilla decode), 1 launch per
+    # sub-batch. algorithmic bytes per launch = compressed f64 bytes read +
+    # 8 B/row decoded output written.
+    raw_f64_bytes = sum(len(pb) - 16 - ((page_rows + 7) // 8)
+                        for _, pages in groups[:per_sb * gpp]
+                        for pb, ct in [pages[1]])
+    alg_bytes_launch = raw_f64_bytes + 8 * sb_rows
+    ms_gorilla_launch = phases[1] / (args.steps * args.sub_batches)
+    achieved = alg_bytes_launch / (ms_gorilla_launch / 1000) if ms_gorilla_launch > 0 else 0
+    peak = 8.0e12
+    traffic = None
+    pmc_path = os.path.join(REPO, "profiles", "pmc_traffic.json")
+    if os.path.exists(pmc_path):
+        try:
+            pmc = json.load(open(pmc_path))
+            if pmc.get("kernel") == "k_seq_f64":
+                traffic = pmc.get("bytes_per_launch")
+        except Exception:
+            pass
+    roofline = {"bound": "hbm", "achieved": achieved, "peak": peak,
+                "unit": "B/s", "frac": achieved / peak, "traffic": traffic}
+
+    cpu_baseline = None
+    if rank == 0 and not args.skip_cpu_baseline:
+        cpu_baseline = cpu_baseline_leg(32, npts, page_rows, lo, hi)
+
+    if rank == 0:
+        line = {
+            "metric": "decoded+filtered values/sec & HBM GB/s, TSBS devops scan, 1/2/4/8 GPU",
+            "value": value,
+            "unit": "values/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": "tsbs-devops-scan (BASELINE configs[1]: 10k series x 1M pts, delta-ts + Gorilla-f64 decode, ts-range filter 50%, fused 5-min max/sum/count)",
+                "series_per_gpu": nseries,
+                "points_per_series": npts,
+                "page_rows": page_rows,
+                "selectivity": 0.5,
+                "n_buckets": nbuckets,
+                "raw_page_bytes_per_gpu": raw_bytes,
+                "rows_per_step_per_gpu": values_per_step,
+                "out_rows_per_step_per_gpu": int(out_rows),
+                "sub_batches": args.sub_batches,
+                "setup_s": round(setup_s, 1),
+                "phase_ms_per_step": {
+                    "decode_ts": phases[0] / args.steps,
+                    "decode_f64": phases[1] / args.steps,
+                    "filter": phases[2] / args.steps,
+                    "compact": phases[3] / args.steps,
+                    "agg": phases[4] / args.steps,
+                },
+                "effective_GBps": value * (raw_bytes / (nseries * npts) + 8) / 1e9,
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line))
+
+    for st_ in sets:
+        st_.free()
+    eng.close()
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

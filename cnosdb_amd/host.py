@@ -1,0 +1,340 @@
+"""ctypes host bindings for libcnosdb_gs.so (see include/cnosdb_gs.h).
+
+The binding mirrors the seam the reference's thin Rust shim would call
+(SURVEY.md §8b): page upload standing where `ColumnGroupReader::read`
+stands, `Engine.decode` standing where `decode_pages` stands
+(tskv/src/tsm/reader.rs:494-560), `Engine.scan` standing where the
+pure-time-range `DataFilter` + downsampling aggregate stand.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+CT_TIME, CT_I64, CT_F64, CT_BOOL, CT_U64 = 0, 1, 2, 3, 4
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+
+
+def lib_path():
+    return os.path.join(_DIR, "libcnosdb_gs.so")
+
+
+class GsTimeRange(ctypes.Structure):
+    _fields_ = [("min_ts", ctypes.c_int64), ("max_ts", ctypes.c_int64)]
+
+
+class GsPageSpec(ctypes.Structure):
+    _fields_ = [
+        ("bytes", ctypes.c_void_p),
+        ("len", ctypes.c_uint64),
+        ("num_values", ctypes.c_uint32),
+        ("ctype", ctypes.c_uint8),
+        ("column_id", ctypes.c_uint32),
+    ]
+
+
+class GsColumnGroupDesc(ctypes.Structure):
+    _fields_ = [
+        ("pages", ctypes.POINTER(GsPageSpec)),
+        ("npages", ctypes.c_uint32),
+        ("series_id", ctypes.c_uint32),
+    ]
+
+
+class GsScanSpec(ctypes.Structure):
+    _fields_ = [
+        ("range", GsTimeRange),
+        ("tombstones", ctypes.POINTER(GsTimeRange)),
+        ("n_tombstones", ctypes.c_size_t),
+        ("bucket_ns", ctypes.c_int64),
+        ("t0", ctypes.c_int64),
+        ("n_buckets", ctypes.c_int32),
+        ("d_agg_max", ctypes.c_void_p),
+        ("d_agg_sum", ctypes.c_void_p),
+        ("d_agg_count", ctypes.c_void_p),
+        ("d_out_ts", ctypes.c_void_p),
+        ("d_out_val", ctypes.c_void_p),
+        ("d_ts", ctypes.c_void_p),
+        ("d_val", ctypes.c_void_p),
+    ]
+
+
+class GsScanResult(ctypes.Structure):
+    _fields_ = [
+        ("out_rows", ctypes.c_int64),
+        ("decoded_rows", ctypes.c_int64),
+        ("ms_decode_ts", ctypes.c_double),
+        ("ms_decode_val", ctypes.c_double),
+        ("ms_filter", ctypes.c_double),
+        ("ms_compact", ctypes.c_double),
+        ("ms_agg", ctypes.c_double),
+    ]
+
+
+class PageLib:
+    """Loads libcnosdb_gs.so and declares signatures. Singleton per process."""
+
+    _inst = None
+
+    def __new__(cls):
+        if cls._inst is None:
+            cls._inst = super().__new__(cls)
+            cls._inst._load()
+        return cls._inst
+
+    def _load(self):
+        path = lib_path()
+        if not os.path.exists(path):
+            raise RuntimeError(
+                f"cnosdb_gs HIP extension not built: {path} missing. "
+                "Run __graft_entry__.build() — the product path has no "
+                "CPU fallback.")
+        lib = ctypes.CDLL(path)
+        self.lib = lib
+        for nm in ("gs_encode_ts", "gs_encode_i64", "gs_encode_f64",
+                   "gs_encode_bool", "gs_build_page"):
+            getattr(lib, nm).restype = ctypes.c_int64
+        lib.gs_crc32.restype = ctypes.c_uint32
+        lib.gs_version.restype = ctypes.c_char_p
+        lib.gs_last_error.restype = ctypes.c_char_p
+        lib.gs_device_count.restype = ctypes.c_int32
+        lib.gs_ctx_create.restype = ctypes.c_void_p
+        lib.gs_ctx_create.argtypes = [ctypes.c_int32]
+        lib.gs_ctx_destroy.argtypes = [ctypes.c_void_p]
+        lib.gs_groups_upload.restype = ctypes.c_void_p
+        lib.gs_groups_upload.argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(GsColumnGroupDesc),
+            ctypes.c_size_t, ctypes.c_int32]
+        lib.gs_groups_free.argtypes = [ctypes.c_void_p]
+        lib.gs_set_rows.restype = ctypes.c_int64
+        lib.gs_set_rows.argtypes = [ctypes.c_void_p]
+        lib.gs_set_groups.restype = ctypes.c_int64
+        lib.gs_set_groups.argtypes = [ctypes.c_void_p]
+        lib.gs_set_row_offsets.restype = ctypes.c_int32
+        lib.gs_set_row_offsets.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+        lib.gs_decode.restype = ctypes.c_int32
+        lib.gs_decode.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                  ctypes.c_uint32, ctypes.c_void_p,
+                                  ctypes.c_void_p]
+        lib.gs_apply_tombstone.restype = ctypes.c_int32
+        lib.gs_apply_tombstone.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_void_p, ctypes.POINTER(GsTimeRange), ctypes.c_size_t]
+        lib.gs_scan.restype = ctypes.c_int32
+        lib.gs_scan.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                ctypes.POINTER(GsScanSpec),
+                                ctypes.POINTER(GsScanResult)]
+
+    def err(self):
+        return self.lib.gs_last_error().decode()
+
+
+def _np_ptr(arr):
+    return arr.ctypes.data_as(ctypes.c_void_p)
+
+
+# ---- host-side encoders (product write path; CPU like the reference's) ----
+
+def encode_ts(values):
+    values = np.ascontiguousarray(values, dtype=np.int64)
+    lib = PageLib().lib
+    buf = np.zeros(values.size * 9 + 64, dtype=np.uint8)
+    n = lib.gs_encode_ts(_np_ptr(values), values.size, _np_ptr(buf), buf.size)
+    if n < 0:
+        raise RuntimeError(f"gs_encode_ts failed: {n}")
+    return buf[:n].tobytes()
+
+
+def encode_i64(values):
+    values = np.ascontiguousarray(values, dtype=np.int64)
+    lib = PageLib().lib
+    buf = np.zeros(values.size * 9 + 64, dtype=np.uint8)
+    n = lib.gs_encode_i64(_np_ptr(values), values.size, _np_ptr(buf), buf.size)
+    if n < 0:
+        raise RuntimeError(f"gs_encode_i64 failed: {n}")
+    return buf[:n].tobytes()
+
+
+def encode_f64(values):
+    values = np.ascontiguousarray(values, dtype=np.float64)
+    lib = PageLib().lib
+    buf = np.zeros(values.size * 12 + 64, dtype=np.uint8)
+    n = lib.gs_encode_f64(_np_ptr(values), values.size, _np_ptr(buf), buf.size)
+    if n < 0:
+        raise RuntimeError(f"gs_encode_f64 failed: {n}")
+    return buf[:n].tobytes()
+
+
+def encode_bool(values):
+    values = np.ascontiguousarray(values, dtype=np.uint8)
+    lib = PageLib().lib
+    buf = np.zeros(values.size + 64, dtype=np.uint8)
+    n = lib.gs_encode_bool(_np_ptr(values), values.size, _np_ptr(buf), buf.size)
+    if n < 0:
+        raise RuntimeError(f"gs_encode_bool failed: {n}")
+    return buf[:n].tobytes()
+
+
+def build_page(data, nrows, bitset=None):
+    """Assemble full page bytes (tsm/page.rs layout). bitset: packed
+    LSB-first validity bytes; None -> all valid."""
+    lib = PageLib().lib
+    if bitset is None:
+        nb = (nrows + 7) // 8
+        bitset = np.full(nb, 0xFF, dtype=np.uint8)
+        if nrows % 8:
+            bitset[-1] = (1 << (nrows % 8)) - 1
+    else:
+        bitset = np.ascontiguousarray(bitset, dtype=np.uint8)
+    data = np.frombuffer(data, dtype=np.uint8) if isinstance(data, (bytes, bytearray)) else np.ascontiguousarray(data, dtype=np.uint8)
+    out = np.zeros(16 + bitset.size + data.size, dtype=np.uint8)
+    n = lib.gs_build_page(_np_ptr(bitset), nrows, _np_ptr(data), data.size,
+                          _np_ptr(out), out.size)
+    if n < 0:
+        raise RuntimeError(f"gs_build_page failed: {n}")
+    return out[:n].tobytes()
+
+
+def page_of(values, ctype, valid=None):
+    """Encode a column + assemble its page. valid: bool array or None.
+    The encoded stream holds only non-null values (page.rs/A.2)."""
+    values = np.asarray(values)
+    nrows = values.size
+    if valid is None:
+        present = values
+        bitset = None
+    else:
+        valid = np.asarray(valid, dtype=bool)
+        present = values[valid]
+        bitset = np.packbits(valid, bitorder="little")
+    if ctype == CT_TIME:
+        data = encode_ts(present.astype(np.int64))
+    elif ctype == CT_I64:
+        data = encode_i64(present.astype(np.int64))
+    elif ctype == CT_F64:
+        data = encode_f64(present.astype(np.float64))
+    elif ctype == CT_BOOL:
+        data = encode_bool(present.astype(np.uint8))
+    else:
+        raise ValueError(ctype)
+    return build_page(data, nrows, bitset)
+
+
+# ------------------------------ device engine ------------------------------
+
+class GroupSet:
+    def __init__(self, engine, handle, nrows, ngroups, keepalive):
+        self._engine = engine
+        self._h = handle
+        self.rows = nrows
+        self.ngroups = ngroups
+        self._keepalive = keepalive  # page byte buffers must outlive upload
+
+    def row_offsets(self):
+        out = np.zeros(self.ngroups, dtype=np.int64)
+        st = PageLib().lib.gs_set_row_offsets(self._h, _np_ptr(out))
+        if st != 0:
+            raise RuntimeError(PageLib().err())
+        return out
+
+    def free(self):
+        if self._h:
+            PageLib().lib.gs_groups_free(self._h)
+            self._h = None
+            self._keepalive = None
+
+
+class Engine:
+    """One GPU device context. Raises at construction if no HIP device —
+    the product path fails loudly rather than falling back to CPU."""
+
+    def __init__(self, device=0):
+        self._pl = PageLib()
+        self.lib = self._pl.lib
+        ctx = self.lib.gs_ctx_create(device)
+        if not ctx:
+            raise RuntimeError(f"gs_ctx_create failed: {self._pl.err()}")
+        self._ctx = ctypes.c_void_p(ctx)
+        self.device = device
+
+    def close(self):
+        if self._ctx:
+            self.lib.gs_ctx_destroy(self._ctx)
+            self._ctx = None
+
+    def upload(self, groups, validate_crc=True):
+        """groups: list of (series_id, [(page_bytes, ctype), ...]);
+        pages[0] must be the time page."""
+        keep = []
+        gdescs = (GsColumnGroupDesc * len(groups))()
+        for gi, (sid, pages) in enumerate(groups):
+            specs = (GsPageSpec * len(pages))()
+            keep.append(specs)
+            for pi, (pb, ctype) in enumerate(pages):
+                if isinstance(pb, (bytes, bytearray)):
+                    pb = np.frombuffer(pb, dtype=np.uint8)
+                keep.append(pb)
+                nrows = int.from_bytes(pb[4:12].tobytes(), "big")
+                specs[pi].bytes = pb.ctypes.data
+                specs[pi].len = pb.size
+                specs[pi].num_values = nrows
+                specs[pi].ctype = ctype
+                specs[pi].column_id = pi
+            gdescs[gi].pages = specs
+            gdescs[gi].npages = len(pages)
+            gdescs[gi].series_id = sid
+        h = self.lib.gs_groups_upload(self._ctx, gdescs, len(groups),
+                                      1 if validate_crc else 0)
+        if not h:
+            raise RuntimeError(f"gs_groups_upload failed: {self._pl.err()}")
+        h = ctypes.c_void_p(h)
+        rows = self.lib.gs_set_rows(h)
+        return GroupSet(self, h, rows, len(groups), keep)
+
+    def decode(self, gset, col, d_out, d_valid=None):
+        """d_out/d_valid: torch CUDA tensors (int64/float64/uint8)."""
+        st = self.lib.gs_decode(self._ctx, gset._h, col,
+                                ctypes.c_void_p(d_out.data_ptr()),
+                                ctypes.c_void_p(d_valid.data_ptr()) if d_valid is not None else None)
+        if st != 0:
+            raise RuntimeError(f"gs_decode failed ({st}): {self._pl.err()}")
+
+    def apply_tombstone(self, gset, d_ts, d_valid, ranges):
+        arr = (GsTimeRange * len(ranges))(*[GsTimeRange(a, b) for a, b in ranges])
+        st = self.lib.gs_apply_tombstone(self._ctx, gset._h,
+                                         ctypes.c_void_p(d_ts.data_ptr()),
+                                         ctypes.c_void_p(d_valid.data_ptr()),
+                                         arr, len(ranges))
+        if st != 0:
+            raise RuntimeError(f"gs_apply_tombstone failed: {self._pl.err()}")
+
+    def scan(self, gset, d_ts, d_val, time_range=None, tombstones=None,
+             d_out_ts=None, d_out_val=None, agg=None):
+        """Fused scan. agg: dict(bucket_ns, t0, n_buckets, d_max, d_sum,
+        d_count).  Returns GsScanResult."""
+        spec = GsScanSpec()
+        lo, hi = time_range if time_range else (-(2**63), 2**63 - 1)
+        spec.range = GsTimeRange(lo, hi)
+        if tombstones:
+            tarr = (GsTimeRange * len(tombstones))(*[GsTimeRange(a, b) for a, b in tombstones])
+            spec.tombstones = tarr
+            spec.n_tombstones = len(tombstones)
+        spec.d_ts = d_ts.data_ptr()
+        spec.d_val = d_val.data_ptr()
+        if d_out_ts is not None:
+            spec.d_out_ts = d_out_ts.data_ptr()
+            spec.d_out_val = d_out_val.data_ptr()
+        if agg:
+            spec.bucket_ns = agg["bucket_ns"]
+            spec.t0 = agg["t0"]
+            spec.n_buckets = agg["n_buckets"]
+            spec.d_agg_max = agg["d_max"].data_ptr()
+            spec.d_agg_sum = agg["d_sum"].data_ptr()
+            spec.d_agg_count = agg["d_count"].data_ptr()
+        res = GsScanResult()
+        st = self.lib.gs_scan(self._ctx, gset._h, ctypes.byref(spec),
+                              ctypes.byref(res))
+        if st != 0:
+            raise RuntimeError(f"gs_scan failed ({st}): {self._pl.err()}")
+        return res

@@ -1,0 +1,194 @@
+/* cnosdb_gs — C ABI of the MI355X-native TSM scan/decode engine.
+ *
+ * This is the drop-in seam described in SURVEY.md §8b: a thin host shim
+ * standing where CnosDB's `ColumnGroupReader::read` + `decode_pages` stand
+ * (tskv/src/reader/column_group/mod.rs:33-70,195-243 and
+ * tskv/src/tsm/reader.rs:494-560) calls these entry points instead of the
+ * Rust CPU codec path.  A Rust shim (cxx/bindgen) would reconstitute
+ * RecordBatches zero-copy from the output buffers; INTEGRATION.md shows the
+ * binding a cnosdb maintainer would add.
+ *
+ * Conventions:
+ *  - all functions return 0 (GS_OK) on success or a negative GsStatus;
+ *    gs_last_error() returns a thread-local message for the last failure.
+ *  - device buffers are raw HIP device pointers owned by the caller
+ *    (e.g. torch tensors); the library only allocates its internal page
+ *    store and scratch.
+ *  - one GsCtx per device; calls on a ctx are serialized on its HIP stream
+ *    and synchronized before returning.
+ */
+#ifndef CNOSDB_GS_H
+#define CNOSDB_GS_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef int32_t GsStatus;
+enum {
+    GS_OK = 0,
+    GS_ERR = -1,          /* generic; see gs_last_error() */
+    GS_ERR_NO_GPU = -2,   /* HIP device unavailable — the product path never
+                             falls back to CPU; callers must treat this as fatal */
+    GS_ERR_FORMAT = -3,   /* malformed page / unknown encoding byte */
+    GS_ERR_CRC = -4,      /* page crc32 mismatch (tsm/page.rs:58-76) */
+    GS_ERR_CAP = -5,      /* output buffer too small */
+    GS_ERR_SENTINEL = -6, /* Gorilla sentinel appeared as input (float.rs:58-60) */
+};
+
+/* Encoding byte, first byte of every encoded data buffer
+ * (common/models/src/codec.rs:39-54) */
+enum {
+    GS_ENC_NULL = 1,
+    GS_ENC_DELTA = 2,    /* i64: zigzag delta + simple8b/RLE  (codec/integer.rs) */
+    GS_ENC_GORILLA = 6,  /* f64 XOR                           (codec/float.rs) */
+    GS_ENC_BITPACK = 10, /* bool                              (codec/boolean.rs) */
+    GS_ENC_DELTATS = 11, /* ts: delta + scaled simple8b/RLE   (codec/timestamp.rs) */
+};
+
+/* physical column type of a page (PhysicalCType, tsm/reader.rs:658-731) */
+enum {
+    GS_CT_TIME = 0, /* i64 timestamps */
+    GS_CT_I64 = 1,
+    GS_CT_F64 = 2,
+    GS_CT_BOOL = 3,
+    GS_CT_U64 = 4, /* bit-cast to i64, unsigned.rs:20-45 */
+};
+
+/* Closed time interval (common/models/src/predicate/domain.rs:36-44) */
+typedef struct {
+    int64_t min_ts;
+    int64_t max_ts;
+} GsTimeRange;
+
+/* One on-disk page: bytes = [u32 BE bitset_len][u64 BE row_count]
+ * [u32 BE crc32(data)][validity bitset, LSB-first][encoded data]
+ * (tsm/page.rs:32-94); num_values mirrors PageMeta.num_values. */
+typedef struct {
+    const uint8_t *bytes;
+    uint64_t len;
+    uint32_t num_values;
+    uint8_t ctype;    /* GS_CT_* */
+    uint32_t column_id;
+} GsPageSpec;
+
+/* One column group = one series' time slab: a time page plus field pages,
+ * all with the same row count (tsm/column_group.rs:10-17). */
+typedef struct {
+    const GsPageSpec *pages; /* pages[0] MUST be the time page */
+    uint32_t npages;
+    uint32_t series_id;
+} GsColumnGroupDesc;
+
+typedef struct GsCtx GsCtx;
+typedef struct GsGroupSet GsGroupSet;
+
+/* ---- context ---- */
+GsCtx *gs_ctx_create(int device);
+void gs_ctx_destroy(GsCtx *ctx);
+const char *gs_last_error(void);
+int gs_device_count(void);
+const char *gs_version(void);
+
+/* ---- write path (host): the re-encode side of the seam
+ * (Page::arrow_array_to_page, tsm/page.rs:100-353 + codec encode fns).
+ * Return encoded length, or a negative GsStatus. ---- */
+int64_t gs_encode_ts(const int64_t *src, size_t n, uint8_t *dst, size_t cap);
+int64_t gs_encode_i64(const int64_t *src, size_t n, uint8_t *dst, size_t cap);
+int64_t gs_encode_f64(const double *src, size_t n, uint8_t *dst, size_t cap);
+int64_t gs_encode_bool(const uint8_t *src, size_t n, uint8_t *dst, size_t cap);
+/* assemble a full page (header + crc + bitset + data), tsm/page.rs:488-497 */
+int64_t gs_build_page(const uint8_t *bitset, int64_t nrows, const uint8_t *data,
+                      size_t data_len, uint8_t *dst, size_t cap);
+uint32_t gs_crc32(const uint8_t *data, size_t len); /* CRC-32/ISO-HDLC */
+
+/* ---- device page store ----
+ * Uploads raw page bytes of `ngroups` column groups to HBM and builds the
+ * device-side page table.  validate_crc runs the crc32 check of
+ * Page::crc_validation on upload (host-side).  Row layout: group g's rows
+ * occupy [row_offset[g], row_offset[g] + num_values_g) in every output
+ * column buffer; gs_set_rows returns the total. */
+GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
+                             size_t ngroups, int validate_crc);
+void gs_groups_free(GsGroupSet *set);
+int64_t gs_set_rows(const GsGroupSet *set);
+int64_t gs_set_groups(const GsGroupSet *set);
+/* copy the per-group row offsets (ngroups entries) into caller buffer */
+GsStatus gs_set_row_offsets(const GsGroupSet *set, int64_t *out);
+
+/* ---- decode (hot loop ①, tsm/reader.rs:494-560) ----
+ * Decodes column slot `col` (index into each group's pages[]) of every
+ * group into d_out (device, 8 B/row for i64/f64/u64, 1 B for bool) at the
+ * group row offsets.  Null slots are written as 0 (arrow builder
+ * append_null semantics).  d_valid (device, 1 B/row, may be NULL) receives
+ * 1 for valid rows, 0 for null rows. */
+GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
+                   uint8_t *d_valid);
+
+/* ---- tombstone masking (tsm/reader.rs:634-656) ----
+ * Clears validity (in d_valid) for rows whose decoded timestamp (d_ts,
+ * from gs_decode of the time page) falls in any deleted closed range.
+ * Applied per group; `ranges` apply to every group in the set. */
+GsStatus gs_apply_tombstone(GsCtx *ctx, GsGroupSet *set, const int64_t *d_ts,
+                            uint8_t *d_valid, const GsTimeRange *ranges,
+                            size_t nranges);
+
+/* ---- fused time-range scan (hot loops ①+②+④) ----
+ * For the TSBS-devops shape: schema = time page + one f64 field page per
+ * group (col 1).  Pipeline per group, entirely on device:
+ *   decode ts -> decode f64 -> closed-interval time filter
+ *   -> EITHER compacted output (d_out_ts/d_out_val at gather offsets)
+ *      OR per-bucket aggregates (max/sum/count, fused, no row output).
+ * Mirrors TableScanStream semantics for a pure time-range predicate
+ * (data_source/batch/tskv.rs:351-371 pushes exactly these) with the
+ * downsampling aggregate that stock DataFusion would run above
+ * (SURVEY.md §8a row "downsampling agg").
+ */
+typedef struct {
+    GsTimeRange range;     /* closed; use INT64_MIN/MAX for no filter */
+    /* deleted time ranges (tombstones) applied to the field column's
+       validity before filter/agg (tsm/reader.rs:529-544) */
+    const GsTimeRange *tombstones;
+    size_t n_tombstones;
+    /* aggregate spec; n_buckets == 0 disables aggregation */
+    int64_t bucket_ns;     /* e.g. 300_000_000_000 for 5-min buckets */
+    int64_t t0;            /* bucket origin: bucket = (ts - t0) / bucket_ns */
+    int32_t n_buckets;
+    /* device outputs for aggregation (caller-allocated, e.g. torch):
+       d_max must be pre-filled with -inf, d_sum/d_count with 0 */
+    double *d_agg_max;
+    double *d_agg_sum;
+    long long *d_agg_count;
+    /* device outputs for compaction; NULL to skip row materialization.
+       capacity must be >= gs_set_rows(set) rows. */
+    int64_t *d_out_ts;
+    double *d_out_val;
+    /* scratch/outputs the caller provides (device):
+       d_ts: decoded time column (gs_set_rows rows)
+       d_val: decoded f64 column (gs_set_rows rows) */
+    int64_t *d_ts;
+    double *d_val;
+} GsScanSpec;
+
+typedef struct {
+    int64_t out_rows;      /* rows selected by the time filter */
+    int64_t decoded_rows;  /* total rows decoded (= gs_set_rows) */
+    /* per-phase kernel times, ms, measured with HIP events on the engine
+       stream (for roofline accounting) */
+    double ms_decode_ts;
+    double ms_decode_val;
+    double ms_filter;
+    double ms_compact;
+    double ms_agg;
+} GsScanResult;
+
+GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
+                 GsScanResult *result);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* CNOSDB_GS_H */

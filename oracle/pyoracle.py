@@ -1,0 +1,175 @@
+"""ORACLE — TEST INFRASTRUCTURE ONLY (see tsm_oracle.c header).
+
+ctypes bindings for liboracle.so plus the numpy restatement of the
+scan-level semantics (filter/aggregate/tombstone) used to check the GPU
+product path.  May be imported ONLY from tests/, __graft_entry__.smoke()
+and bench.py's cpu_baseline leg.
+"""
+import ctypes
+import os
+import subprocess
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "liboracle.so")
+
+
+def _ensure_built():
+    if not os.path.exists(_SO):
+        subprocess.check_call(["make", "-C", _DIR])
+
+
+class Oracle:
+    _inst = None
+
+    def __new__(cls):
+        if cls._inst is None:
+            cls._inst = super().__new__(cls)
+            cls._inst._load()
+        return cls._inst
+
+    def _load(self):
+        _ensure_built()
+        lib = ctypes.CDLL(_SO)
+        self.lib = lib
+        for nm in ("orc_ts_encode", "orc_i64_encode", "orc_f64_encode",
+                   "orc_bool_encode", "orc_null_encode_i64", "orc_build_page"):
+            getattr(lib, nm).restype = ctypes.c_int64
+        lib.orc_crc32.restype = ctypes.c_uint32
+        for nm in ("orc_decode_i64", "orc_decode_f64", "orc_decode_bool",
+                   "orc_page_check", "orc_decode_pages_omp"):
+            getattr(lib, nm).restype = ctypes.c_int32
+
+
+def _ptr(a):
+    return a.ctypes.data_as(ctypes.c_void_p)
+
+
+def _bitset(nrows, valid=None):
+    if valid is None:
+        nb = (nrows + 7) // 8
+        bs = np.full(nb, 0xFF, dtype=np.uint8)
+        if nrows % 8:
+            bs[-1] = (1 << (nrows % 8)) - 1
+        return bs
+    return np.packbits(np.asarray(valid, dtype=bool), bitorder="little")
+
+
+def encode_ts(values):
+    o = Oracle().lib
+    v = np.ascontiguousarray(values, dtype=np.int64)
+    buf = np.zeros(v.size * 9 + 64, dtype=np.uint8)
+    n = o.orc_ts_encode(_ptr(v), v.size, _ptr(buf), buf.size)
+    assert n >= 0, n
+    return buf[:n].tobytes()
+
+
+def encode_i64(values):
+    o = Oracle().lib
+    v = np.ascontiguousarray(values, dtype=np.int64)
+    buf = np.zeros(v.size * 9 + 64, dtype=np.uint8)
+    n = o.orc_i64_encode(_ptr(v), v.size, _ptr(buf), buf.size)
+    assert n >= 0, n
+    return buf[:n].tobytes()
+
+
+def encode_f64(values):
+    o = Oracle().lib
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    buf = np.zeros(v.size * 12 + 64, dtype=np.uint8)
+    n = o.orc_f64_encode(_ptr(v), v.size, _ptr(buf), buf.size)
+    assert n >= 0, n
+    return buf[:n].tobytes()
+
+
+def encode_bool(values):
+    o = Oracle().lib
+    v = np.ascontiguousarray(values, dtype=np.uint8)
+    buf = np.zeros(v.size + 64, dtype=np.uint8)
+    n = o.orc_bool_encode(_ptr(v), v.size, _ptr(buf), buf.size)
+    assert n >= 0, n
+    return buf[:n].tobytes()
+
+
+def decode_i64(data, nrows, valid=None):
+    """Mirror of *_decode_to_array for ts/i64 pages: returns int64 array of
+    nrows with null slots 0 (arrow builder semantics)."""
+    o = Oracle().lib
+    d = np.frombuffer(data, dtype=np.uint8)
+    bs = _bitset(nrows, valid)
+    out = np.zeros(nrows, dtype=np.int64)
+    st = o.orc_decode_i64(_ptr(d), d.size, _ptr(bs), nrows, _ptr(out))
+    if st != 0:
+        raise RuntimeError(f"orc_decode_i64: {st}")
+    return out
+
+
+def decode_f64(data, nrows, valid=None):
+    o = Oracle().lib
+    d = np.frombuffer(data, dtype=np.uint8)
+    bs = _bitset(nrows, valid)
+    out = np.zeros(nrows, dtype=np.float64)
+    st = o.orc_decode_f64(_ptr(d), d.size, _ptr(bs), nrows, _ptr(out))
+    if st != 0:
+        raise RuntimeError(f"orc_decode_f64: {st}")
+    return out
+
+
+def decode_bool(data, nrows, valid=None):
+    o = Oracle().lib
+    d = np.frombuffer(data, dtype=np.uint8)
+    bs = _bitset(nrows, valid)
+    out = np.zeros(nrows, dtype=np.uint8)
+    st = o.orc_decode_bool(_ptr(d), d.size, _ptr(bs), nrows, _ptr(out))
+    if st != 0:
+        raise RuntimeError(f"orc_decode_bool: {st}")
+    return out
+
+
+def crc32(data):
+    d = np.frombuffer(data, dtype=np.uint8)
+    return Oracle().lib.orc_crc32(_ptr(d), d.size)
+
+
+# ------------- scan-level restatement (numpy; cites tskv sources) ----------
+
+def update_nullbits(ts, ranges, valid):
+    """tsm/reader.rs:634-656: clear validity for rows whose ts lies in a
+    deleted CLOSED range. ts sorted; valid is a bool array (modified copy
+    returned)."""
+    valid = np.asarray(valid, dtype=bool).copy()
+    ts = np.asarray(ts)
+    for mn, mx in ranges:
+        start = np.searchsorted(ts, mn, side="left")
+        i = np.searchsorted(ts, mx, side="left")
+        end = i + 1 if i < ts.size and ts[i] == mx else i
+        valid[start:end] = False
+    return valid
+
+
+def time_span(ts, mn, mx):
+    """closed-interval [mn,mx] span on sorted ts (TimeRange semantics,
+    common/models/src/predicate/domain.rs:36-44)."""
+    s = np.searchsorted(ts, mn, side="left")
+    e = np.searchsorted(ts, mx, side="right")
+    return s, e - s
+
+
+def bucket_agg(ts, vals, valid, t0, bucket_ns, n_buckets):
+    """stock-DataFusion-style per-bucket max/sum/count over non-null rows
+    (the downsampling aggregate run above TskvExec)."""
+    ts = np.asarray(ts)
+    vals = np.asarray(vals)
+    m = np.asarray(valid, dtype=bool) if valid is not None else np.ones(ts.size, bool)
+    b = (ts - t0) // bucket_ns
+    m = m & (b >= 0) & (b < n_buckets)
+    b = b[m]
+    v = vals[m]
+    mx = np.full(n_buckets, -np.inf)
+    sm = np.zeros(n_buckets)
+    ct = np.zeros(n_buckets, dtype=np.int64)
+    np.maximum.at(mx, b, v)
+    np.add.at(sm, b, v)
+    np.add.at(ct, b, 1)
+    return mx, sm, ct
