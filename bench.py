@@ -177,7 +177,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--series", type=int, default=10000)
     ap.add_argument("--npts", type=int, default=1_000_000)
-    ap.add_argument("--page-rows", type=int, default=125_000)
+    ap.add_argument("--page-rows", type=int, default=15_625)
     ap.add_argument("--sub-batches", type=int, default=4)
     ap.add_argument("--unique", type=int, default=256)
     ap.add_argument("--skip-cpu-baseline", action="store_true")

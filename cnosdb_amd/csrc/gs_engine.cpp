@@ -126,49 +126,53 @@ struct DevS8b {
     }
 };
 
-/* Gorilla bit reader (float.rs:418-606): 64-bit rotate-left cursor over a
- * BE byte stream, tail refill places valid bits at the top. */
+/* Gorilla bit reader: MSB-first over the BE byte stream, semantically
+ * identical bit order to the reference's rotate-left cursor
+ * (float.rs:418-606) but held as a 128-bit MSB-aligned window so a field
+ * read is one compare + shift pair instead of the reference's split-read
+ * masking — the hot-kernel form of the same contract. */
 struct DevBitReader {
     const uint8_t *src;
-    uint32_t len, i;
-    uint64_t cache;
-    uint32_t valid;
-    __device__ bool refill() {
-        uint32_t rem = len - i;
-        if (rem >= 8) {
-            cache = dev_be64(src + i);
-            valid = 64;
-            i += 8;
-            return true;
-        }
-        if (rem > 0) {
-            uint64_t v = 0;
-            for (uint32_t k = i; k < len; k++) v = (v << 8) | src[k];
-            unsigned bits = rem * 8;
-            cache = (v >> bits) | (v << (64 - bits));
-            valid = bits;
-            i = len;
-            return true;
-        }
-        return false;
+    uint32_t len, pos; /* byte cursor */
+    uint64_t hi, lo;   /* valid bits MSB-aligned across hi then lo */
+    int n;             /* number of valid bits in the window */
+    __device__ void init(const uint8_t *s, uint32_t l, uint32_t p) {
+        src = s; len = l; pos = p; hi = 0; lo = 0; n = 0;
     }
-    __device__ bool read(unsigned cnt, uint64_t *out) {
-        if (valid == 0 && !refill()) return false;
-        if (valid >= cnt) {
-            valid -= cnt;
-            cache = dev_rotl64(cache, cnt);
-            *out = cache;
-            return true;
+    __device__ __forceinline__ void insert_word() {
+        /* pull the next <=8 bytes into the window below the n valid bits
+           (call only with n <= 64) */
+        uint32_t rem = len - pos;
+        uint64_t x;
+        int bits;
+        if (__builtin_expect(rem >= 8, 1)) {
+            x = dev_be64(src + pos);
+            pos += 8;
+            bits = 64;
+        } else if (rem > 0) {
+            x = 0;
+            for (uint32_t k = pos; k < len; k++) x = (x << 8) | src[k];
+            x <<= (8 - rem) * 8;
+            bits = int(rem) * 8;
+            pos = len;
+        } else {
+            return;
         }
-        unsigned m = cnt;
-        uint64_t bits = 0;
-        if (valid > 0) { m -= valid; bits = dev_rotl64(cache, cnt); }
-        if (!refill()) return false;
-        cache = dev_rotl64(cache, m);
-        valid -= m;
-        uint64_t mask = (m & 0x3f) ? ((1ULL << (m & 0x3f)) - 1) : ~0ULL;
-        bits &= ~mask;
-        bits |= cache & mask;
+        if (n == 0) { hi = x; lo = 0; }
+        else { hi |= x >> n; lo = x << (64 - n); }
+        n += bits;
+    }
+    /* read cnt (1..=64) MSB-first bits into the LOW bits of *out.
+       false = stream exhausted (reference: "unexpected end of block") */
+    __device__ __forceinline__ bool read(unsigned cnt, uint64_t *out) {
+        while (n < int(cnt)) {
+            if (pos >= len) return false;
+            insert_word();
+        }
+        uint64_t bits = (cnt == 64) ? hi : (hi >> (64 - cnt));
+        if (cnt == 64) { hi = lo; lo = 0; }
+        else { hi = (hi << cnt) | (lo >> (64 - cnt)); lo <<= cnt; }
+        n -= int(cnt);
         *out = bits;
         return true;
     }
@@ -337,8 +341,8 @@ __global__ void k_seq_f64(const uint8_t *__restrict__ blob,
         uint32_t slen = pg.data_len - 1;
         if (slen < 9) { atomicOr(err, DERR_SHORT); continue; }
         uint64_t val = dev_be64(s + 1);
-        DevBitReader br{s, slen, 9, 0, 0};
-        if (!br.refill()) { atomicOr(err, DERR_SHORT); continue; }
+        DevBitReader br;
+        br.init(s, slen, 9);
         uint32_t trailing_n = 0, meaningful_n = 64;
         uint32_t r = 0;
         bool bad = false;
@@ -541,11 +545,14 @@ __global__ void k_compact(const DevGroup *__restrict__ groups, int n,
  * BASELINE config #3 requires max exact and sum within stated tolerance,
  * and the reference's DataFusion hash-agg is sequential per partition.
  *
- * Phase 1 (k_agg_partial): block per group.  The decoded ts of a group is
- * sorted, so each bucket's rows form a contiguous range found by binary
- * search; the block's waves split the bucket list, each wave reduces its
- * bucket's rows with a shuffle tree and writes the (group,bucket) partial.
- * Every (group,bucket) cell is written exactly once — no init needed.
+ * Phase 1 (k_agg_partial): block per AGGREGATION GROUP (consecutive
+ * page-groups of one series merged at upload — their rows are contiguous
+ * and time-ordered, tsm/column_group.rs:55-63).  The decoded ts of a
+ * group is sorted, so each bucket's rows form a contiguous range found by
+ * binary search (clipped to the scan's closed time range); the block's
+ * waves split the bucket list, each wave reduces its bucket's rows with a
+ * shuffle tree and writes the (group,bucket) partial.  Every cell is
+ * written exactly once — no init needed, no atomics anywhere.
  *
  * Phase 2 (k_agg_merge): one wave per bucket strides the groups, reducing
  * partials in a fixed lane order, and accumulates into the caller's
@@ -555,8 +562,7 @@ __global__ void k_agg_partial(const DevGroup *__restrict__ groups, int n,
                               const int64_t *__restrict__ ts,
                               const double *__restrict__ val,
                               const uint8_t *__restrict__ valid,
-                              const int64_t *__restrict__ sp_start,
-                              const int64_t *__restrict__ sp_cnt, int64_t t0,
+                              int64_t range_lo, int64_t range_hi, int64_t t0,
                               int64_t bucket_ns, int nbuckets,
                               double *__restrict__ pmax,
                               double *__restrict__ psum,
@@ -564,22 +570,50 @@ __global__ void k_agg_partial(const DevGroup *__restrict__ groups, int n,
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int nwaves = blockDim.x >> 6;
+    /* bucket-boundary row offsets staged in LDS: one cooperative binary
+       search per boundary (256 in flight per block) instead of 2 dependent
+       searches per (wave,bucket) — the searches were the latency hot spot */
+    constexpr int MAXB = 8192;
+    __shared__ uint32_t soff[MAXB + 1];
+    const bool use_lds = nbuckets <= MAXB;
+    const int64_t hi_open = (range_hi == INT64_MAX) ? INT64_MAX : range_hi + 1;
     for (int g = blockIdx.x; g < n; g += gridDim.x) {
-        const int64_t base = groups[g].row_off + sp_start[g];
-        const int64_t cnt = sp_cnt[g];
+        const int64_t base = groups[g].row_off;
+        const int64_t cnt = groups[g].nrows;
         const int64_t *t = ts + base;
         const double *v = val + base;
         const uint8_t *vd = valid ? valid + base : nullptr;
+        if (use_lds) {
+            for (int bi = threadIdx.x; bi <= nbuckets; bi += blockDim.x) {
+                int64_t bound = t0 + int64_t(bi) * bucket_ns;
+                if (bound < range_lo) bound = range_lo;
+                if (bound > hi_open) bound = hi_open;
+                int64_t lo = 0, hi = cnt;
+                while (lo < hi) {
+                    int64_t m = (lo + hi) >> 1;
+                    if (t[m] < bound) lo = m + 1; else hi = m;
+                }
+                soff[bi] = uint32_t(lo);
+            }
+            __syncthreads();
+        }
         for (int b = wave; b < nbuckets; b += nwaves) {
-            /* row range of bucket b within the span (ts sorted) */
-            int64_t blo = t0 + int64_t(b) * bucket_ns;
-            int64_t bhi = blo + bucket_ns;
-            int64_t lo = 0, hi = cnt;
-            while (lo < hi) { int64_t m = (lo + hi) >> 1; if (t[m] < blo) lo = m + 1; else hi = m; }
-            int64_t s = lo;
-            lo = s; hi = cnt;
-            while (lo < hi) { int64_t m = (lo + hi) >> 1; if (t[m] < bhi) lo = m + 1; else hi = m; }
-            int64_t e = lo;
+            int64_t s, e;
+            if (use_lds) {
+                s = soff[b];
+                e = soff[b + 1];
+            } else {
+                int64_t blo = t0 + int64_t(b) * bucket_ns;
+                int64_t bhi = blo + bucket_ns;
+                if (blo < range_lo) blo = range_lo;
+                int64_t bhi_cl = (bhi < hi_open) ? bhi : hi_open;
+                int64_t lo = 0, hi = cnt;
+                while (lo < hi) { int64_t m = (lo + hi) >> 1; if (t[m] < blo) lo = m + 1; else hi = m; }
+                s = lo;
+                lo = s; hi = cnt;
+                while (lo < hi) { int64_t m = (lo + hi) >> 1; if (t[m] < bhi_cl) lo = m + 1; else hi = m; }
+                e = lo;
+            }
             double mx = -__builtin_inf(), sm = 0.0;
             long long c = 0;
             for (int64_t r = s + lane; r < e; r += 64) {
@@ -605,6 +639,7 @@ __global__ void k_agg_partial(const DevGroup *__restrict__ groups, int n,
                 pcnt[idx] = c;
             }
         }
+        if (use_lds) __syncthreads();
     }
 }
 
@@ -670,6 +705,9 @@ struct GsGroupSet {
     std::vector<int64_t> row_offsets;
     std::vector<SlotPages> slots;
     DevGroup *d_groups = nullptr;
+    DevGroup *d_sgroups = nullptr; /* series-level (consecutive same-series
+                                      groups merged) for aggregation */
+    int nsgroups = 0;
     int64_t *d_sp_start = nullptr;
     int64_t *d_sp_cnt = nullptr;
     int64_t *d_out_off = nullptr;
@@ -886,6 +924,24 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         hg[g].nrows = int32_t(groups[g].pages[0].num_values);
         hg[g].pad = 0;
     }
+    /* series-level groups: merge consecutive page-groups of one series
+       (their rows are contiguous and time-ordered) */
+    std::vector<DevGroup> hsg;
+    for (size_t g = 0; g < ngroups; g++) {
+        if (!hsg.empty() && g > 0 &&
+            groups[g].series_id == groups[g - 1].series_id) {
+            hsg.back().nrows += hg[g].nrows;
+        } else {
+            hsg.push_back(hg[g]);
+        }
+    }
+    set->nsgroups = int(hsg.size());
+    if (hipMalloc(&set->d_sgroups, hsg.size() * sizeof(DevGroup)) != hipSuccess) {
+        fail(GS_ERR, "hipMalloc sgroup table failed");
+        delete set; return nullptr;
+    }
+    hipMemcpyAsync(set->d_sgroups, hsg.data(), hsg.size() * sizeof(DevGroup),
+                   hipMemcpyHostToDevice, ctx->stream);
     if (hipMalloc(&set->d_groups, ngroups * sizeof(DevGroup)) != hipSuccess ||
         hipMalloc(&set->d_sp_start, ngroups * sizeof(int64_t)) != hipSuccess ||
         hipMalloc(&set->d_sp_cnt, ngroups * sizeof(int64_t)) != hipSuccess ||
@@ -908,6 +964,7 @@ void gs_groups_free(GsGroupSet *set) {
             if (sp.dev[k]) hipFree(sp.dev[k]);
     hipFree(set->d_blob);
     hipFree(set->d_groups);
+    if (set->d_sgroups) hipFree(set->d_sgroups);
     hipFree(set->d_sp_start);
     hipFree(set->d_sp_cnt);
     hipFree(set->d_out_off);
@@ -1077,7 +1134,8 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
     if (spec->n_buckets > 0) {
         if (!spec->d_agg_max || !spec->d_agg_sum || !spec->d_agg_count)
             return fail(GS_ERR, "agg outputs missing");
-        size_t cells = size_t(ng) * size_t(spec->n_buckets);
+        int nsg = set->nsgroups;
+        size_t cells = size_t(nsg) * size_t(spec->n_buckets);
         if (set->partials_cap < cells) {
             if (set->d_pmax) hipFree(set->d_pmax);
             if (set->d_psum) hipFree(set->d_psum);
@@ -1088,15 +1146,16 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                 return fail(GS_ERR, "hipMalloc agg partials failed");
             set->partials_cap = cells;
         }
-        hipLaunchKernelGGL(k_agg_partial, dim3(ng > 2048 ? 2048 : ng),
-                           dim3(256), 0, ctx->stream, set->d_groups, ng,
-                           spec->d_ts, spec->d_val, d_valid, set->d_sp_start,
-                           set->d_sp_cnt, spec->t0, spec->bucket_ns,
+        hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 2048 ? 2048 : nsg),
+                           dim3(256), 0, ctx->stream, set->d_sgroups, nsg,
+                           spec->d_ts, spec->d_val, d_valid,
+                           spec->range.min_ts, spec->range.max_ts,
+                           spec->t0, spec->bucket_ns,
                            spec->n_buckets, set->d_pmax, set->d_psum,
                            set->d_pcnt);
         int mb = (spec->n_buckets + 3) / 4;
         hipLaunchKernelGGL(k_agg_merge, dim3(mb > 2048 ? 2048 : mb), dim3(256),
-                           0, ctx->stream, ng, spec->n_buckets, set->d_pmax,
+                           0, ctx->stream, nsg, spec->n_buckets, set->d_pmax,
                            set->d_psum, set->d_pcnt, spec->d_agg_max,
                            spec->d_agg_sum, spec->d_agg_count);
     }
