@@ -303,7 +303,86 @@ __global__ void k_seq_i64(const uint8_t *__restrict__ blob,
 
 /* Gorilla f64 pages: one thread per page (the bitstream carries a strict
  * sequential dependency — float.rs:445-463; parallelism comes from the
- * page count).  Null-free fast path avoids the bitset walk. */
+ * page count).  The hot loop is a flat 128-bit-window parser: one 13-bit
+ * peek covers ctrl bits + leading/meaningful, whole-word refills run over
+ * the blob's tail pad with a bit-budget guard standing in for the
+ * reference's end-of-block refill error — identical bit consumption and
+ * identical error surface on truncated streams. */
+template <bool ALLVALID>
+__device__ __forceinline__ void gorilla_decode_page(
+    const uint8_t *data, uint32_t data_len, const uint8_t *bs, uint32_t n,
+    double *o, uint8_t *vd, unsigned *err) {
+    const uint8_t *s = data + 1;
+    uint32_t slen = data_len - 1;
+    if (slen < 9) { atomicOr(err, DERR_SHORT); return; }
+    uint64_t val = dev_be64(s + 1);
+    const uint8_t *p = s + 9;
+    int64_t budget = int64_t(slen - 9) * 8; /* bits the stream really holds */
+    uint64_t hi = 0, lo = 0;
+    int nb = 0; /* valid window bits (may exceed budget over the pad) */
+    uint32_t trailing_n = 0, meaningful_n = 64;
+    uint32_t r = 0;
+    bool bad = false;
+
+    auto emit = [&](uint64_t bits_) {
+        if (!ALLVALID) {
+            while (r < n && !dev_bit(bs, r)) { o[r] = 0.0; if (vd) vd[r] = 0; r++; }
+        }
+        if (r < n) {
+            o[r] = __longlong_as_double(int64_t(bits_));
+            if (vd) vd[r] = 1;
+            r++;
+        }
+    };
+    auto topup = [&]() { /* call only with nb <= 64 */
+        uint64_t x = dev_be64(p);
+        p += 8;
+        if (nb == 0) { hi = x; lo = 0; }
+        else { hi |= x >> nb; lo = x << (64 - nb); }
+        nb += 64;
+    };
+    auto consume = [&](unsigned k) { /* k in 1..=64 */
+        hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
+        lo = (k == 64) ? 0 : (lo << k);
+        nb -= int(k);
+        budget -= int64_t(k);
+    };
+
+    emit(val);
+    for (;;) {
+        if (nb <= 64) topup(); /* >= 65 bits: enough for ctrl+meta (13) */
+        if (budget <= 0) { bad = true; break; }
+        uint32_t top13 = uint32_t(hi >> 51);
+        if (!(top13 & 0x1000)) { /* ctrl 0: repeat */
+            consume(1);
+            emit(val);
+            continue;
+        }
+        if (top13 & 0x0800) { /* ctrl 11: new window */
+            uint32_t lead = (top13 >> 6) & 0x1f;
+            meaningful_n = top13 & 0x3f;
+            if (meaningful_n > 0) trailing_n = 64 - lead - meaningful_n;
+            else { trailing_n = 0; meaningful_n = 64; }
+            consume(13);
+        } else { /* ctrl 10: reuse window */
+            consume(2);
+        }
+        while (nb < int(meaningful_n)) topup();
+        uint64_t sbits = (meaningful_n == 64) ? hi : (hi >> (64 - meaningful_n));
+        consume(meaningful_n);
+        if (budget < 0) { bad = true; break; }
+        val ^= sbits << trailing_n;
+        if (val == GORILLA_SENTINEL) break;
+        emit(val);
+    }
+    if (bad) { atomicOr(err, DERR_SHORT); return; }
+    for (; r < n; r++) {
+        if (ALLVALID || dev_bit(bs, r)) { atomicOr(err, DERR_SHORT); break; }
+        o[r] = 0.0;
+        if (vd) vd[r] = 0;
+    }
+}
+
 __global__ void k_seq_f64(const uint8_t *__restrict__ blob,
                           const DevPage *__restrict__ pages, int npages,
                           double *__restrict__ out,
@@ -337,55 +416,10 @@ __global__ void k_seq_f64(const uint8_t *__restrict__ blob,
             continue;
         }
         if (enc != GS_ENC_GORILLA) { atomicOr(err, DERR_FORMAT); continue; }
-        const uint8_t *s = data + 1;
-        uint32_t slen = pg.data_len - 1;
-        if (slen < 9) { atomicOr(err, DERR_SHORT); continue; }
-        uint64_t val = dev_be64(s + 1);
-        DevBitReader br;
-        br.init(s, slen, 9);
-        uint32_t trailing_n = 0, meaningful_n = 64;
-        uint32_t r = 0;
-        bool bad = false;
-        /* emit helper: scatter through validity */
-        auto emit = [&](uint64_t bits_) {
-            if (!pg.all_valid) {
-                while (r < n && !dev_bit(bs, r)) { o[r] = 0.0; if (vd) vd[r] = 0; r++; }
-            }
-            if (r < n) {
-                o[r] = __longlong_as_double(int64_t(bits_));
-                if (vd) vd[r] = 1;
-                r++;
-            }
-        };
-        emit(val);
-        for (;;) {
-            uint64_t b;
-            if (!br.read(1, &b)) { bad = true; break; }
-            if ((b & 1) == 0) { emit(val); continue; }
-            if (!br.read(1, &b)) { bad = true; break; }
-            if (b & 1) {
-                uint64_t lm;
-                if (!br.read(11, &lm)) { bad = true; break; }
-                lm &= 0x7ff;
-                uint32_t leading_n = uint32_t(lm >> 6) & 0x1f;
-                meaningful_n = uint32_t(lm & 0x3f);
-                if (meaningful_n > 0) trailing_n = 64 - leading_n - meaningful_n;
-                else { trailing_n = 0; meaningful_n = 64; }
-            }
-            uint64_t sbits;
-            if (!br.read(meaningful_n, &sbits)) { bad = true; break; }
-            if (meaningful_n & 0x3f) sbits &= (1ULL << (meaningful_n & 0x3f)) - 1;
-            val ^= sbits << (trailing_n & 0x3f);
-            if (val == GORILLA_SENTINEL) break;
-            emit(val);
-        }
-        if (bad) { atomicOr(err, DERR_SHORT); continue; }
-        /* remaining rows must be null */
-        for (; r < n; r++) {
-            if (pg.all_valid || dev_bit(bs, r)) { atomicOr(err, DERR_SHORT); break; }
-            o[r] = 0.0;
-            if (vd) vd[r] = 0;
-        }
+        if (pg.all_valid)
+            gorilla_decode_page<true>(data, pg.data_len, bs, n, o, vd, err);
+        else
+            gorilla_decode_page<false>(data, pg.data_len, bs, n, o, vd, err);
     }
 }
 
@@ -850,7 +884,9 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
             hp.dp.data_off = blob;
             blob += data_len;
             blob = align16(blob);
-            blob += 16; /* tail pad so unaligned 8-B loads never run off */
+            blob += 32; /* tail pad: the Gorilla fast path reads whole
+                           words up to ~24 B past the data end (guarded by
+                           its bit budget, garbage never interpreted) */
             hp.dp.row_off = rows;
             hp.dp.data_len = uint32_t(data_len);
             hp.dp.nrows = ps.num_values;
