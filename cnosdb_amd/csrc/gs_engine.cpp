@@ -334,7 +334,7 @@ __device__ __forceinline__ void gorilla_decode_page(
             r++;
         }
     };
-    auto topup = [&]() { /* call only with nb <= 64 */
+    auto topup = [&]() { /* call only with nb < 64 (x >> 64 is UB/mod-64) */
         uint64_t x = dev_be64(p);
         p += 8;
         if (nb == 0) { hi = x; lo = 0; }
@@ -350,7 +350,7 @@ __device__ __forceinline__ void gorilla_decode_page(
 
     emit(val);
     for (;;) {
-        if (nb <= 64) topup(); /* >= 65 bits: enough for ctrl+meta (13) */
+        if (nb < 64) topup(); /* >= 64 bits: enough for any single field */
         if (budget <= 0) { bad = true; break; }
         uint32_t top13 = uint32_t(hi >> 51);
         if (!(top13 & 0x1000)) { /* ctrl 0: repeat */
