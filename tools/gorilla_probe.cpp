@@ -1,0 +1,199 @@
+/* Standalone ablation probe for the Gorilla decode kernel on gfx950.
+ * Variants isolate compute vs store vs load cost. Build:
+ *   hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/gorilla_probe.cpp \
+ *         cnosdb_amd/csrc/gs_encode.cpp -o tools/gorilla_probe
+ */
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdint>
+#include <cstring>
+#include <vector>
+#include <cmath>
+#include <random>
+
+extern "C" int64_t gs_encode_f64(const double *, size_t, uint8_t *, size_t);
+
+#define SENT 0x7ff8000000000ffULL
+
+__device__ __forceinline__ uint64_t dbe64(const uint8_t *p) {
+    uint64_t v;
+    __builtin_memcpy(&v, p, 8);
+    return __builtin_bswap64(v);
+}
+
+struct PD { uint64_t off; uint32_t len; uint64_t row; };
+
+template <int MODE> /* 0=full 1=nostore 2=2pages-interleaved */
+__global__ void k_gor(const uint8_t *__restrict__ blob,
+                      const PD *__restrict__ pages, int npages,
+                      double *__restrict__ out, unsigned *__restrict__ err) {
+    for (int p0 = blockIdx.x * blockDim.x + threadIdx.x; p0 < npages;
+         p0 += gridDim.x * blockDim.x) {
+        PD pg = pages[p0];
+        const uint8_t *data = blob + pg.off;
+        double *o = out + pg.row;
+        const uint8_t *s = data + 1;
+        uint32_t slen = pg.len - 1;
+        uint64_t val = dbe64(s + 1);
+        const uint8_t *p = s + 9;
+        int64_t budget = int64_t(slen - 9) * 8;
+        uint64_t hi = 0, lo = 0;
+        int nb = 0;
+        uint32_t trailing = 0, meaningful = 64;
+        uint32_t r = 0;
+        uint64_t acc = 0;
+        uint64_t nextw = dbe64(p);
+        p += 8;
+        auto topup = [&]() {
+            uint64_t x = nextw;
+            nextw = dbe64(p);
+            p += 8;
+            if (nb == 0) { hi = x; lo = 0; }
+            else { hi |= x >> nb; lo = x << (64 - nb); }
+            nb += 64;
+        };
+        auto consume = [&](unsigned k) {
+            hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
+            lo = (k == 64) ? 0 : (lo << k);
+            nb -= int(k);
+            budget -= int64_t(k);
+        };
+        auto emit = [&](uint64_t v) {
+            if (MODE == 0) o[r++] = __longlong_as_double((long long)v);
+            else acc ^= v + r++;
+        };
+        emit(val);
+        for (;;) {
+            if (nb < 64) topup();
+            if (budget <= 0) { atomicOr(err, 2u); break; }
+            uint32_t top13 = uint32_t(hi >> 51);
+            if (!(top13 & 0x1000)) {
+                consume(1);
+            } else {
+                if (top13 & 0x0800) {
+                    uint32_t lead = (top13 >> 6) & 0x1f;
+                    meaningful = top13 & 0x3f;
+                    if (meaningful > 0) trailing = 64 - lead - meaningful;
+                    else { trailing = 0; meaningful = 64; }
+                    consume(13);
+                } else consume(2);
+                while (nb < int(meaningful)) topup();
+                uint64_t sb = (meaningful == 64) ? hi : (hi >> (64 - meaningful));
+                consume(meaningful);
+                if (budget < 0) { atomicOr(err, 2u); break; }
+                val ^= sb << trailing;
+                if (val == SENT) break;
+            }
+            emit(val);
+        }
+        if (MODE != 0 && acc == 0xdeadbeef) o[pg.row % 64] = 1.0; /* keep acc */
+    }
+}
+
+/* store-only: same store pattern, no decode */
+__global__ void k_store(const PD *__restrict__ pages, int npages, int rows,
+                        double *__restrict__ out) {
+    for (int p0 = blockIdx.x * blockDim.x + threadIdx.x; p0 < npages;
+         p0 += gridDim.x * blockDim.x) {
+        double *o = out + pages[p0].row;
+        for (int r = 0; r < rows; r++) o[r] = double(r);
+    }
+}
+
+/* load-only: stream the compressed bytes, no decode */
+__global__ void k_load(const uint8_t *__restrict__ blob,
+                       const PD *__restrict__ pages, int npages,
+                       double *__restrict__ out) {
+    for (int p0 = blockIdx.x * blockDim.x + threadIdx.x; p0 < npages;
+         p0 += gridDim.x * blockDim.x) {
+        PD pg = pages[p0];
+        const uint8_t *p = blob + pg.off;
+        uint64_t acc = 0;
+        for (uint32_t i = 0; i + 8 <= pg.len; i += 8) acc ^= dbe64(p + i);
+        if (acc == 0xdeadbeef) out[pg.row % 64] = 1.0;
+    }
+}
+
+#define CHK(x) do { auto e=(x); if (e!=hipSuccess){printf("ERR %s %s\n",#x,hipGetErrorString(e)); return 1;} } while(0)
+
+int main(int argc, char **argv) {
+    int npages = argc > 1 ? atoi(argv[1]) : 160000;
+    int rows = argc > 2 ? atoi(argv[2]) : 15625;
+    int block = argc > 3 ? atoi(argv[3]) : 256;
+    printf("npages=%d rows=%d block=%d\n", npages, rows, block);
+
+    /* host: encode a handful of unique pages, tile them */
+    std::mt19937_64 rng(231);
+    std::normal_distribution<double> nd(0, 0.5);
+    int uniq = 64;
+    std::vector<std::vector<uint8_t>> enc(uniq);
+    std::vector<double> v(rows);
+    for (int u = 0; u < uniq; u++) {
+        double w = 50;
+        for (int i = 0; i < rows; i++) {
+            w += nd(rng);
+            if (w < 0) w = 0; if (w > 100) w = 100;
+            v[i] = std::round(w * 10) / 10;
+        }
+        enc[u].resize(rows * 12 + 64);
+        int64_t n = gs_encode_f64(v.data(), rows, enc[u].data(), enc[u].size());
+        enc[u].resize(n);
+    }
+    size_t per = 0;
+    for (auto &e : enc) per = std::max(per, e.size());
+    per = (per + 63) & ~size_t(63);
+    size_t blobsz = size_t(npages) * per + 64;
+    uint8_t *d_blob;
+    double *d_out;
+    PD *d_pd;
+    unsigned *d_err;
+    CHK(hipMalloc(&d_blob, blobsz));
+    CHK(hipMalloc(&d_out, size_t(npages) * rows * 8));
+    CHK(hipMalloc(&d_pd, npages * sizeof(PD)));
+    CHK(hipMalloc(&d_err, 4));
+    CHK(hipMemset(d_err, 0, 4));
+    std::vector<PD> pd(npages);
+    std::vector<uint8_t> hb(blobsz);
+    double avg_len = 0;
+    for (int i = 0; i < npages; i++) {
+        auto &e = enc[i % uniq];
+        memcpy(hb.data() + size_t(i) * per, e.data(), e.size());
+        pd[i] = {size_t(i) * per, uint32_t(e.size()), uint64_t(i) * rows};
+        avg_len += e.size();
+    }
+    printf("bits/val %.2f\n", avg_len / npages * 8 / rows);
+    CHK(hipMemcpy(d_blob, hb.data(), blobsz, hipMemcpyHostToDevice));
+    CHK(hipMemcpy(d_pd, pd.data(), npages * sizeof(PD), hipMemcpyDeviceToHost == 99 ? hipMemcpyHostToDevice : hipMemcpyHostToDevice));
+
+    int grid = std::min((npages + block - 1) / block, 16384);
+    hipEvent_t a, b;
+    hipEventCreate(&a);
+    hipEventCreate(&b);
+    double vals = double(npages) * rows;
+    auto run = [&](const char *name, auto fn) {
+        fn(); /* warmup */
+        CHK(hipDeviceSynchronize());
+        hipEventRecord(a);
+        for (int it = 0; it < 3; it++) fn();
+        hipEventRecord(b);
+        CHK(hipDeviceSynchronize());
+        float ms;
+        hipEventElapsedTime(&ms, a, b);
+        ms /= 3;
+        printf("%-12s %8.2f ms  %7.1f Gval/s  %7.1f GB/s(out8B)\n", name, ms,
+               vals / ms / 1e6, vals * 8 / ms / 1e6);
+        return 0;
+    };
+    run("full", [&] { k_gor<0><<<grid, block>>>(d_blob, d_pd, npages, d_out, d_err); });
+    run("nostore", [&] { k_gor<1><<<grid, block>>>(d_blob, d_pd, npages, d_out, d_err); });
+    run("storeonly", [&] { k_store<<<grid, block>>>(d_pd, npages, rows, d_out); });
+    run("loadonly", [&] { k_load<<<grid, block>>>(d_blob, d_pd, npages, d_out); });
+    unsigned derr;
+    CHK(hipMemcpy(&derr, d_err, 4, hipMemcpyDeviceToHost));
+    printf("err=%u (expect 0)\n", derr);
+    /* verify full output of page 0 */
+    std::vector<double> o(rows);
+    CHK(hipMemcpy(o.data(), d_out, rows * 8, hipMemcpyDeviceToHost));
+    printf("out[0..3]=%.1f %.1f %.1f %.1f\n", o[0], o[1], o[2], o[3]);
+    return 0;
+}
