@@ -30,7 +30,8 @@ enum PageClass {
     PC_SEQ = 0,     /* universal sequential thread-per-page decoder */
     PC_RLE_TS = 1,  /* DeltaTs + RLE + all-valid: closed-form parallel */
     PC_RLE_I64 = 2, /* Delta + RLE + all-valid: closed-form parallel */
-    PC_NCLASS = 3,
+    PC_GOR = 3,     /* Gorilla + all-valid: LDS-staged cooperative stores */
+    PC_NCLASS = 4,
 };
 
 extern "C" {

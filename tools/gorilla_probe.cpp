@@ -90,6 +90,114 @@ __global__ void k_gor(const uint8_t *__restrict__ blob,
     }
 }
 
+/* MODE 0 decode with LDS ring staging: each lane accumulates RING decoded
+ * values in LDS; when all lanes' rings fill (lanes run in lockstep, one
+ * value per iteration), the wave flushes lane-by-lane with page-contiguous
+ * coalesced stores. */
+template <int RING>
+__global__ void k_gor_lds(const uint8_t *__restrict__ blob,
+                          const PD *__restrict__ pages, int npages,
+                          double *__restrict__ out,
+                          unsigned *__restrict__ err) {
+    __shared__ double ring[256 / 64][RING][64 + 1]; /* [wave][slot][lane] */
+    const int lane = threadIdx.x & 63;
+    const int wv = threadIdx.x >> 6;
+    auto rslot = ring[wv];
+    int stride = gridDim.x * blockDim.x;
+    int base_id = blockIdx.x * blockDim.x + threadIdx.x;
+    int rounds = (npages + stride - 1) / stride;
+    for (int rd = 0; rd < rounds; rd++) {
+        int p0 = base_id + rd * stride;
+        bool have = p0 < npages;
+        PD pg = pages[have ? p0 : 0];
+        const uint8_t *data = blob + pg.off;
+        double *o = out + pg.row;
+        const uint8_t *s = data + 1;
+        uint32_t slen = pg.len - 1;
+        uint64_t val = dbe64(s + 1);
+        const uint8_t *p = s + 9;
+        int64_t budget = int64_t(slen - 9) * 8;
+        uint64_t hi = 0, lo = 0;
+        int nb = 0;
+        uint32_t trailing = 0, meaningful = 64;
+        int r = 0;     /* rows emitted (incl. staged) */
+        int rfill = 0; /* staged in ring */
+        bool done = !have;
+        uint64_t nextw = dbe64(p);
+        p += 8;
+        auto topup = [&]() {
+            uint64_t x = nextw;
+            nextw = dbe64(p);
+            p += 8;
+            if (nb == 0) { hi = x; lo = 0; }
+            else { hi |= x >> nb; lo = x << (64 - nb); }
+            nb += 64;
+        };
+        auto consume = [&](unsigned k) {
+            hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
+            lo = (k == 64) ? 0 : (lo << k);
+            nb -= int(k);
+            budget -= int64_t(k);
+        };
+        auto flush = [&]() {
+            /* cooperative: store each source lane's staged run as a
+               page-contiguous coalesced store (all 64 lanes alive here) */
+            for (int sl = 0; sl < 64; sl++) {
+                unsigned long long ob =
+                    __shfl((unsigned long long)(uintptr_t)o, sl, 64);
+                int cnt = __shfl(rfill, sl, 64);
+                int row0 = __shfl(r, sl, 64) - cnt;
+                if (lane < cnt)
+                    ((double *)(uintptr_t)ob)[row0 + lane] = rslot[lane][sl];
+            }
+            rfill = 0;
+        };
+        /* stage the first value */
+        if (!done) {
+            rslot[rfill][lane] = __longlong_as_double((long long)val);
+            rfill++;
+            r++;
+        }
+        while (!__all(done)) {
+            if (!done) {
+                if (nb < 64) topup();
+                if (budget <= 0) { atomicOr(err, 2u); done = true; }
+            }
+            if (!done) {
+                uint32_t top13 = uint32_t(hi >> 51);
+                bool stage = true;
+                if (!(top13 & 0x1000)) {
+                    consume(1);
+                } else {
+                    if (top13 & 0x0800) {
+                        uint32_t lead = (top13 >> 6) & 0x1f;
+                        meaningful = top13 & 0x3f;
+                        if (meaningful > 0) trailing = 64 - lead - meaningful;
+                        else { trailing = 0; meaningful = 64; }
+                        consume(13);
+                    } else consume(2);
+                    while (nb < int(meaningful)) topup();
+                    uint64_t sb =
+                        (meaningful == 64) ? hi : (hi >> (64 - meaningful));
+                    consume(meaningful);
+                    if (budget < 0) { atomicOr(err, 2u); done = true; stage = false; }
+                    else {
+                        val ^= sb << trailing;
+                        if (val == SENT) { done = true; stage = false; }
+                    }
+                }
+                if (stage) {
+                    rslot[rfill][lane] = __longlong_as_double((long long)val);
+                    rfill++;
+                    r++;
+                }
+            }
+            if (__any(rfill == RING)) flush();
+        }
+        flush();
+    }
+}
+
 /* store-only: same store pattern, no decode */
 __global__ void k_store(const PD *__restrict__ pages, int npages, int rows,
                         double *__restrict__ out) {
@@ -188,6 +296,19 @@ int main(int argc, char **argv) {
     run("nostore", [&] { k_gor<1><<<grid, block>>>(d_blob, d_pd, npages, d_out, d_err); });
     run("storeonly", [&] { k_store<<<grid, block>>>(d_pd, npages, rows, d_out); });
     run("loadonly", [&] { k_load<<<grid, block>>>(d_blob, d_pd, npages, d_out); });
+    run("lds16", [&] { k_gor_lds<16><<<grid, block>>>(d_blob, d_pd, npages, d_out, d_err); });
+    run("lds32", [&] { k_gor_lds<32><<<grid, block>>>(d_blob, d_pd, npages, d_out, d_err); });
+    /* verify lds16 output matches full */
+    {
+        std::vector<double> a(rows), c(rows);
+        hipMemcpy(a.data(), d_out, rows * 8, hipMemcpyDeviceToHost);
+        hipLaunchKernelGGL((k_gor<0>), dim3(grid), dim3(block), 0, 0, d_blob, d_pd, npages, d_out, d_err);
+        hipDeviceSynchronize();
+        hipMemcpy(c.data(), d_out, rows * 8, hipMemcpyDeviceToHost);
+        int bad = 0;
+        for (int i = 0; i < rows; i++) if (a[i] != c[i]) bad++;
+        printf("lds-vs-full mismatches(page0): %d\n", bad);
+    }
     unsigned derr;
     CHK(hipMemcpy(&derr, d_err, 4, hipMemcpyDeviceToHost));
     printf("err=%u (expect 0)\n", derr);
