@@ -263,6 +263,49 @@ class Engine:
             self.lib.gs_ctx_destroy(self._ctx)
             self._ctx = None
 
+    def upload_packed(self, buf, page_off, page_len, num_values, ctypes_arr,
+                      series_ids, pages_per_group, validate_crc=True):
+        """Vectorized upload: all pages live in one contiguous host buffer.
+        page_off/page_len/num_values/ctypes_arr are per-page numpy arrays in
+        group-major order (group g's pages at [g*ppg, (g+1)*ppg), pages[0]
+        of each group = time page); series_ids is per-group."""
+        buf = np.frombuffer(buf, dtype=np.uint8) if isinstance(buf, (bytes, bytearray)) else buf
+        npages = page_off.size
+        ngroups = npages // pages_per_group
+        spec_dt = np.dtype({
+            "names": ["ptr", "len", "nv", "ct", "cid"],
+            "formats": ["<u8", "<u8", "<u4", "u1", "<u4"],
+            "offsets": [0, 8, 16, 20, 24],
+            "itemsize": ctypes.sizeof(GsPageSpec),
+        })
+        specs = np.zeros(npages, dtype=spec_dt)
+        specs["ptr"] = buf.ctypes.data + page_off.astype(np.uint64)
+        specs["len"] = page_len.astype(np.uint64)
+        specs["nv"] = num_values.astype(np.uint32)
+        specs["ct"] = ctypes_arr.astype(np.uint8)
+        specs["cid"] = np.tile(np.arange(pages_per_group, dtype=np.uint32), ngroups)
+        gdesc_dt = np.dtype({
+            "names": ["pages", "npages", "sid"],
+            "formats": ["<u8", "<u4", "<u4"],
+            "offsets": [0, 8, 12],
+            "itemsize": ctypes.sizeof(GsColumnGroupDesc),
+        })
+        gdescs = np.zeros(ngroups, dtype=gdesc_dt)
+        gdescs["pages"] = specs.ctypes.data + \
+            (np.arange(ngroups, dtype=np.uint64) * pages_per_group *
+             ctypes.sizeof(GsPageSpec))
+        gdescs["npages"] = pages_per_group
+        gdescs["sid"] = series_ids.astype(np.uint32)
+        h = self.lib.gs_groups_upload(
+            self._ctx,
+            ctypes.cast(gdescs.ctypes.data, ctypes.POINTER(GsColumnGroupDesc)),
+            ngroups, 1 if validate_crc else 0)
+        if not h:
+            raise RuntimeError(f"gs_groups_upload failed: {self._pl.err()}")
+        h = ctypes.c_void_p(h)
+        rows = self.lib.gs_set_rows(h)
+        return GroupSet(self, h, rows, ngroups, (buf, specs, gdescs))
+
     def upload(self, groups, validate_crc=True):
         """groups: list of (series_id, [(page_bytes, ctype), ...]);
         pages[0] must be the time page."""
