@@ -6,8 +6,8 @@ TSBS devops scan") on BASELINE.json configs[1] — the north-star workload:
 
   10,000 series x 1,000,000 points per GPU; timestamps at 1 s spacing
   (DeltaTs/RLE pages); f64 gauge values = clipped random walk in [0,100],
-  90% of series quantized to 1/8 (measurement precision, ~1 B/val Gorilla)
-  and 10% full-precision (~8 B/val) — blended ~1.7 B/val, the 12-20
+  80% of series quantized to 1/4 (measurement precision) and 20%
+  full-precision (~8 B/val) — blended ~1.5 B/val, the 12-20
   bits/val TSBS-devops regime SURVEY.md §8d assumes; pages of 4,000 rows;
   time-range filter ts in [25%,75%] of the range (50% selectivity); fused
   per-5-minute max/sum/count buckets.
@@ -68,13 +68,13 @@ def build_workload(nseries, npts, page_rows, unique, sub_batches, seed=231):
         vals = np.empty((cu, npts))
         for k in range(cu):
             walk = np.clip(np.cumsum(rng.normal(0, 0.5, npts)) + 50, 0, 100)
-            if (u + k) % 10 == 0:
+            if (u + k) % 5 == 0:
                 vals[k] = walk  # full-precision gauge (hard pages, ~8 B/val)
             else:
-                # measurement-precision gauge: 1/8 quantization (~1 B/val);
-                # blended ~1.7 B/val, the 12-20 bits/val TSBS-devops regime
+                # measurement-precision gauge: 1/4 quantization; blended
+                # ~11-14 bits/val, the 12-20 bits/val TSBS-devops regime
                 # SURVEY.md §8d assumes
-                vals[k] = np.round(walk * 8) / 8
+                vals[k] = np.round(walk * 4) / 4
         flat = np.ascontiguousarray(vals.reshape(-1))
         total_pages = cu * npages
         enc = np.zeros(total_pages * cap, dtype=np.uint8)
@@ -138,7 +138,7 @@ def cpu_baseline_leg(nseries_sample, npts, page_rows, lo, hi, seed=231):
         ts_datas.append(gs.encode_ts(ts))
     for s in range(nseries_sample):
         walk = np.clip(np.cumsum(rng.normal(0, 0.5, npts)) + 50, 0, 100)
-        v = walk if s % 10 == 0 else np.round(walk * 8) / 8
+        v = walk if s % 5 == 0 else np.round(walk * 4) / 4
         for p in range(npages):
             val_datas.append(gs.encode_f64(v[p * page_rows:(p + 1) * page_rows]))
 
@@ -202,7 +202,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--series", type=int, default=10000)
     ap.add_argument("--npts", type=int, default=1_000_000)
-    ap.add_argument("--page-rows", type=int, default=15_625)
+    ap.add_argument("--page-rows", type=int, default=4000)
     ap.add_argument("--sub-batches", type=int, default=4)
     ap.add_argument("--unique", type=int, default=256)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
