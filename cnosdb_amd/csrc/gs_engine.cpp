@@ -843,6 +843,142 @@ __global__ void k_agg_merge(int ngroups, int nbuckets,
     }
 }
 
+/* ------------------------------------------------- compaction merge (k-way)
+ * Config #5: k overlapping L0 column groups per series -> one merged,
+ * deduped stream (compact.rs:271-404, comapcting_block_meta_group.rs:
+ * 52-208).  Streams are ordered oldest -> newest (file_id order,
+ * iterator.rs:488); equal timestamps collapse, per column the newest
+ * non-null value wins (batch_builder.rs:106-155).
+ *
+ * GPU mapping: no loser tree — ranks by binary search.  An element is the
+ * row OWNER iff no newer stream contains its ts; the merged position of
+ * an owner is the sum over streams of "owners with smaller ts", obtained
+ * from per-stream exclusive scans of the owner flags.  Dedup value
+ * selection walks streams newest -> oldest at scatter time. */
+
+#define GS_MAX_STREAMS 16
+
+struct CompactArgs {
+    const int64_t *ts[GS_MAX_STREAMS];
+    const double *val[GS_MAX_STREAMS];
+    const uint8_t *valid[GS_MAX_STREAMS];
+    const DevGroup *groups[GS_MAX_STREAMS];
+    uint8_t *flags[GS_MAX_STREAMS];
+    int32_t *prefix[GS_MAX_STREAMS];
+    int64_t *counts; /* [nsets][nseries] owner counts */
+    int64_t *out_off; /* [nseries] output row offsets */
+};
+
+__device__ __forceinline__ bool dev_ts_contains(const int64_t *t, int64_t n,
+                                                int64_t x, int64_t *pos) {
+    int64_t lo = 0, hi = n;
+    while (lo < hi) { int64_t m = (lo + hi) >> 1; if (t[m] < x) lo = m + 1; else hi = m; }
+    *pos = lo;
+    return lo < n && t[lo] == x;
+}
+
+/* block per (set, series): owner flags + per-series owner count */
+__global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
+                           unsigned *__restrict__ err) {
+    int f = blockIdx.y;
+    for (int s = blockIdx.x; s < nseries; s += gridDim.x) {
+        DevGroup g = a.groups[f][s];
+        const int64_t *t = a.ts[f] + g.row_off;
+        uint8_t *fl = a.flags[f] + g.row_off;
+        long long cnt = 0;
+        for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
+            int64_t x = t[j];
+            if (j > 0 && t[j - 1] >= x) atomicOr(err, DERR_FORMAT); /* "data in stream is not sorted" */
+            bool owner = true;
+            for (int f2 = f + 1; f2 < nsets; f2++) {
+                DevGroup g2 = a.groups[f2][s];
+                int64_t pos;
+                if (dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &pos)) {
+                    owner = false;
+                    break;
+                }
+            }
+            fl[j] = owner;
+            cnt += owner;
+        }
+        /* block reduce cnt -> counts[f][s] */
+        __shared__ long long sred[256];
+        sred[threadIdx.x] = cnt;
+        __syncthreads();
+        for (int w = blockDim.x >> 1; w > 0; w >>= 1) {
+            if (threadIdx.x < w) sred[threadIdx.x] += sred[threadIdx.x + w];
+            __syncthreads();
+        }
+        if (threadIdx.x == 0)
+            a.counts[size_t(f) * nseries + s] = sred[0];
+    }
+}
+
+/* thread per (set, series): exclusive scan of owner flags */
+__global__ void k_cm_prefix(CompactArgs a, int nsets, int nseries) {
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    int total = nsets * nseries;
+    for (; idx < total; idx += gridDim.x * blockDim.x) {
+        int f = idx / nseries, s = idx % nseries;
+        DevGroup g = a.groups[f][s];
+        const uint8_t *fl = a.flags[f] + g.row_off;
+        int32_t *pf = a.prefix[f] + g.row_off;
+        int32_t acc = 0;
+        for (int64_t j = 0; j < g.nrows; j++) { pf[j] = acc; acc += fl[j]; }
+    }
+}
+
+/* block per (set, series): scatter owners to merged positions with
+ * newest-non-null value selection */
+__global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
+                             int64_t *__restrict__ out_ts,
+                             double *__restrict__ out_val,
+                             uint8_t *__restrict__ out_valid) {
+    int f = blockIdx.y;
+    for (int s = blockIdx.x; s < nseries; s += gridDim.x) {
+        DevGroup g = a.groups[f][s];
+        const int64_t *t = a.ts[f] + g.row_off;
+        const uint8_t *fl = a.flags[f] + g.row_off;
+        const int32_t *pf = a.prefix[f] + g.row_off;
+        int64_t base = a.out_off[s];
+        for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
+            if (!fl[j]) continue;
+            int64_t x = t[j];
+            /* merged position: owners with smaller ts across all streams */
+            int64_t pos = base + pf[j];
+            for (int f2 = 0; f2 < nsets; f2++) {
+                if (f2 == f) continue;
+                DevGroup g2 = a.groups[f2][s];
+                int64_t ins;
+                dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &ins);
+                /* owners in f2 below insertion point */
+                pos += (ins > 0) ? a.prefix[f2][g2.row_off + ins - 1] +
+                                       a.flags[f2][g2.row_off + ins - 1]
+                                 : 0;
+            }
+            /* value: newest stream containing x with a non-null value
+               (batch_builder.rs:139-151); all-null -> null, slot 0 */
+            double v = 0.0;
+            uint8_t ok = 0;
+            for (int f2 = nsets - 1; f2 >= 0; f2--) {
+                DevGroup g2 = a.groups[f2][s];
+                int64_t ins;
+                if (!dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &ins))
+                    continue;
+                const uint8_t *vd2 = a.valid[f2];
+                if (!vd2 || vd2[g2.row_off + ins]) {
+                    v = a.val[f2][g2.row_off + ins];
+                    ok = 1;
+                    break;
+                }
+            }
+            out_ts[pos] = x;
+            out_val[pos] = v;
+            if (out_valid) out_valid[pos] = ok;
+        }
+    }
+}
+
 /* ------------------------------------------------------------- host state */
 
 struct GsCtx {
@@ -1185,6 +1321,7 @@ void gs_groups_free(GsGroupSet *set) {
 }
 
 int64_t gs_set_rows(const GsGroupSet *set) { return set ? set->total_rows : -1; }
+int64_t gs_set_series(const GsGroupSet *set) { return set ? set->nsgroups : -1; }
 int64_t gs_set_groups(const GsGroupSet *set) { return set ? int64_t(set->ngroups) : -1; }
 GsStatus gs_set_row_offsets(const GsGroupSet *set, int64_t *out) {
     if (!set || !out) return fail(GS_ERR, "bad args");
@@ -1400,6 +1537,85 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
     }
     result->out_rows = out_rows;
     result->decoded_rows = set->total_rows;
+    return GS_OK;
+}
+
+GsStatus gs_compact_merge(GsCtx *ctx, GsGroupSet *const *sets, int32_t nsets,
+                          const int64_t *const *d_ts,
+                          const double *const *d_val,
+                          const uint8_t *const *d_valid, int64_t *d_out_ts,
+                          double *d_out_val, uint8_t *d_out_valid,
+                          int64_t *h_out_offsets, int64_t *out_rows) {
+    if (!ctx || !sets || nsets < 1 || nsets > GS_MAX_STREAMS || !d_ts ||
+        !d_val || !d_out_ts || !d_out_val)
+        return fail(GS_ERR, "bad args to gs_compact_merge");
+    int nseries = sets[0]->nsgroups;
+    for (int f = 0; f < nsets; f++)
+        if (sets[f]->nsgroups != nseries)
+            return fail(GS_ERR, "all streams must cover the same series list");
+    HIP_TRY(hipSetDevice(ctx->device));
+
+    CompactArgs a;
+    memset(&a, 0, sizeof(a));
+    std::vector<void *> scratch;
+    auto cleanup = [&]() { for (void *p : scratch) hipFree(p); };
+    for (int f = 0; f < nsets; f++) {
+        a.ts[f] = d_ts[f];
+        a.val[f] = d_val[f];
+        a.valid[f] = d_valid ? d_valid[f] : nullptr;
+        a.groups[f] = sets[f]->d_sgroups;
+        size_t rows = size_t(sets[f]->total_rows);
+        uint8_t *fl;
+        int32_t *pf;
+        if (hipMalloc(&fl, rows ? rows : 1) != hipSuccess ||
+            hipMalloc(&pf, (rows ? rows : 1) * 4) != hipSuccess) {
+            cleanup();
+            return fail(GS_ERR, "hipMalloc compact scratch failed");
+        }
+        scratch.push_back(fl);
+        scratch.push_back(pf);
+        a.flags[f] = fl;
+        a.prefix[f] = pf;
+    }
+    int64_t *d_counts, *d_ooff;
+    if (hipMalloc(&d_counts, size_t(nsets) * nseries * 8) != hipSuccess ||
+        hipMalloc(&d_ooff, size_t(nseries) * 8) != hipSuccess) {
+        cleanup();
+        return fail(GS_ERR, "hipMalloc compact tables failed");
+    }
+    scratch.push_back(d_counts);
+    scratch.push_back(d_ooff);
+    a.counts = d_counts;
+    a.out_off = d_ooff;
+
+    int gx = nseries > 2048 ? 2048 : nseries;
+    hipLaunchKernelGGL(k_cm_flags, dim3(gx, nsets), dim3(256), 0, ctx->stream,
+                       a, nsets, nseries, ctx->d_err);
+    int total = nsets * nseries;
+    hipLaunchKernelGGL(k_cm_prefix, dim3(grid_for(total, 256)), dim3(256), 0,
+                       ctx->stream, a, nsets, nseries);
+    std::vector<int64_t> counts(size_t(nsets) * nseries), ooff(nseries + 1);
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    HIP_TRY(hipMemcpy(counts.data(), d_counts, counts.size() * 8,
+                      hipMemcpyDeviceToHost));
+    int64_t acc = 0;
+    for (int s = 0; s < nseries; s++) {
+        ooff[s] = acc;
+        for (int f = 0; f < nsets; f++) acc += counts[size_t(f) * nseries + s];
+    }
+    ooff[nseries] = acc;
+    HIP_TRY(hipMemcpyAsync(d_ooff, ooff.data(), size_t(nseries) * 8,
+                           hipMemcpyHostToDevice, ctx->stream));
+    hipLaunchKernelGGL(k_cm_scatter, dim3(gx, nsets), dim3(256), 0,
+                       ctx->stream, a, nsets, nseries, d_out_ts, d_out_val,
+                       d_out_valid);
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    cleanup();
+    GsStatus st = check_dev_err(ctx);
+    if (st != GS_OK) return st;
+    if (h_out_offsets)
+        memcpy(h_out_offsets, ooff.data(), size_t(nseries + 1) * 8);
+    if (out_rows) *out_rows = acc;
     return GS_OK;
 }
 

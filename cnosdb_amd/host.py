@@ -111,6 +111,8 @@ class PageLib:
         lib.gs_set_rows.argtypes = [ctypes.c_void_p]
         lib.gs_set_groups.restype = ctypes.c_int64
         lib.gs_set_groups.argtypes = [ctypes.c_void_p]
+        lib.gs_set_series.restype = ctypes.c_int64
+        lib.gs_set_series.argtypes = [ctypes.c_void_p]
         lib.gs_set_row_offsets.restype = ctypes.c_int32
         lib.gs_set_row_offsets.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
         lib.gs_decode.restype = ctypes.c_int32
@@ -125,6 +127,12 @@ class PageLib:
         lib.gs_scan.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                 ctypes.POINTER(GsScanSpec),
                                 ctypes.POINTER(GsScanResult)]
+        lib.gs_compact_merge.restype = ctypes.c_int32
+        lib.gs_compact_merge.argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+            ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_void_p),
+            ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(ctypes.c_int64)]
 
     def err(self):
         return self.lib.gs_last_error().decode()
@@ -351,6 +359,30 @@ class Engine:
                                          arr, len(ranges))
         if st != 0:
             raise RuntimeError(f"gs_apply_tombstone failed: {self._pl.err()}")
+
+    def compact_merge(self, gsets, d_ts_list, d_val_list, d_valid_list,
+                      d_out_ts, d_out_val, d_out_valid=None):
+        """k-way merge+dedup (BASELINE config #5). gsets oldest->newest;
+        d_*_list: per-stream decoded torch tensors (valid entries may be
+        None). Returns (out_rows, per-series offsets np.array)."""
+        k = len(gsets)
+        sets = (ctypes.c_void_p * k)(*[g._h for g in gsets])
+        ts_p = (ctypes.c_void_p * k)(*[t.data_ptr() for t in d_ts_list])
+        val_p = (ctypes.c_void_p * k)(*[t.data_ptr() for t in d_val_list])
+        vd_p = (ctypes.c_void_p * k)(
+            *[(t.data_ptr() if t is not None else None) for t in d_valid_list])
+        nseries = self.lib.gs_set_series(gsets[0]._h)
+        offs = np.zeros(nseries + 1, dtype=np.int64)
+        rows = ctypes.c_int64(0)
+        st = self.lib.gs_compact_merge(
+            self._ctx, sets, k, ts_p, val_p, vd_p,
+            ctypes.c_void_p(d_out_ts.data_ptr()),
+            ctypes.c_void_p(d_out_val.data_ptr()),
+            ctypes.c_void_p(d_out_valid.data_ptr()) if d_out_valid is not None else None,
+            _np_ptr(offs), ctypes.byref(rows))
+        if st != 0:
+            raise RuntimeError(f"gs_compact_merge failed ({st}): {self._pl.err()}")
+        return rows.value, offs
 
     def scan(self, gset, d_ts, d_val, time_range=None, tombstones=None,
              d_out_ts=None, d_out_val=None, agg=None):

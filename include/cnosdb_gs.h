@@ -116,6 +116,8 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
 void gs_groups_free(GsGroupSet *set);
 int64_t gs_set_rows(const GsGroupSet *set);
 int64_t gs_set_groups(const GsGroupSet *set);
+/* series-level group count (consecutive same-series groups merged) */
+int64_t gs_set_series(const GsGroupSet *set);
 /* copy the per-group row offsets (ngroups entries) into caller buffer */
 GsStatus gs_set_row_offsets(const GsGroupSet *set, int64_t *out);
 
@@ -187,6 +189,22 @@ typedef struct {
 
 GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                  GsScanResult *result);
+
+/* ---- compaction merge (BASELINE config #5) ----
+ * k overlapping L0 column-group streams per series -> one merged, deduped
+ * (ts, value, validity) stream per series (tskv/src/compaction/compact.rs:
+ * 271-404, comapcting_block_meta_group.rs:52-208; dedup rule
+ * reader/batch_builder.rs:106-155).  sets[f] (f = 0 oldest .. nsets-1
+ * newest, file_id order) must cover the same series list; d_ts/d_val/
+ * d_valid[f] are that stream's decoded columns (from gs_decode).  Outputs
+ * are caller device buffers sized >= the sum of input rows;
+ * h_out_offsets[nseries+1] receives per-series output row offsets. */
+GsStatus gs_compact_merge(GsCtx *ctx, GsGroupSet *const *sets, int32_t nsets,
+                          const int64_t *const *d_ts,
+                          const double *const *d_val,
+                          const uint8_t *const *d_valid, int64_t *d_out_ts,
+                          double *d_out_val, uint8_t *d_out_valid,
+                          int64_t *h_out_offsets, int64_t *out_rows);
 
 #ifdef __cplusplus
 }

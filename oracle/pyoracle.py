@@ -156,6 +156,31 @@ def time_span(ts, mn, mx):
     return s, e - s
 
 
+def merge_dedup(streams):
+    """sort_merge + BatchMergeBuilder dedup (reader/sort_merge.rs:152-343,
+    reader/batch_builder.rs:106-155): k time-sorted streams ordered
+    oldest -> newest (grouped chunks sorted by file_id, iterator.rs:488;
+    loser-tree ties break toward the lower stream index,
+    sort_merge.rs:305-312).  Equal-ts rows collapse to one; per column the
+    value comes from the NEWEST stream containing that ts with a non-null
+    value, else the row is null (value slot 0).  streams: list of
+    (ts, val, valid-or-None)."""
+    all_ts = np.concatenate([np.asarray(t) for t, _, _ in streams])
+    uts = np.unique(all_ts)
+    out_val = np.zeros(uts.size, dtype=np.float64)
+    out_valid = np.zeros(uts.size, dtype=bool)
+    for t, v, vd in streams:  # oldest -> newest: later valid rows overwrite
+        idx = np.searchsorted(uts, np.asarray(t))
+        if vd is None:
+            out_val[idx] = v
+            out_valid[idx] = True
+        else:
+            m = np.asarray(vd, dtype=bool)
+            out_val[idx[m]] = np.asarray(v)[m]
+            out_valid[idx[m]] = True
+    return uts, out_val, out_valid
+
+
 def bucket_agg(ts, vals, valid, t0, bucket_ns, n_buckets):
     """stock-DataFusion-style per-bucket max/sum/count over non-null rows
     (the downsampling aggregate run above TskvExec)."""
