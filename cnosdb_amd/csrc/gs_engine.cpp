@@ -979,6 +979,302 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
     }
 }
 
+/* --------------------------------------------------- GPU page re-encode
+ * The write side of compaction (tsm/writer.rs:249-314 via
+ * Page::arrow_array_to_page, tsm/page.rs:100-353): one thread per output
+ * page, byte-exact with the host encoders in gs_encode.cpp (which are
+ * pinned by the reference's golden vectors).  ts/i64 multi-pass encoders
+ * require all-valid input (time columns are never null; i64 nulls are a
+ * later row); Gorilla streams past nulls inline. */
+
+struct EncPageSpec {
+    int64_t row_off;
+    int32_t nrows;
+    int32_t pad;
+};
+
+/* MSB-first bit appender emitting bytes to global memory */
+struct DevBitWriter {
+    uint8_t *dst;
+    int64_t nbytes;
+    uint64_t acc; /* filled from the top */
+    int nfill;
+    __device__ void init(uint8_t *d) { dst = d; nbytes = 0; acc = 0; nfill = 0; }
+    __device__ __forceinline__ void put(uint64_t x, int l) { /* l <= 32 */
+        acc |= (x & ((l == 64) ? ~0ULL : ((1ULL << l) - 1))) << (64 - nfill - l);
+        nfill += l;
+        while (nfill >= 8) {
+            dst[nbytes++] = uint8_t(acc >> 56);
+            acc <<= 8;
+            nfill -= 8;
+        }
+    }
+    __device__ __forceinline__ void put64(uint64_t x, int l) {
+        if (l > 32) { put(x >> 32, l - 32); put(x & 0xffffffffULL, 32); }
+        else if (l > 0) put(x, l);
+    }
+    __device__ int64_t finish() { /* pad final partial byte with zeros */
+        if (nfill > 0) { dst[nbytes++] = uint8_t(acc >> 56); acc = 0; nfill = 0; }
+        return nbytes;
+    }
+};
+
+__device__ __forceinline__ void dev_put_be64(uint8_t *p, uint64_t v) {
+    for (int i = 7; i >= 0; i--) { p[i] = uint8_t(v); v >>= 8; }
+}
+__device__ __forceinline__ uint64_t dev_zzenc(int64_t v) {
+    return (uint64_t(v) << 1) ^ uint64_t(v >> 63);
+}
+__device__ int dev_varint_put(uint8_t *dst, uint64_t v) {
+    int n = 0;
+    while (v >= 0x80) { dst[n++] = uint8_t(v | 0x80); v >>= 7; }
+    dst[n++] = uint8_t(v);
+    return n;
+}
+
+__constant__ uint8_t DEV_S8B_NUM_BITS[14][2] = {
+    {60, 1}, {30, 2}, {20, 3}, {15, 4}, {12, 5}, {10, 6}, {8, 7},
+    {7, 8},  {6, 10}, {5, 12}, {4, 15}, {3, 20}, {2, 30}, {1, 60},
+};
+
+/* Gorilla f64 encode of one page's non-null values (float.rs:32-243);
+ * returns data length or -1 (sentinel input) */
+__device__ int64_t dev_enc_gorilla(const double *v, const uint8_t *valid,
+                                   int64_t row_off, int32_t nrows,
+                                   uint8_t *dst) {
+    /* find first non-null */
+    int32_t r0 = 0;
+    if (valid) while (r0 < nrows && !valid[row_off + r0]) r0++;
+    if (r0 >= nrows) return 0; /* empty input -> empty buffer */
+    dst[0] = GS_ENC_GORILLA;
+    dst[1] = 1 << 4;
+    uint64_t prev = (uint64_t)__double_as_longlong(v[row_off + r0]);
+    dev_put_be64(dst + 2, prev);
+    DevBitWriter bw;
+    bw.init(dst + 10);
+    uint64_t prev_lead = ~0ULL, prev_trail = 0;
+    int32_t r = r0 + 1;
+    for (;;) {
+        uint64_t cur;
+        if (r < nrows) {
+            if (valid && !valid[row_off + r]) { r++; continue; }
+            cur = (uint64_t)__double_as_longlong(v[row_off + r]);
+            r++;
+            if (cur == GORILLA_SENTINEL) return -1;
+        } else {
+            cur = GORILLA_SENTINEL;
+        }
+        uint64_t x = cur ^ prev;
+        if (x == 0) {
+            bw.put(0, 1);
+            prev = cur;
+            if (cur == GORILLA_SENTINEL) break;
+            continue;
+        }
+        bw.put(1, 1);
+        uint64_t lead = uint64_t(__builtin_clzll(x)) & 0x1f;
+        uint64_t trail = uint64_t(__builtin_ctzll(x));
+        if (prev_lead != ~0ULL && lead >= prev_lead && trail >= prev_trail) {
+            bw.put(0, 1);
+            int l = int(64 - prev_lead - prev_trail);
+            bw.put64(x >> prev_trail, l);
+        } else {
+            prev_lead = lead;
+            prev_trail = trail;
+            bw.put(1, 1);
+            bw.put(lead, 5);
+            uint64_t sig = 64 - lead - trail;
+            bw.put(sig & 0x3f, 6); /* 64 encodes as 0 */
+            bw.put64(x >> trail, int(sig));
+        }
+        prev = cur;
+        if (cur == GORILLA_SENTINEL) break;
+    }
+    return 10 + bw.finish();
+}
+
+/* delta/zigzag int encode of one all-valid page (timestamp.rs:51-122 /
+ * integer.rs:40-96), multi-pass over the input; returns data length */
+__device__ int64_t dev_enc_int(const int64_t *v, int64_t row_off,
+                               int32_t n, uint8_t *dst, bool is_ts) {
+    if (n == 0) return 0;
+    const int64_t *s = v + row_off;
+    auto rawd = [&](int32_t i) { /* wrapping diff, u64 */
+        return uint64_t(s[i]) - uint64_t(s[i - 1]);
+    };
+    size_t w = 0;
+    dst[w++] = is_ts ? GS_ENC_DELTATS : GS_ENC_DELTA;
+    /* pass 1: max + RLE check */
+    uint64_t max = 0, d1 = 0;
+    bool use_rle = true;
+    for (int32_t i = 1; i < n; i++) {
+        uint64_t d = rawd(i);
+        if (!is_ts) d = dev_zzenc(int64_t(d));
+        if (i == 1) d1 = d;
+        else if (d != d1) use_rle = false;
+        if (d > max) max = d;
+    }
+    uint64_t d0 = is_ts ? uint64_t(s[0]) : dev_zzenc(s[0]);
+    if ((is_ts && n > 1 && use_rle) || (!is_ts && n > 2 && use_rle)) {
+        dst[w++] = 0;
+        dev_put_be64(dst + w, d0);
+        w += 8;
+        if (is_ts) {
+            uint64_t div = 1000000000000ULL;
+            while (div > 1 && d1 % div != 0) div /= 10;
+            if (div > 1) {
+                unsigned sc = 0;
+                for (uint64_t x = div; x > 1; x /= 10) sc++;
+                dst[1] |= uint8_t(sc);
+                w += dev_varint_put(dst + w, d1 / div);
+            } else {
+                w += dev_varint_put(dst + w, d1);
+            }
+            w += dev_varint_put(dst + w, uint64_t(n));
+        } else {
+            w += dev_varint_put(dst + w, d1);
+            w += dev_varint_put(dst + w, uint64_t(n) - 1);
+        }
+        dst[1] |= uint8_t(2 << 4);
+        return int64_t(w);
+    }
+    if (max > ((1ULL << 60) - 1)) { /* uncompressed */
+        dst[w++] = 0;
+        dev_put_be64(dst + w, d0);
+        w += 8;
+        for (int32_t i = 1; i < n; i++) {
+            uint64_t d = rawd(i);
+            if (!is_ts) d = dev_zzenc(int64_t(d));
+            dev_put_be64(dst + w, d);
+            w += 8;
+        }
+        return int64_t(w);
+    }
+    /* simple8b; ts applies the power-of-10 divisor (timestamp.rs:97-121) */
+    uint64_t div = 1;
+    unsigned sc = 0;
+    if (is_ts) {
+        div = 1000000000000ULL;
+        for (int32_t i = 1; i < n && div > 1; i++)
+            while (div > 1 && rawd(i) % div != 0) div /= 10;
+        for (uint64_t x = div; x > 1; x /= 10) sc++;
+    }
+    auto packed = [&](int32_t j) { /* j in 0..n-2 */
+        uint64_t d = rawd(j + 1);
+        return is_ts ? d / div : dev_zzenc(int64_t(d));
+    };
+    dst[w++] = uint8_t((1 << 4) | sc);
+    dev_put_be64(dst + w, d0);
+    w += 8;
+    /* simple8b greedy packer (simple8b.rs:26-76) */
+    int32_t m = n - 1, i = 0;
+    while (i < m) {
+        int32_t remain = m - i;
+        if (remain >= 120) {
+            int32_t lim = remain >= 240 ? 240 : 120;
+            int32_t k = 0;
+            while (k < lim && packed(i + k) == 1) k++;
+            if (k == 240) {
+                for (int q = 0; q < 8; q++) dst[w + q] = 0;
+                w += 8; i += 240;
+                continue;
+            }
+            if (k >= 120) {
+                dev_put_be64(dst + w, 1ULL << 60);
+                w += 8; i += 120;
+                continue;
+            }
+        }
+        bool ok = false;
+        for (int idx = 0; idx < 14; idx++) {
+            int32_t int_n = DEV_S8B_NUM_BITS[idx][0];
+            unsigned bit_n = DEV_S8B_NUM_BITS[idx][1];
+            if (int_n > remain) continue;
+            uint64_t max_val = 1ULL << (bit_n & 0x3f);
+            uint64_t word = (uint64_t(idx) + 2) << 60;
+            bool fits = true;
+            for (int32_t q = 0; q < int_n; q++) {
+                uint64_t pv = packed(i + q);
+                if (pv >= max_val) { fits = false; break; }
+                word |= pv << ((unsigned(q) * bit_n) & 0x3f);
+            }
+            if (!fits) continue;
+            dev_put_be64(dst + w, word);
+            w += 8; i += int_n;
+            ok = true;
+            break;
+        }
+        if (!ok) return -2; /* value out of bounds */
+    }
+    return int64_t(w);
+}
+
+/* one thread per output page: bitset + encode + crc + header
+ * (page layout tsm/page.rs:488-497) */
+__global__ void k_encode_pages(int kind /*0=ts 1=i64 2=f64*/,
+                               const void *__restrict__ vals,
+                               const uint8_t *__restrict__ valid,
+                               const EncPageSpec *__restrict__ pages,
+                               int npages, uint8_t *__restrict__ out,
+                               int64_t cap, int64_t *__restrict__ lens,
+                               unsigned *__restrict__ err) {
+    /* crc32 table built cooperatively in LDS (CRC-32/ISO-HDLC) */
+    __shared__ uint32_t crct[256];
+    for (int i = threadIdx.x; i < 256; i += blockDim.x) {
+        uint32_t c = i;
+        for (int k = 0; k < 8; k++) c = (c & 1) ? 0xEDB88320u ^ (c >> 1) : c >> 1;
+        crct[i] = c;
+    }
+    __syncthreads();
+    for (int p = blockIdx.x * blockDim.x + threadIdx.x; p < npages;
+         p += gridDim.x * blockDim.x) {
+        EncPageSpec ps = pages[p];
+        uint8_t *pg = out + int64_t(p) * cap;
+        uint32_t bl = uint32_t(ps.nrows + 7) / 8;
+        /* validity bitset, LSB-first */
+        uint8_t *bs = pg + 16;
+        for (uint32_t b = 0; b < bl; b++) {
+            uint8_t byte = 0;
+            for (int k = 0; k < 8; k++) {
+                int32_t r = int32_t(b) * 8 + k;
+                if (r < ps.nrows && (!valid || valid[ps.row_off + r]))
+                    byte |= uint8_t(1u << k);
+            }
+            bs[b] = byte;
+        }
+        uint8_t *data = pg + 16 + bl;
+        int64_t dl;
+        if (kind == 2) {
+            dl = dev_enc_gorilla((const double *)vals, valid, ps.row_off,
+                                 ps.nrows, data);
+        } else {
+            if (valid) { /* int encoders need all-valid input (see header) */
+                bool av = true;
+                for (int32_t r = 0; r < ps.nrows; r++)
+                    if (!valid[ps.row_off + r]) { av = false; break; }
+                if (!av) { atomicOr(err, DERR_FORMAT); lens[p] = -3; continue; }
+            }
+            dl = dev_enc_int((const int64_t *)vals, ps.row_off, ps.nrows,
+                             data, kind == 0);
+        }
+        if (dl < 0) {
+            atomicOr(err, DERR_FORMAT);
+            lens[p] = dl;
+            continue;
+        }
+        uint32_t crc = 0xFFFFFFFFu;
+        for (int64_t i = 0; i < dl; i++)
+            crc = crct[(crc ^ data[i]) & 0xFF] ^ (crc >> 8);
+        crc ^= 0xFFFFFFFFu;
+        pg[0] = uint8_t(bl >> 24); pg[1] = uint8_t(bl >> 16);
+        pg[2] = uint8_t(bl >> 8); pg[3] = uint8_t(bl);
+        dev_put_be64(pg + 4, uint64_t(ps.nrows));
+        pg[12] = uint8_t(crc >> 24); pg[13] = uint8_t(crc >> 16);
+        pg[14] = uint8_t(crc >> 8); pg[15] = uint8_t(crc);
+        lens[p] = 16 + int64_t(bl) + dl;
+    }
+}
+
 /* ------------------------------------------------------------- host state */
 
 struct GsCtx {
@@ -1538,6 +1834,49 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
     result->out_rows = out_rows;
     result->decoded_rows = set->total_rows;
     return GS_OK;
+}
+
+GsStatus gs_encode_pages_dev(GsCtx *ctx, int32_t kind, const void *d_vals,
+                             const uint8_t *d_valid, const int64_t *h_row_off,
+                             const int32_t *h_rows, int32_t npages,
+                             uint8_t *d_out, int64_t cap_per_page,
+                             int64_t *h_lens) {
+    if (!ctx || !d_vals || !h_row_off || !h_rows || npages <= 0 || !d_out ||
+        !h_lens || kind < 0 || kind > 2)
+        return fail(GS_ERR, "bad args to gs_encode_pages_dev");
+    int32_t maxr = 0;
+    for (int32_t p = 0; p < npages; p++) if (h_rows[p] > maxr) maxr = h_rows[p];
+    int64_t need = 16 + (int64_t(maxr) + 7) / 8 + 2 + 8 +
+                   (int64_t(maxr) + 1) * 10 + 16;
+    if (cap_per_page < need)
+        return fail(GS_ERR_CAP, "cap_per_page below worst-case encoded size");
+    HIP_TRY(hipSetDevice(ctx->device));
+    std::vector<EncPageSpec> specs(npages);
+    for (int32_t p = 0; p < npages; p++) {
+        specs[p].row_off = h_row_off[p];
+        specs[p].nrows = h_rows[p];
+        specs[p].pad = 0;
+    }
+    EncPageSpec *d_specs;
+    int64_t *d_lens;
+    HIP_TRY(hipMalloc(&d_specs, size_t(npages) * sizeof(EncPageSpec)));
+    if (hipMalloc(&d_lens, size_t(npages) * 8) != hipSuccess) {
+        hipFree(d_specs);
+        return fail(GS_ERR, "hipMalloc lens failed");
+    }
+    hipMemcpyAsync(d_specs, specs.data(), size_t(npages) * sizeof(EncPageSpec),
+                   hipMemcpyHostToDevice, ctx->stream);
+    hipLaunchKernelGGL(k_encode_pages, dim3(grid_for(npages, 256)), dim3(256),
+                       0, ctx->stream, kind, d_vals, d_valid, d_specs, npages,
+                       d_out, cap_per_page, d_lens, ctx->d_err);
+    GsStatus st = GS_OK;
+    if (hipStreamSynchronize(ctx->stream) != hipSuccess)
+        st = fail(GS_ERR, "encode kernel failed");
+    hipMemcpy(h_lens, d_lens, size_t(npages) * 8, hipMemcpyDeviceToHost);
+    hipFree(d_specs);
+    hipFree(d_lens);
+    if (st != GS_OK) return st;
+    return check_dev_err(ctx);
 }
 
 GsStatus gs_compact_merge(GsCtx *ctx, GsGroupSet *const *sets, int32_t nsets,

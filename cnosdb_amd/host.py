@@ -127,6 +127,11 @@ class PageLib:
         lib.gs_scan.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                 ctypes.POINTER(GsScanSpec),
                                 ctypes.POINTER(GsScanResult)]
+        lib.gs_encode_pages_dev.restype = ctypes.c_int32
+        lib.gs_encode_pages_dev.argtypes = [
+            ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p,
+            ctypes.c_int64, ctypes.c_void_p]
         lib.gs_compact_merge.restype = ctypes.c_int32
         lib.gs_compact_merge.argtypes = [
             ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
@@ -383,6 +388,23 @@ class Engine:
         if st != 0:
             raise RuntimeError(f"gs_compact_merge failed ({st}): {self._pl.err()}")
         return rows.value, offs
+
+    def encode_pages_dev(self, kind, d_vals, row_off, rows, d_out,
+                         cap_per_page, d_valid=None):
+        """GPU page re-encode. kind: 0=ts 1=i64 2=f64. row_off/rows: host
+        numpy arrays; d_out capacity npages*cap_per_page. Returns lens."""
+        row_off = np.ascontiguousarray(row_off, dtype=np.int64)
+        rows = np.ascontiguousarray(rows, dtype=np.int32)
+        npages = row_off.size
+        lens = np.zeros(npages, dtype=np.int64)
+        st = self.lib.gs_encode_pages_dev(
+            self._ctx, kind, ctypes.c_void_p(d_vals.data_ptr()),
+            ctypes.c_void_p(d_valid.data_ptr()) if d_valid is not None else None,
+            _np_ptr(row_off), _np_ptr(rows), npages,
+            ctypes.c_void_p(d_out.data_ptr()), cap_per_page, _np_ptr(lens))
+        if st != 0:
+            raise RuntimeError(f"gs_encode_pages_dev failed ({st}): {self._pl.err()}")
+        return lens
 
     def scan(self, gset, d_ts, d_val, time_range=None, tombstones=None,
              d_out_ts=None, d_out_val=None, agg=None):

@@ -190,6 +190,20 @@ typedef struct {
 GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                  GsScanResult *result);
 
+/* ---- GPU page re-encode (the write side of compaction/flush:
+ * Page::arrow_array_to_page, tsm/page.rs:100-353 + tsm/writer.rs:249-314).
+ * kind: 0 = ts (DeltaTs), 1 = i64 (Delta), 2 = f64 (Gorilla).  d_vals is a
+ * device column; pages are [h_row_off[p], +h_rows[p]) slices.  Each full
+ * page (header+crc+bitset+data, byte-exact with the host encoder) is
+ * written at d_out + p*cap_per_page; h_lens receives encoded lengths.
+ * Int encoders (kind 0/1) require all-valid slices — the time column is
+ * never null; null-carrying i64 re-encode stays on the host encoder. */
+GsStatus gs_encode_pages_dev(GsCtx *ctx, int32_t kind, const void *d_vals,
+                             const uint8_t *d_valid, const int64_t *h_row_off,
+                             const int32_t *h_rows, int32_t npages,
+                             uint8_t *d_out, int64_t cap_per_page,
+                             int64_t *h_lens);
+
 /* ---- compaction merge (BASELINE config #5) ----
  * k overlapping L0 column-group streams per series -> one merged, deduped
  * (ts, value, validity) stream per series (tskv/src/compaction/compact.rs:
