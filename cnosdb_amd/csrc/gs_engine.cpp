@@ -1846,8 +1846,10 @@ GsStatus gs_encode_pages_dev(GsCtx *ctx, int32_t kind, const void *d_vals,
         return fail(GS_ERR, "bad args to gs_encode_pages_dev");
     int32_t maxr = 0;
     for (int32_t p = 0; p < npages; p++) if (h_rows[p] > maxr) maxr = h_rows[p];
-    int64_t need = 16 + (int64_t(maxr) + 7) / 8 + 2 + 8 +
-                   (int64_t(maxr) + 1) * 10 + 16;
+    /* worst case: ints -> uncompressed 8 B/val; f64 -> ~77 bits/val */
+    int64_t need = 16 + (int64_t(maxr) + 7) / 8 +
+                   (kind == 2 ? (2 + 8 + (int64_t(maxr) + 1) * 10 + 16)
+                              : (2 + 8 * int64_t(maxr) + 32));
     if (cap_per_page < need)
         return fail(GS_ERR_CAP, "cap_per_page below worst-case encoded size");
     HIP_TRY(hipSetDevice(ctx->device));
