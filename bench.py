@@ -195,8 +195,106 @@ def cpu_baseline_leg(nseries_sample, npts, page_rows, lo, hi, seed=231):
                       f"(decode ts+f64, filter, compact; oracle C+OpenMP)"}
 
 
+def bench_compact(args):
+    """BASELINE config #5: per series k=8 overlapping L0 column groups,
+    GPU merge + dedup-by-ts (newest non-null wins) + re-encode.  One step =
+    merge + re-encode of the whole resident set.  Secondary benchmark line
+    (the driver's headline line is the scan mode)."""
+    import torch
+    import cnosdb_amd as gs
+    rng = np.random.default_rng(231)
+    k, nseries, grid_n = 8, args.series, args.npts
+    eng = gs.Engine(0)
+    grid = T0 + np.arange(grid_n, dtype=np.int64) * NS
+    # per-stream ts layout shared across series (20% collisions via
+    # overlapping random subsets, config #5)
+    t_setup = time.perf_counter()
+    stream_ts = []
+    for f in range(k):
+        take = rng.random(grid_n) < 0.25
+        stream_ts.append(grid[take])
+    uniq = min(64, nseries)
+    gsets, tss, vls = [], [], []
+    total = 0
+    for f in range(k):
+        ts = stream_ts[f]
+        tpage = gs.page_of(ts, gs.CT_TIME)
+        vpages = [gs.page_of(np.round(np.clip(
+            np.cumsum(rng.normal(0, 1, ts.size)) + 50, 0, 100), 2), gs.CT_F64)
+            for _ in range(uniq)]
+        groups = [(s, [(tpage, gs.CT_TIME), (vpages[s % uniq], gs.CT_F64)])
+                  for s in range(nseries)]
+        gset = eng.upload(groups, validate_crc=False)
+        d_ts = torch.zeros(gset.rows, dtype=torch.int64, device="cuda")
+        d_val = torch.zeros(gset.rows, dtype=torch.float64, device="cuda")
+        eng.decode(gset, 0, d_ts)
+        eng.decode(gset, 1, d_val)
+        gsets.append(gset)
+        tss.append(d_ts)
+        vls.append(d_val)
+        total += gset.rows
+    d_ots = torch.zeros(total, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(total, dtype=torch.float64, device="cuda")
+    block_rows = 15625
+    cap = block_rows * 12 + 128
+    setup_s = time.perf_counter() - t_setup
+
+    def step():
+        out_rows, offs = eng.compact_merge(gsets, tss, vls, [None] * k,
+                                           d_ots, d_oval)
+        row_off, rows_arr = [], []
+        for s in range(nseries):
+            r = offs[s]
+            while r < offs[s + 1]:
+                n = min(block_rows, offs[s + 1] - r)
+                row_off.append(r)
+                rows_arr.append(n)
+                r += n
+        npg = len(row_off)
+        d_enc_ts = torch.zeros(npg * cap, dtype=torch.uint8, device="cuda")
+        d_enc_v = torch.zeros(npg * cap, dtype=torch.uint8, device="cuda")
+        eng.encode_pages_dev(0, d_ots, np.array(row_off), np.array(rows_arr),
+                             d_enc_ts, cap)
+        lens = eng.encode_pages_dev(2, d_oval, np.array(row_off),
+                                    np.array(rows_arr), d_enc_v, cap)
+        return out_rows, int(lens.sum())
+
+    for _ in range(args.warmup):
+        out_rows, enc_bytes = step()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        out_rows, enc_bytes = step()
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    value = total * args.steps / dt
+    print(json.dumps({
+        "metric": "compaction merge rows/sec (config #5: k=8 overlapping L0 groups, dedup-by-ts, re-encode)",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1000,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "f64",
+        "data": "synthetic",
+        "config": {"workload": "compaction-merge (BASELINE configs[4])",
+                   "k": k, "series": nseries,
+                   "rows_in": int(total), "rows_out": int(out_rows),
+                   "encoded_out_bytes": enc_bytes,
+                   "block_rows": block_rows, "setup_s": round(setup_s, 1)},
+    }))
+    for g in gsets:
+        g.free()
+    eng.close()
+
+
 def main():
     ap = argparse.ArgumentParser()
+    ap.add_argument("--mode", choices=["scan", "compact"], default="scan")
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
@@ -207,6 +305,14 @@ def main():
     ap.add_argument("--unique", type=int, default=256)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
+
+    if args.mode == "compact":
+        if args.series == 10000:
+            args.series = 512
+        if args.npts == 1_000_000:
+            args.npts = 100_000
+        bench_compact(args)
+        return
 
     import torch
     import cnosdb_amd as gs
