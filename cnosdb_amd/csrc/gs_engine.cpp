@@ -979,7 +979,202 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
     }
 }
 
-/* --------------------------------------------------- GPU page re-encode
+/* --------------------------------------------------- fused filtered scan
+ * Fast path of gs_scan when every ts page is RLE (the TSBS shape), every
+ * field page is all-valid Gorilla and no tombstones apply: the span of
+ * the closed time range is computed from the RLE header in closed form
+ * (no ts materialization), the Gorilla decoder writes ONLY the selected
+ * rows, already compacted, and the aggregate runs over the compacted
+ * output — the decode work is identical (every record is still parsed;
+ * only the stores of filtered-out rows are skipped), which is exactly the
+ * operator fusion the north star asks for. */
+
+/* closed-form span of closed [lo,hi] on an RLE ts page */
+__global__ void k_spans_rle(const uint8_t *__restrict__ blob,
+                            const DevPage *__restrict__ pages, int npages,
+                            int64_t lo, int64_t hi,
+                            int64_t *__restrict__ sp_start,
+                            int64_t *__restrict__ sp_cnt,
+                            unsigned *__restrict__ err) {
+    for (int p = blockIdx.x * blockDim.x + threadIdx.x; p < npages;
+         p += gridDim.x * blockDim.x) {
+        DevPage pg = pages[p];
+        const uint8_t *s = blob + pg.data_off + 1;
+        uint64_t scaler = 1;
+        unsigned s10 = s[0] & 0x0f;
+        for (unsigned k = 0; k < s10; k++) scaler *= 10;
+        const uint8_t *q = s + 1;
+        int64_t first = int64_t(dev_be64(q));
+        uint64_t dv; uint32_t nr;
+        dev_varint(q + 8, pg.data_len - 10, &dv, &nr);
+        int64_t delta = int64_t(dv * scaler);
+        int64_t n = pg.nrows;
+        int64_t st, en; /* selected rows = [st, en) */
+        if (delta < 0) { atomicOr(err, DERR_FORMAT); st = en = 0; }
+        else if (delta == 0) {
+            bool in = first >= lo && first <= hi;
+            st = 0; en = in ? n : 0;
+        } else {
+            __int128 d = delta;
+            __int128 a = (__int128)lo - first;
+            __int128 b = (__int128)hi - first;
+            __int128 s0 = a <= 0 ? 0 : (a + d - 1) / d; /* ceil */
+            __int128 e0 = b < 0 ? 0 : b / d + 1;        /* floor + 1 */
+            if (s0 > n) s0 = n;
+            if (e0 > n) e0 = n;
+            st = int64_t(s0);
+            en = int64_t(e0);
+        }
+        if (en < st) en = st;
+        sp_start[p] = st;
+        sp_cnt[p] = en - st;
+    }
+}
+
+/* RLE ts generation of only the selected span, compacted */
+__global__ void k_rle_ts_filtered(const uint8_t *__restrict__ blob,
+                                  const DevPage *__restrict__ pages,
+                                  int npages,
+                                  const int64_t *__restrict__ sp_start,
+                                  const int64_t *__restrict__ sp_cnt,
+                                  const int64_t *__restrict__ out_off,
+                                  int64_t *__restrict__ out_ts) {
+    for (int p = blockIdx.x; p < npages; p += gridDim.x) {
+        DevPage pg = pages[p];
+        const uint8_t *s = blob + pg.data_off + 1;
+        uint64_t scaler = 1;
+        unsigned s10 = s[0] & 0x0f;
+        for (unsigned k = 0; k < s10; k++) scaler *= 10;
+        const uint8_t *q = s + 1;
+        int64_t first = int64_t(dev_be64(q));
+        uint64_t dv; uint32_t nr;
+        dev_varint(q + 8, pg.data_len - 10, &dv, &nr);
+        int64_t delta = int64_t(dv * scaler);
+        int64_t st = sp_start[p], cnt = sp_cnt[p];
+        int64_t *o = out_ts + out_off[p];
+        for (int64_t j = threadIdx.x; j < cnt; j += blockDim.x)
+            o[j] = int64_t(uint64_t(first) + uint64_t(st + j) * uint64_t(delta));
+    }
+}
+
+/* Gorilla decode writing only the selected span, compacted (LDS-staged
+ * cooperative flush as k_gor_lds; each lane's staged run is contiguous in
+ * the output by construction) */
+__global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
+                                   const DevPage *__restrict__ pages,
+                                   int npages,
+                                   const int64_t *__restrict__ sp_start,
+                                   const int64_t *__restrict__ sp_cnt,
+                                   const int64_t *__restrict__ out_off,
+                                   double *__restrict__ out,
+                                   unsigned *__restrict__ err) {
+    __shared__ double ring[256 / 64][GS_RING][64 + 1];
+    const int lane = threadIdx.x & 63;
+    const int wv = threadIdx.x >> 6;
+    auto rslot = ring[wv];
+    int stride = gridDim.x * blockDim.x;
+    int base_id = blockIdx.x * blockDim.x + threadIdx.x;
+    int rounds = (npages + stride - 1) / stride;
+    for (int rd = 0; rd < rounds; rd++) {
+        int p0 = base_id + rd * stride;
+        bool have = p0 < npages;
+        DevPage pg = pages[have ? p0 : 0];
+        const uint8_t *s = blob + pg.data_off + 1;
+        int64_t sel_lo = have ? sp_start[p0] : 0;
+        int64_t sel_hi = have ? sel_lo + sp_cnt[p0] : 0;
+        double *o = have ? out + out_off[p0] - sel_lo : out; /* o[r] valid for r in span */
+        uint32_t nrows = pg.nrows;
+        uint32_t slen = pg.data_len - 1;
+        uint64_t val = dev_be64(s + 1);
+        const uint8_t *p = s + 9;
+        int64_t budget = int64_t(slen - 9) * 8;
+        uint64_t hi = 0, lo = 0;
+        int nb = 0;
+        uint32_t trailing = 0, meaningful = 64;
+        int64_t r = 0;   /* rows produced */
+        int rfill = 0;   /* staged */
+        int64_t run0 = 0; /* output row of first staged entry */
+        bool done = !have;
+        if (have && slen < 9) { atomicOr(err, DERR_SHORT); done = true; }
+        uint64_t nextw = dev_be64(p);
+        p += 8;
+        auto topup = [&]() {
+            uint64_t x = nextw;
+            nextw = dev_be64(p);
+            p += 8;
+            if (nb == 0) { hi = x; lo = 0; }
+            else { hi |= x >> nb; lo = x << (64 - nb); }
+            nb += 64;
+        };
+        auto consume = [&](unsigned k) {
+            hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
+            lo = (k == 64) ? 0 : (lo << k);
+            nb -= int(k);
+            budget -= int64_t(k);
+        };
+        auto flush = [&]() {
+            for (int sl = 0; sl < 64; sl++) {
+                unsigned long long ob =
+                    __shfl((unsigned long long)(uintptr_t)o, sl, 64);
+                int cnt = __shfl(rfill, sl, 64);
+                long long r0 = __shfl((long long)run0, sl, 64);
+                if (lane < cnt)
+                    ((double *)(uintptr_t)ob)[r0 + lane] = rslot[lane][sl];
+            }
+            rfill = 0;
+        };
+        auto stage_row = [&](uint64_t bits_) {
+            if (r >= sel_lo && r < sel_hi) {
+                if (rfill == 0) run0 = r;
+                rslot[rfill][lane] = __longlong_as_double((long long)bits_);
+                rfill++;
+            }
+            r++;
+        };
+        if (!done && r < int64_t(nrows)) stage_row(val);
+        while (!__all(done)) {
+            if (!done) {
+                if (nb < 64) topup();
+                if (budget <= 0) { atomicOr(err, DERR_SHORT); done = true; }
+            }
+            if (!done) {
+                uint32_t top13 = uint32_t(hi >> 51);
+                bool stg = true;
+                if (!(top13 & 0x1000)) {
+                    consume(1);
+                } else {
+                    if (top13 & 0x0800) {
+                        uint32_t lead = (top13 >> 6) & 0x1f;
+                        meaningful = top13 & 0x3f;
+                        if (meaningful > 0) trailing = 64 - lead - meaningful;
+                        else { trailing = 0; meaningful = 64; }
+                        consume(13);
+                    } else {
+                        consume(2);
+                    }
+                    while (nb < int(meaningful)) topup();
+                    uint64_t sb =
+                        (meaningful == 64) ? hi : (hi >> (64 - meaningful));
+                    consume(meaningful);
+                    if (budget < 0) {
+                        atomicOr(err, DERR_SHORT);
+                        done = true;
+                        stg = false;
+                    } else {
+                        val ^= sb << trailing;
+                        if (val == GORILLA_SENTINEL) { done = true; stg = false; }
+                    }
+                }
+                if (stg && r < int64_t(nrows)) stage_row(val);
+            }
+            if (__any(rfill == GS_RING)) flush();
+        }
+        flush();
+        if (have && r < int64_t(nrows)) atomicOr(err, DERR_SHORT);
+    }
+}
+
+/* --------------------------------------------------- GPU page re-encode */
  * The write side of compaction (tsm/writer.rs:249-314 via
  * Page::arrow_array_to_page, tsm/page.rs:100-353): one thread per output
  * page, byte-exact with the host encoders in gs_encode.cpp (which are
@@ -1303,6 +1498,9 @@ struct GsGroupSet {
     DevGroup *d_sgroups = nullptr; /* series-level (consecutive same-series
                                       groups merged) for aggregation */
     int nsgroups = 0;
+    std::vector<int32_t> sgroup_span; /* page-groups per series-group */
+    DevGroup *d_sgroups_out = nullptr; /* series-group layout of the fused
+                                          compacted output (rebuilt per scan) */
     int64_t *d_sp_start = nullptr;
     int64_t *d_sp_cnt = nullptr;
     int64_t *d_out_off = nullptr;
@@ -1571,8 +1769,10 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         if (!hsg.empty() && g > 0 &&
             groups[g].series_id == groups[g - 1].series_id) {
             hsg.back().nrows += hg[g].nrows;
+            set->sgroup_span.back()++;
         } else {
             hsg.push_back(hg[g]);
+            set->sgroup_span.push_back(1);
         }
     }
     set->nsgroups = int(hsg.size());
@@ -1605,6 +1805,7 @@ void gs_groups_free(GsGroupSet *set) {
     hipFree(set->d_blob);
     hipFree(set->d_groups);
     if (set->d_sgroups) hipFree(set->d_sgroups);
+    if (set->d_sgroups_out) hipFree(set->d_sgroups_out);
     hipFree(set->d_sp_start);
     hipFree(set->d_sp_cnt);
     hipFree(set->d_out_off);
@@ -1727,6 +1928,110 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
         hipEvent_t *e;
         ~EvGuard() { for (int k = 0; k < 6; k++) hipEventDestroy(e[k]); }
     } guard{ev};
+
+    /* fused filtered path: spans closed-form from RLE ts headers, Gorilla
+     * decode writes only the selected rows already compacted, aggregate
+     * runs over the compacted output.  Preconditions: every ts page RLE,
+     * every field page all-valid Gorilla, no tombstones, compacted
+     * outputs requested.  Falls back to the general path otherwise. */
+    bool fused = spec->d_out_ts && spec->d_out_val &&
+                 spec->n_tombstones == 0 && !set->any_nulls_field &&
+                 set->ncols == 2 &&
+                 set->slots[0].n[PC_RLE_TS] == int(set->ngroups) &&
+                 set->slots[1].n[PC_GOR] == int(set->ngroups);
+    if (fused) {
+        int ng = int(set->ngroups);
+        const DevPage *ts_pages = set->slots[0].dev[PC_RLE_TS];
+        const DevPage *f_pages = set->slots[1].dev[PC_GOR];
+        HIP_TRY(hipEventRecord(ev[0], ctx->stream));
+        hipLaunchKernelGGL(k_spans_rle, dim3(grid_for(ng, 256)), dim3(256), 0,
+                           ctx->stream, set->d_blob, ts_pages, ng,
+                           spec->range.min_ts, spec->range.max_ts,
+                           set->d_sp_start, set->d_sp_cnt, ctx->d_err);
+        std::vector<int64_t> cnt(set->ngroups), off(set->ngroups);
+        HIP_TRY(hipStreamSynchronize(ctx->stream));
+        HIP_TRY(hipMemcpy(cnt.data(), set->d_sp_cnt,
+                          set->ngroups * sizeof(int64_t),
+                          hipMemcpyDeviceToHost));
+        int64_t acc = 0;
+        for (size_t g = 0; g < set->ngroups; g++) { off[g] = acc; acc += cnt[g]; }
+        HIP_TRY(hipMemcpyAsync(set->d_out_off, off.data(),
+                               set->ngroups * sizeof(int64_t),
+                               hipMemcpyHostToDevice, ctx->stream));
+        HIP_TRY(hipEventRecord(ev[1], ctx->stream));
+        hipLaunchKernelGGL(k_rle_ts_filtered, dim3(ng > 2048 ? 2048 : ng),
+                           dim3(256), 0, ctx->stream, set->d_blob, ts_pages,
+                           ng, set->d_sp_start, set->d_sp_cnt, set->d_out_off,
+                           spec->d_out_ts);
+        HIP_TRY(hipEventRecord(ev[2], ctx->stream));
+        hipLaunchKernelGGL(k_gor_lds_filtered, dim3(grid_for(ng, 256)),
+                           dim3(256), 0, ctx->stream, set->d_blob, f_pages,
+                           ng, set->d_sp_start, set->d_sp_cnt, set->d_out_off,
+                           spec->d_out_val, ctx->d_err);
+        HIP_TRY(hipEventRecord(ev[3], ctx->stream));
+        HIP_TRY(hipEventRecord(ev[4], ctx->stream));
+        if (spec->n_buckets > 0) {
+            if (!spec->d_agg_max || !spec->d_agg_sum || !spec->d_agg_count)
+                return fail(GS_ERR, "agg outputs missing");
+            /* series-group layout of the compacted output */
+            int nsg = set->nsgroups;
+            std::vector<DevGroup> hso(nsg);
+            size_t g = 0;
+            for (int s = 0; s < nsg; s++) {
+                hso[s].row_off = off[g];
+                int64_t rows = 0;
+                for (int32_t k = 0; k < set->sgroup_span[s]; k++, g++)
+                    rows += cnt[g];
+                hso[s].nrows = int32_t(rows);
+                hso[s].pad = 0;
+            }
+            if (!set->d_sgroups_out)
+                HIP_TRY(hipMalloc(&set->d_sgroups_out, nsg * sizeof(DevGroup)));
+            HIP_TRY(hipMemcpyAsync(set->d_sgroups_out, hso.data(),
+                                   nsg * sizeof(DevGroup),
+                                   hipMemcpyHostToDevice, ctx->stream));
+            size_t cells = size_t(nsg) * size_t(spec->n_buckets);
+            if (set->partials_cap < cells) {
+                if (set->d_pmax) hipFree(set->d_pmax);
+                if (set->d_psum) hipFree(set->d_psum);
+                if (set->d_pcnt) hipFree(set->d_pcnt);
+                if (hipMalloc(&set->d_pmax, cells * 8) != hipSuccess ||
+                    hipMalloc(&set->d_psum, cells * 8) != hipSuccess ||
+                    hipMalloc(&set->d_pcnt, cells * 8) != hipSuccess)
+                    return fail(GS_ERR, "hipMalloc agg partials failed");
+                set->partials_cap = cells;
+            }
+            hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 2048 ? 2048 : nsg),
+                               dim3(256), 0, ctx->stream, set->d_sgroups_out,
+                               nsg, spec->d_out_ts, spec->d_out_val, nullptr,
+                               INT64_MIN, INT64_MAX, spec->t0, spec->bucket_ns,
+                               spec->n_buckets, set->d_pmax, set->d_psum,
+                               set->d_pcnt);
+            int mb = (spec->n_buckets + 3) / 4;
+            hipLaunchKernelGGL(k_agg_merge, dim3(mb > 2048 ? 2048 : mb),
+                               dim3(256), 0, ctx->stream, nsg, spec->n_buckets,
+                               set->d_pmax, set->d_psum, set->d_pcnt,
+                               spec->d_agg_max, spec->d_agg_sum,
+                               spec->d_agg_count);
+        }
+        HIP_TRY(hipEventRecord(ev[5], ctx->stream));
+        HIP_TRY(hipStreamSynchronize(ctx->stream));
+        GsStatus st = check_dev_err(ctx);
+        if (st != GS_OK) return st;
+        float ms;
+        HIP_TRY(hipEventElapsedTime(&ms, ev[0], ev[1]));
+        result->ms_filter = ms;
+        HIP_TRY(hipEventElapsedTime(&ms, ev[1], ev[2]));
+        result->ms_decode_ts = ms;
+        HIP_TRY(hipEventElapsedTime(&ms, ev[2], ev[3]));
+        result->ms_decode_val = ms;
+        result->ms_compact = 0.0;
+        HIP_TRY(hipEventElapsedTime(&ms, ev[4], ev[5]));
+        result->ms_agg = ms;
+        result->out_rows = acc;
+        result->decoded_rows = set->total_rows;
+        return GS_OK;
+    }
 
     /* validity bytes needed if the field has nulls or tombstones apply */
     uint8_t *d_valid = nullptr;
