@@ -1174,7 +1174,7 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
     }
 }
 
-/* --------------------------------------------------- GPU page re-encode */
+/* --------------------------------------------------- GPU page re-encode
  * The write side of compaction (tsm/writer.rs:249-314 via
  * Page::arrow_array_to_page, tsm/page.rs:100-353): one thread per output
  * page, byte-exact with the host encoders in gs_encode.cpp (which are
