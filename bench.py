@@ -406,12 +406,15 @@ def main():
     value = total_values / dt
     ms_per_step = dt / args.steps * 1000
 
-    # roofline: dominant kernel = k_gor_lds (Gorilla decode), 1 launch per
-    # sub-batch. algorithmic bytes per launch = compressed f64 bytes read +
-    # 8 B/row decoded output written.
+    # roofline: dominant kernel = k_gor_lds_filtered (fused Gorilla decode),
+    # 1 launch per sub-batch. algorithmic bytes per launch = compressed f64
+    # bytes read + 8 B per SELECTED row written (the fused kernel skips the
+    # stores of filtered-out rows; SURVEY.md §8d: count bytes actually
+    # moved).  The unfused number (8 B x all rows) is stated in DESIGN.md.
     raw_f64_bytes = raw_f64_bytes_sb0 - \
         (16 + (page_rows + 7) // 8) * (sb_rows // page_rows)
-    alg_bytes_launch = raw_f64_bytes + 8 * sb_rows
+    sel_rows_sb = int(out_rows) // args.sub_batches
+    alg_bytes_launch = raw_f64_bytes + 8 * sel_rows_sb
     ms_gorilla_launch = phases[1] / (args.steps * args.sub_batches)
     achieved = alg_bytes_launch / (ms_gorilla_launch / 1000) if ms_gorilla_launch > 0 else 0
     peak = 8.0e12
@@ -420,7 +423,7 @@ def main():
     if os.path.exists(pmc_path):
         try:
             pmc = json.load(open(pmc_path))
-            if pmc.get("kernel") == "k_gor_lds":
+            if pmc.get("kernel") == "k_gor_lds_filtered":
                 traffic = pmc.get("bytes_per_launch")
         except Exception:
             pass

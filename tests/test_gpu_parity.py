@@ -300,6 +300,75 @@ def test_scan_tombstones(engine):
     gset.free()
 
 
+def test_scan_filter_compact_irregular_ts(engine):
+    """Irregular (simple8b) timestamps force the general scan path (the
+    fused path requires RLE ts pages); outputs must still match oracle."""
+    r = np.random.default_rng(13)
+    t0 = 1_700_000_000_000_000_000
+    groups, truth = [], []
+    for s in range(16):
+        ts = t0 + np.sort(r.choice(np.arange(20000, dtype=np.int64),
+                                   4096, replace=False)) * 1_000_000_000
+        vals = np.round(np.clip(np.cumsum(r.normal(0, 0.5, 4096)) + 50, 0, 100), 1)
+        groups.append((s, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (gs.page_of(vals, gs.CT_F64), gs.CT_F64)]))
+        truth.append((ts, vals))
+    gset = engine.upload(groups)
+    rows = gset.rows
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    lo = t0 + 2000 * 10**9
+    hi = t0 + 15000 * 10**9
+    res = engine.scan(gset, d_ts, d_val, time_range=(lo, hi),
+                      d_out_ts=d_ots, d_out_val=d_oval)
+    exp_ts, exp_val = [], []
+    for ts, vals in truth:
+        s0, c = orc.time_span(ts, lo, hi)
+        exp_ts.append(ts[s0:s0 + c])
+        exp_val.append(vals[s0:s0 + c])
+    exp_ts = np.concatenate(exp_ts)
+    exp_val = np.concatenate(exp_val).astype(np.float64)
+    assert res.out_rows == exp_ts.size
+    assert (d_ots[:res.out_rows].cpu().numpy() == exp_ts).all()
+    got = d_oval[:res.out_rows].cpu().numpy()
+    assert got.view(np.uint64).tolist() == exp_val.view(np.uint64).tolist()
+    gset.free()
+
+
+def test_scan_fused_agg_equals_general(engine):
+    """The fused path's aggregates must equal the general path's (the
+    general path is forced by omitting compacted outputs)."""
+    gset, truth, t0 = _mk_scan_set(engine, nseries=12, npts=4096, seed=3)
+    rows = gset.rows
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    bucket_ns = 300_000_000_000
+    nb = 16
+    lo = t0 + 500 * 10**9
+    hi = t0 + 3900 * 10**9
+    results = []
+    for with_compact in (False, True):
+        d_max = torch.full((nb,), -np.inf, dtype=torch.float64, device="cuda")
+        d_sum = torch.zeros(nb, dtype=torch.float64, device="cuda")
+        d_cnt = torch.zeros(nb, dtype=torch.int64, device="cuda")
+        kw = {}
+        if with_compact:
+            kw["d_out_ts"] = torch.zeros(rows, dtype=torch.int64, device="cuda")
+            kw["d_out_val"] = torch.zeros(rows, dtype=torch.float64, device="cuda")
+        engine.scan(gset, d_ts, d_val, time_range=(lo, hi),
+                    agg=dict(bucket_ns=bucket_ns, t0=t0, n_buckets=nb,
+                             d_max=d_max, d_sum=d_sum, d_count=d_cnt), **kw)
+        results.append((d_max.cpu().numpy(), d_sum.cpu().numpy(),
+                        d_cnt.cpu().numpy()))
+    (m0, s0, c0), (m1, s1, c1) = results
+    assert (c0 == c1).all()
+    assert (m0[c0 > 0] == m1[c0 > 0]).all()
+    assert np.allclose(s0, s1, rtol=1e-12)
+    gset.free()
+
+
 def test_crc_validation_rejects_corruption(engine):
     ts = np.arange(100, dtype=np.int64)
     page = bytearray(gs.page_of(ts, gs.CT_TIME))
