@@ -989,6 +989,78 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
  * only the stores of filtered-out rows are skipped), which is exactly the
  * operator fusion the north star asks for. */
 
+/* two-level device exclusive scan of int64 counts (n <= 2048*2048);
+ * out has n+1 entries, out[n] = total */
+#define SCAN_BLOCK 256
+#define SCAN_ITEMS 8
+__global__ void k_scan_partials(const int64_t *__restrict__ in, int n,
+                                int64_t *__restrict__ out,
+                                int64_t *__restrict__ blocksums) {
+    __shared__ int64_t sh[SCAN_BLOCK];
+    int chunk = SCAN_BLOCK * SCAN_ITEMS;
+    int base = blockIdx.x * chunk;
+    int64_t vals[SCAN_ITEMS];
+    int64_t acc = 0;
+    for (int k = 0; k < SCAN_ITEMS; k++) {
+        int i = base + threadIdx.x * SCAN_ITEMS + k;
+        vals[k] = acc;
+        acc += (i < n) ? in[i] : 0;
+    }
+    sh[threadIdx.x] = acc;
+    __syncthreads();
+    /* block scan (exclusive) over per-thread sums */
+    for (int off = 1; off < SCAN_BLOCK; off <<= 1) {
+        int64_t v = (threadIdx.x >= off) ? sh[threadIdx.x - off] : 0;
+        __syncthreads();
+        sh[threadIdx.x] += v;
+        __syncthreads();
+    }
+    int64_t tbase = (threadIdx.x > 0) ? sh[threadIdx.x - 1] : 0;
+    for (int k = 0; k < SCAN_ITEMS; k++) {
+        int i = base + threadIdx.x * SCAN_ITEMS + k;
+        if (i < n) out[i] = tbase + vals[k];
+    }
+    if (threadIdx.x == SCAN_BLOCK - 1) blocksums[blockIdx.x] = sh[threadIdx.x];
+}
+
+__global__ void k_scan_fixup(int n, int64_t *__restrict__ out,
+                             int64_t *__restrict__ blocksums, int nblocks) {
+    /* single thread: scan the (<=2048) block sums in place + write total */
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        int64_t acc = 0;
+        for (int i = 0; i < nblocks; i++) {
+            int64_t v = blocksums[i];
+            blocksums[i] = acc;
+            acc += v;
+        }
+        out[n] = acc; /* total */
+    }
+}
+
+__global__ void k_scan_add(int n, int64_t *__restrict__ out,
+                           const int64_t *__restrict__ blocksums) {
+    int chunk = SCAN_BLOCK * SCAN_ITEMS;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x)
+        out[i] += blocksums[i / chunk];
+}
+
+/* series-group layout of the fused compacted output, built on device:
+ * sgroup s covers page-groups [first[s], first[s+1]) */
+__global__ void k_build_sgroups_out(const int32_t *__restrict__ first,
+                                    int nsg,
+                                    const int64_t *__restrict__ out_off,
+                                    DevGroup *__restrict__ sg) {
+    for (int s = blockIdx.x * blockDim.x + threadIdx.x; s < nsg;
+         s += gridDim.x * blockDim.x) {
+        int64_t a = out_off[first[s]];
+        int64_t b = out_off[first[s + 1]];
+        sg[s].row_off = a;
+        sg[s].nrows = int32_t(b - a);
+        sg[s].pad = 0;
+    }
+}
+
 /* closed-form span of closed [lo,hi] on an RLE ts page */
 __global__ void k_spans_rle(const uint8_t *__restrict__ blob,
                             const DevPage *__restrict__ pages, int npages,
@@ -1501,6 +1573,8 @@ struct GsGroupSet {
     std::vector<int32_t> sgroup_span; /* page-groups per series-group */
     DevGroup *d_sgroups_out = nullptr; /* series-group layout of the fused
                                           compacted output (rebuilt per scan) */
+    int32_t *d_sgroup_first = nullptr; /* [nsgroups+1] first page-group idx */
+    int64_t *d_blocksums = nullptr;    /* device-scan scratch */
     int64_t *d_sp_start = nullptr;
     int64_t *d_sp_cnt = nullptr;
     int64_t *d_out_off = nullptr;
@@ -1782,13 +1856,30 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
     }
     hipMemcpyAsync(set->d_sgroups, hsg.data(), hsg.size() * sizeof(DevGroup),
                    hipMemcpyHostToDevice, ctx->stream);
+    size_t nblocks = (ngroups + SCAN_BLOCK * SCAN_ITEMS - 1) /
+                     (SCAN_BLOCK * SCAN_ITEMS);
+    std::vector<int32_t> sgfirst(hsg.size() + 1);
+    {
+        int32_t acc32 = 0;
+        for (size_t s = 0; s < hsg.size(); s++) {
+            sgfirst[s] = acc32;
+            acc32 += set->sgroup_span[s];
+        }
+        sgfirst[hsg.size()] = acc32;
+    }
     if (hipMalloc(&set->d_groups, ngroups * sizeof(DevGroup)) != hipSuccess ||
         hipMalloc(&set->d_sp_start, ngroups * sizeof(int64_t)) != hipSuccess ||
         hipMalloc(&set->d_sp_cnt, ngroups * sizeof(int64_t)) != hipSuccess ||
-        hipMalloc(&set->d_out_off, ngroups * sizeof(int64_t)) != hipSuccess) {
+        hipMalloc(&set->d_out_off, (ngroups + 1) * sizeof(int64_t)) != hipSuccess ||
+        hipMalloc(&set->d_blocksums, (nblocks + 1) * sizeof(int64_t)) != hipSuccess ||
+        hipMalloc(&set->d_sgroup_first, sgfirst.size() * sizeof(int32_t)) != hipSuccess ||
+        hipMalloc(&set->d_sgroups_out, hsg.size() * sizeof(DevGroup)) != hipSuccess) {
         fail(GS_ERR, "hipMalloc group tables failed");
         delete set; return nullptr;
     }
+    hipMemcpyAsync(set->d_sgroup_first, sgfirst.data(),
+                   sgfirst.size() * sizeof(int32_t), hipMemcpyHostToDevice,
+                   ctx->stream);
     hipMemcpyAsync(set->d_groups, hg.data(), ngroups * sizeof(DevGroup),
                    hipMemcpyHostToDevice, ctx->stream);
     HIP_TRY_NULL(hipStreamSynchronize(ctx->stream));
@@ -1806,6 +1897,8 @@ void gs_groups_free(GsGroupSet *set) {
     hipFree(set->d_groups);
     if (set->d_sgroups) hipFree(set->d_sgroups);
     if (set->d_sgroups_out) hipFree(set->d_sgroups_out);
+    if (set->d_sgroup_first) hipFree(set->d_sgroup_first);
+    if (set->d_blocksums) hipFree(set->d_blocksums);
     hipFree(set->d_sp_start);
     hipFree(set->d_sp_cnt);
     hipFree(set->d_out_off);
@@ -1943,21 +2036,23 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
         int ng = int(set->ngroups);
         const DevPage *ts_pages = set->slots[0].dev[PC_RLE_TS];
         const DevPage *f_pages = set->slots[1].dev[PC_GOR];
+        int nblocks = (ng + SCAN_BLOCK * SCAN_ITEMS - 1) /
+                      (SCAN_BLOCK * SCAN_ITEMS);
+        if (nblocks > 2048)
+            return fail(GS_ERR, "too many groups for the fused span scan");
         HIP_TRY(hipEventRecord(ev[0], ctx->stream));
         hipLaunchKernelGGL(k_spans_rle, dim3(grid_for(ng, 256)), dim3(256), 0,
                            ctx->stream, set->d_blob, ts_pages, ng,
                            spec->range.min_ts, spec->range.max_ts,
                            set->d_sp_start, set->d_sp_cnt, ctx->d_err);
-        std::vector<int64_t> cnt(set->ngroups), off(set->ngroups);
-        HIP_TRY(hipStreamSynchronize(ctx->stream));
-        HIP_TRY(hipMemcpy(cnt.data(), set->d_sp_cnt,
-                          set->ngroups * sizeof(int64_t),
-                          hipMemcpyDeviceToHost));
-        int64_t acc = 0;
-        for (size_t g = 0; g < set->ngroups; g++) { off[g] = acc; acc += cnt[g]; }
-        HIP_TRY(hipMemcpyAsync(set->d_out_off, off.data(),
-                               set->ngroups * sizeof(int64_t),
-                               hipMemcpyHostToDevice, ctx->stream));
+        /* device exclusive scan of span counts -> output offsets (+total) */
+        hipLaunchKernelGGL(k_scan_partials, dim3(nblocks), dim3(SCAN_BLOCK), 0,
+                           ctx->stream, set->d_sp_cnt, ng, set->d_out_off,
+                           set->d_blocksums);
+        hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(1), 0, ctx->stream,
+                           ng, set->d_out_off, set->d_blocksums, nblocks);
+        hipLaunchKernelGGL(k_scan_add, dim3(grid_for(ng, 256)), dim3(256), 0,
+                           ctx->stream, ng, set->d_out_off, set->d_blocksums);
         HIP_TRY(hipEventRecord(ev[1], ctx->stream));
         hipLaunchKernelGGL(k_rle_ts_filtered, dim3(ng > 2048 ? 2048 : ng),
                            dim3(256), 0, ctx->stream, set->d_blob, ts_pages,
@@ -1973,23 +2068,11 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
         if (spec->n_buckets > 0) {
             if (!spec->d_agg_max || !spec->d_agg_sum || !spec->d_agg_count)
                 return fail(GS_ERR, "agg outputs missing");
-            /* series-group layout of the compacted output */
+            /* series-group layout of the compacted output, built on device */
             int nsg = set->nsgroups;
-            std::vector<DevGroup> hso(nsg);
-            size_t g = 0;
-            for (int s = 0; s < nsg; s++) {
-                hso[s].row_off = off[g];
-                int64_t rows = 0;
-                for (int32_t k = 0; k < set->sgroup_span[s]; k++, g++)
-                    rows += cnt[g];
-                hso[s].nrows = int32_t(rows);
-                hso[s].pad = 0;
-            }
-            if (!set->d_sgroups_out)
-                HIP_TRY(hipMalloc(&set->d_sgroups_out, nsg * sizeof(DevGroup)));
-            HIP_TRY(hipMemcpyAsync(set->d_sgroups_out, hso.data(),
-                                   nsg * sizeof(DevGroup),
-                                   hipMemcpyHostToDevice, ctx->stream));
+            hipLaunchKernelGGL(k_build_sgroups_out, dim3(grid_for(nsg, 256)),
+                               dim3(256), 0, ctx->stream, set->d_sgroup_first,
+                               nsg, set->d_out_off, set->d_sgroups_out);
             size_t cells = size_t(nsg) * size_t(spec->n_buckets);
             if (set->partials_cap < cells) {
                 if (set->d_pmax) hipFree(set->d_pmax);
@@ -2018,6 +2101,9 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
         HIP_TRY(hipStreamSynchronize(ctx->stream));
         GsStatus st = check_dev_err(ctx);
         if (st != GS_OK) return st;
+        int64_t acc = 0;
+        HIP_TRY(hipMemcpy(&acc, set->d_out_off + ng, sizeof(int64_t),
+                          hipMemcpyDeviceToHost));
         float ms;
         HIP_TRY(hipEventElapsedTime(&ms, ev[0], ev[1]));
         result->ms_filter = ms;
