@@ -1575,6 +1575,9 @@ struct GsGroupSet {
                                           compacted output (rebuilt per scan) */
     int32_t *d_sgroup_first = nullptr; /* [nsgroups+1] first page-group idx */
     int64_t *d_blocksums = nullptr;    /* device-scan scratch */
+    hipEvent_t sev[6];                 /* fused-scan phase events */
+    bool sev_init = false;
+    bool pending = false;              /* an async fused scan is in flight */
     int64_t *d_sp_start = nullptr;
     int64_t *d_sp_cnt = nullptr;
     int64_t *d_out_off = nullptr;
@@ -1899,6 +1902,8 @@ void gs_groups_free(GsGroupSet *set) {
     if (set->d_sgroups_out) hipFree(set->d_sgroups_out);
     if (set->d_sgroup_first) hipFree(set->d_sgroup_first);
     if (set->d_blocksums) hipFree(set->d_blocksums);
+    if (set->sev_init)
+        for (int k = 0; k < 6; k++) hipEventDestroy(set->sev[k]);
     hipFree(set->d_sp_start);
     hipFree(set->d_sp_cnt);
     hipFree(set->d_out_off);
@@ -2004,6 +2009,131 @@ GsStatus gs_apply_tombstone(GsCtx *ctx, GsGroupSet *set, const int64_t *d_ts,
     return GS_OK;
 }
 
+static bool fused_capable(GsGroupSet *set, const GsScanSpec *spec) {
+    return spec->d_out_ts && spec->d_out_val && spec->n_tombstones == 0 &&
+           !set->any_nulls_field && set->ncols == 2 &&
+           set->slots[0].n[PC_RLE_TS] == int(set->ngroups) &&
+           set->slots[1].n[PC_GOR] == int(set->ngroups);
+}
+
+static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
+                                  const GsScanSpec *spec) {
+    HIP_TRY(hipSetDevice(ctx->device));
+    if (!set->sev_init) {
+        for (int k = 0; k < 6; k++) HIP_TRY(hipEventCreate(&set->sev[k]));
+        set->sev_init = true;
+    }
+    hipEvent_t *ev = set->sev;
+    int ng = int(set->ngroups);
+    const DevPage *ts_pages = set->slots[0].dev[PC_RLE_TS];
+    const DevPage *f_pages = set->slots[1].dev[PC_GOR];
+    int nblocks = (ng + SCAN_BLOCK * SCAN_ITEMS - 1) /
+                  (SCAN_BLOCK * SCAN_ITEMS);
+    if (nblocks > 2048)
+        return fail(GS_ERR, "too many groups for the fused span scan");
+    HIP_TRY(hipEventRecord(ev[0], ctx->stream));
+    hipLaunchKernelGGL(k_spans_rle, dim3(grid_for(ng, 256)), dim3(256), 0,
+                       ctx->stream, set->d_blob, ts_pages, ng,
+                       spec->range.min_ts, spec->range.max_ts,
+                       set->d_sp_start, set->d_sp_cnt, ctx->d_err);
+    /* device exclusive scan of span counts -> output offsets (+total) */
+    hipLaunchKernelGGL(k_scan_partials, dim3(nblocks), dim3(SCAN_BLOCK), 0,
+                       ctx->stream, set->d_sp_cnt, ng, set->d_out_off,
+                       set->d_blocksums);
+    hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(1), 0, ctx->stream, ng,
+                       set->d_out_off, set->d_blocksums, nblocks);
+    hipLaunchKernelGGL(k_scan_add, dim3(grid_for(ng, 256)), dim3(256), 0,
+                       ctx->stream, ng, set->d_out_off, set->d_blocksums);
+    HIP_TRY(hipEventRecord(ev[1], ctx->stream));
+    hipLaunchKernelGGL(k_rle_ts_filtered, dim3(ng > 2048 ? 2048 : ng),
+                       dim3(256), 0, ctx->stream, set->d_blob, ts_pages, ng,
+                       set->d_sp_start, set->d_sp_cnt, set->d_out_off,
+                       spec->d_out_ts);
+    HIP_TRY(hipEventRecord(ev[2], ctx->stream));
+    hipLaunchKernelGGL(k_gor_lds_filtered, dim3(grid_for(ng, 256)), dim3(256),
+                       0, ctx->stream, set->d_blob, f_pages, ng,
+                       set->d_sp_start, set->d_sp_cnt, set->d_out_off,
+                       spec->d_out_val, ctx->d_err);
+    HIP_TRY(hipEventRecord(ev[3], ctx->stream));
+    HIP_TRY(hipEventRecord(ev[4], ctx->stream));
+    if (spec->n_buckets > 0) {
+        if (!spec->d_agg_max || !spec->d_agg_sum || !spec->d_agg_count)
+            return fail(GS_ERR, "agg outputs missing");
+        int nsg = set->nsgroups;
+        hipLaunchKernelGGL(k_build_sgroups_out, dim3(grid_for(nsg, 256)),
+                           dim3(256), 0, ctx->stream, set->d_sgroup_first,
+                           nsg, set->d_out_off, set->d_sgroups_out);
+        size_t cells = size_t(nsg) * size_t(spec->n_buckets);
+        if (set->partials_cap < cells) {
+            if (set->d_pmax) hipFree(set->d_pmax);
+            if (set->d_psum) hipFree(set->d_psum);
+            if (set->d_pcnt) hipFree(set->d_pcnt);
+            if (hipMalloc(&set->d_pmax, cells * 8) != hipSuccess ||
+                hipMalloc(&set->d_psum, cells * 8) != hipSuccess ||
+                hipMalloc(&set->d_pcnt, cells * 8) != hipSuccess)
+                return fail(GS_ERR, "hipMalloc agg partials failed");
+            set->partials_cap = cells;
+        }
+        hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 2048 ? 2048 : nsg),
+                           dim3(256), 0, ctx->stream, set->d_sgroups_out, nsg,
+                           spec->d_out_ts, spec->d_out_val, nullptr,
+                           INT64_MIN, INT64_MAX, spec->t0, spec->bucket_ns,
+                           spec->n_buckets, set->d_pmax, set->d_psum,
+                           set->d_pcnt);
+        int mb = (spec->n_buckets + 3) / 4;
+        hipLaunchKernelGGL(k_agg_merge, dim3(mb > 2048 ? 2048 : mb), dim3(256),
+                           0, ctx->stream, nsg, spec->n_buckets, set->d_pmax,
+                           set->d_psum, set->d_pcnt, spec->d_agg_max,
+                           spec->d_agg_sum, spec->d_agg_count);
+    }
+    HIP_TRY(hipEventRecord(ev[5], ctx->stream));
+    set->pending = true;
+    return GS_OK;
+}
+
+static GsStatus scan_fused_wait(GsCtx *ctx, GsGroupSet *set,
+                                GsScanResult *result) {
+    if (!set->pending)
+        return fail(GS_ERR, "no fused scan in flight on this set");
+    set->pending = false;
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    GsStatus st = check_dev_err(ctx);
+    if (st != GS_OK) return st;
+    int64_t acc = 0;
+    HIP_TRY(hipMemcpy(&acc, set->d_out_off + set->ngroups, sizeof(int64_t),
+                      hipMemcpyDeviceToHost));
+    hipEvent_t *ev = set->sev;
+    float ms;
+    HIP_TRY(hipEventElapsedTime(&ms, ev[0], ev[1]));
+    result->ms_filter = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, ev[1], ev[2]));
+    result->ms_decode_ts = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, ev[2], ev[3]));
+    result->ms_decode_val = ms;
+    result->ms_compact = 0.0;
+    HIP_TRY(hipEventElapsedTime(&ms, ev[4], ev[5]));
+    result->ms_agg = ms;
+    result->out_rows = acc;
+    result->decoded_rows = set->total_rows;
+    return GS_OK;
+}
+
+GsStatus gs_scan_async(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec) {
+    if (!ctx || !set || !spec)
+        return fail(GS_ERR, "bad args to gs_scan_async");
+    if (!fused_capable(set, spec))
+        return fail(GS_ERR, "gs_scan_async requires the fused-capable shape "
+                            "(RLE ts + all-valid Gorilla, no tombstones, "
+                            "compacted outputs)");
+    return scan_fused_launch(ctx, set, spec);
+}
+
+GsStatus gs_scan_wait(GsCtx *ctx, GsGroupSet *set, GsScanResult *result) {
+    if (!ctx || !set || !result)
+        return fail(GS_ERR, "bad args to gs_scan_wait");
+    return scan_fused_wait(ctx, set, result);
+}
+
 GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                  GsScanResult *result) {
     if (!ctx || !set || !spec || !result || !spec->d_ts || !spec->d_val)
@@ -2027,96 +2157,10 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
      * runs over the compacted output.  Preconditions: every ts page RLE,
      * every field page all-valid Gorilla, no tombstones, compacted
      * outputs requested.  Falls back to the general path otherwise. */
-    bool fused = spec->d_out_ts && spec->d_out_val &&
-                 spec->n_tombstones == 0 && !set->any_nulls_field &&
-                 set->ncols == 2 &&
-                 set->slots[0].n[PC_RLE_TS] == int(set->ngroups) &&
-                 set->slots[1].n[PC_GOR] == int(set->ngroups);
-    if (fused) {
-        int ng = int(set->ngroups);
-        const DevPage *ts_pages = set->slots[0].dev[PC_RLE_TS];
-        const DevPage *f_pages = set->slots[1].dev[PC_GOR];
-        int nblocks = (ng + SCAN_BLOCK * SCAN_ITEMS - 1) /
-                      (SCAN_BLOCK * SCAN_ITEMS);
-        if (nblocks > 2048)
-            return fail(GS_ERR, "too many groups for the fused span scan");
-        HIP_TRY(hipEventRecord(ev[0], ctx->stream));
-        hipLaunchKernelGGL(k_spans_rle, dim3(grid_for(ng, 256)), dim3(256), 0,
-                           ctx->stream, set->d_blob, ts_pages, ng,
-                           spec->range.min_ts, spec->range.max_ts,
-                           set->d_sp_start, set->d_sp_cnt, ctx->d_err);
-        /* device exclusive scan of span counts -> output offsets (+total) */
-        hipLaunchKernelGGL(k_scan_partials, dim3(nblocks), dim3(SCAN_BLOCK), 0,
-                           ctx->stream, set->d_sp_cnt, ng, set->d_out_off,
-                           set->d_blocksums);
-        hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(1), 0, ctx->stream,
-                           ng, set->d_out_off, set->d_blocksums, nblocks);
-        hipLaunchKernelGGL(k_scan_add, dim3(grid_for(ng, 256)), dim3(256), 0,
-                           ctx->stream, ng, set->d_out_off, set->d_blocksums);
-        HIP_TRY(hipEventRecord(ev[1], ctx->stream));
-        hipLaunchKernelGGL(k_rle_ts_filtered, dim3(ng > 2048 ? 2048 : ng),
-                           dim3(256), 0, ctx->stream, set->d_blob, ts_pages,
-                           ng, set->d_sp_start, set->d_sp_cnt, set->d_out_off,
-                           spec->d_out_ts);
-        HIP_TRY(hipEventRecord(ev[2], ctx->stream));
-        hipLaunchKernelGGL(k_gor_lds_filtered, dim3(grid_for(ng, 256)),
-                           dim3(256), 0, ctx->stream, set->d_blob, f_pages,
-                           ng, set->d_sp_start, set->d_sp_cnt, set->d_out_off,
-                           spec->d_out_val, ctx->d_err);
-        HIP_TRY(hipEventRecord(ev[3], ctx->stream));
-        HIP_TRY(hipEventRecord(ev[4], ctx->stream));
-        if (spec->n_buckets > 0) {
-            if (!spec->d_agg_max || !spec->d_agg_sum || !spec->d_agg_count)
-                return fail(GS_ERR, "agg outputs missing");
-            /* series-group layout of the compacted output, built on device */
-            int nsg = set->nsgroups;
-            hipLaunchKernelGGL(k_build_sgroups_out, dim3(grid_for(nsg, 256)),
-                               dim3(256), 0, ctx->stream, set->d_sgroup_first,
-                               nsg, set->d_out_off, set->d_sgroups_out);
-            size_t cells = size_t(nsg) * size_t(spec->n_buckets);
-            if (set->partials_cap < cells) {
-                if (set->d_pmax) hipFree(set->d_pmax);
-                if (set->d_psum) hipFree(set->d_psum);
-                if (set->d_pcnt) hipFree(set->d_pcnt);
-                if (hipMalloc(&set->d_pmax, cells * 8) != hipSuccess ||
-                    hipMalloc(&set->d_psum, cells * 8) != hipSuccess ||
-                    hipMalloc(&set->d_pcnt, cells * 8) != hipSuccess)
-                    return fail(GS_ERR, "hipMalloc agg partials failed");
-                set->partials_cap = cells;
-            }
-            hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 2048 ? 2048 : nsg),
-                               dim3(256), 0, ctx->stream, set->d_sgroups_out,
-                               nsg, spec->d_out_ts, spec->d_out_val, nullptr,
-                               INT64_MIN, INT64_MAX, spec->t0, spec->bucket_ns,
-                               spec->n_buckets, set->d_pmax, set->d_psum,
-                               set->d_pcnt);
-            int mb = (spec->n_buckets + 3) / 4;
-            hipLaunchKernelGGL(k_agg_merge, dim3(mb > 2048 ? 2048 : mb),
-                               dim3(256), 0, ctx->stream, nsg, spec->n_buckets,
-                               set->d_pmax, set->d_psum, set->d_pcnt,
-                               spec->d_agg_max, spec->d_agg_sum,
-                               spec->d_agg_count);
-        }
-        HIP_TRY(hipEventRecord(ev[5], ctx->stream));
-        HIP_TRY(hipStreamSynchronize(ctx->stream));
-        GsStatus st = check_dev_err(ctx);
-        if (st != GS_OK) return st;
-        int64_t acc = 0;
-        HIP_TRY(hipMemcpy(&acc, set->d_out_off + ng, sizeof(int64_t),
-                          hipMemcpyDeviceToHost));
-        float ms;
-        HIP_TRY(hipEventElapsedTime(&ms, ev[0], ev[1]));
-        result->ms_filter = ms;
-        HIP_TRY(hipEventElapsedTime(&ms, ev[1], ev[2]));
-        result->ms_decode_ts = ms;
-        HIP_TRY(hipEventElapsedTime(&ms, ev[2], ev[3]));
-        result->ms_decode_val = ms;
-        result->ms_compact = 0.0;
-        HIP_TRY(hipEventElapsedTime(&ms, ev[4], ev[5]));
-        result->ms_agg = ms;
-        result->out_rows = acc;
-        result->decoded_rows = set->total_rows;
-        return GS_OK;
+    if (fused_capable(set, spec)) {
+        GsStatus fst = scan_fused_launch(ctx, set, spec);
+        if (fst != GS_OK) return fst;
+        return scan_fused_wait(ctx, set, result);
     }
 
     /* validity bytes needed if the field has nulls or tombstones apply */

@@ -127,6 +127,12 @@ class PageLib:
         lib.gs_scan.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                 ctypes.POINTER(GsScanSpec),
                                 ctypes.POINTER(GsScanResult)]
+        lib.gs_scan_async.restype = ctypes.c_int32
+        lib.gs_scan_async.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                      ctypes.POINTER(GsScanSpec)]
+        lib.gs_scan_wait.restype = ctypes.c_int32
+        lib.gs_scan_wait.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                     ctypes.POINTER(GsScanResult)]
         lib.gs_encode_pages_dev.restype = ctypes.c_int32
         lib.gs_encode_pages_dev.argtypes = [
             ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p,
@@ -405,6 +411,47 @@ class Engine:
         if st != 0:
             raise RuntimeError(f"gs_encode_pages_dev failed ({st}): {self._pl.err()}")
         return lens
+
+    def _mk_spec(self, d_ts, d_val, time_range, tombstones, d_out_ts,
+                 d_out_val, agg):
+        spec = GsScanSpec()
+        lo, hi = time_range if time_range else (-(2**63), 2**63 - 1)
+        spec.range = GsTimeRange(lo, hi)
+        if tombstones:
+            tarr = (GsTimeRange * len(tombstones))(*[GsTimeRange(a, b) for a, b in tombstones])
+            spec.tombstones = tarr
+            spec.n_tombstones = len(tombstones)
+            spec._keep = tarr
+        if d_ts is not None:
+            spec.d_ts = d_ts.data_ptr()
+            spec.d_val = d_val.data_ptr()
+        if d_out_ts is not None:
+            spec.d_out_ts = d_out_ts.data_ptr()
+            spec.d_out_val = d_out_val.data_ptr()
+        if agg:
+            spec.bucket_ns = agg["bucket_ns"]
+            spec.t0 = agg["t0"]
+            spec.n_buckets = agg["n_buckets"]
+            spec.d_agg_max = agg["d_max"].data_ptr()
+            spec.d_agg_sum = agg["d_sum"].data_ptr()
+            spec.d_agg_count = agg["d_count"].data_ptr()
+        return spec
+
+    def scan_async(self, gset, d_out_ts, d_out_val, time_range=None, agg=None):
+        """Enqueue the fused scan without synchronizing (fused-capable
+        shapes only); pair with scan_wait."""
+        spec = self._mk_spec(None, None, time_range, None, d_out_ts,
+                             d_out_val, agg)
+        st = self.lib.gs_scan_async(self._ctx, gset._h, ctypes.byref(spec))
+        if st != 0:
+            raise RuntimeError(f"gs_scan_async failed ({st}): {self._pl.err()}")
+
+    def scan_wait(self, gset):
+        res = GsScanResult()
+        st = self.lib.gs_scan_wait(self._ctx, gset._h, ctypes.byref(res))
+        if st != 0:
+            raise RuntimeError(f"gs_scan_wait failed ({st}): {self._pl.err()}")
+        return res
 
     def scan(self, gset, d_ts, d_val, time_range=None, tombstones=None,
              d_out_ts=None, d_out_val=None, agg=None):

@@ -190,6 +190,16 @@ typedef struct {
 GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                  GsScanResult *result);
 
+/* Async variant of the fused scan: gs_scan_async enqueues the whole
+ * fused pipeline (requires the fused-capable shape: RLE ts pages,
+ * all-valid Gorilla field pages, no tombstones, compacted outputs — the
+ * TSBS shape; `d_ts`/`d_val` scratch is not used) and returns without
+ * synchronizing; gs_scan_wait synchronizes the ctx stream and fills the
+ * result.  Lets a caller overlap sub-batches on two ctx streams (the
+ * Gorilla decode is ALU-bound, the aggregate HBM-bound). */
+GsStatus gs_scan_async(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec);
+GsStatus gs_scan_wait(GsCtx *ctx, GsGroupSet *set, GsScanResult *result);
+
 /* ---- GPU page re-encode (the write side of compaction/flush:
  * Page::arrow_array_to_page, tsm/page.rs:100-353 + tsm/writer.rs:249-314).
  * kind: 0 = ts (DeltaTs), 1 = i64 (Delta), 2 = f64 (Gorilla).  d_vals is a
