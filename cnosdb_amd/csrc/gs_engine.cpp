@@ -398,13 +398,15 @@ __device__ __forceinline__ void gorilla_decode_page(
  * Lanes run in lockstep (one value per iteration), so rings fill together;
  * lanes whose page ended keep cooperating in flushes until all are done. */
 #define GS_RING 16
+#define GS_GOR_BLOCK 128 /* 2 waves: halves LDS per block -> 8 waves/SIMD
+                            occupancy (PMC showed 50% wait at 4 waves) */
 
 __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
                           const DevPage *__restrict__ pages, int npages,
                           double *__restrict__ out,
                           uint8_t *__restrict__ valid,
                           unsigned *__restrict__ err) {
-    __shared__ double ring[256 / 64][GS_RING][64 + 1]; /* [wave][slot][lane] */
+    __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1]; /* [wave][slot][lane] */
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     auto rslot = ring[wv];
@@ -430,10 +432,12 @@ __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
         bool done = !have;
         if (have && slen < 9) { atomicOr(err, DERR_SHORT); done = true; }
         uint64_t nextw = dev_be64(p);
-        p += 8;
+        uint64_t nextw2 = dev_be64(p + 8);
+        p += 16;
         auto topup = [&]() { /* only with nb < 64 */
             uint64_t x = nextw;
-            nextw = dev_be64(p);
+            nextw = nextw2;
+            nextw2 = dev_be64(p);
             p += 8;
             if (nb == 0) { hi = x; lo = 0; }
             else { hi |= x >> nb; lo = x << (64 - nb); }
@@ -1140,7 +1144,7 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
                                    const int64_t *__restrict__ out_off,
                                    double *__restrict__ out,
                                    unsigned *__restrict__ err) {
-    __shared__ double ring[256 / 64][GS_RING][64 + 1];
+    __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1];
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     auto rslot = ring[wv];
@@ -1169,10 +1173,12 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
         bool done = !have;
         if (have && slen < 9) { atomicOr(err, DERR_SHORT); done = true; }
         uint64_t nextw = dev_be64(p);
-        p += 8;
+        uint64_t nextw2 = dev_be64(p + 8);
+        p += 16;
         auto topup = [&]() {
             uint64_t x = nextw;
-            nextw = dev_be64(p);
+            nextw = nextw2;
+            nextw2 = dev_be64(p);
             p += 8;
             if (nb == 0) { hi = x; lo = 0; }
             else { hi |= x >> nb; lo = x << (64 - nb); }
@@ -1968,9 +1974,10 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
     }
     if (sp.n[PC_GOR]) {
         int n = sp.n[PC_GOR];
-        hipLaunchKernelGGL(k_gor_lds, dim3(grid_for(n, 256)), dim3(256), 0,
-                           ctx->stream, set->d_blob, sp.dev[PC_GOR], n,
-                           (double *)d_out, d_valid, ctx->d_err);
+        hipLaunchKernelGGL(k_gor_lds, dim3(grid_for(n, GS_GOR_BLOCK)),
+                           dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
+                           sp.dev[PC_GOR], n, (double *)d_out, d_valid,
+                           ctx->d_err);
     }
     if (sp.n[PC_RLE_TS])
         hipLaunchKernelGGL(k_rle_par,
@@ -2050,10 +2057,10 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                        set->d_sp_start, set->d_sp_cnt, set->d_out_off,
                        spec->d_out_ts);
     HIP_TRY(hipEventRecord(ev[2], ctx->stream));
-    hipLaunchKernelGGL(k_gor_lds_filtered, dim3(grid_for(ng, 256)), dim3(256),
-                       0, ctx->stream, set->d_blob, f_pages, ng,
-                       set->d_sp_start, set->d_sp_cnt, set->d_out_off,
-                       spec->d_out_val, ctx->d_err);
+    hipLaunchKernelGGL(k_gor_lds_filtered, dim3(grid_for(ng, GS_GOR_BLOCK)),
+                       dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
+                       f_pages, ng, set->d_sp_start, set->d_sp_cnt,
+                       set->d_out_off, spec->d_out_val, ctx->d_err);
     HIP_TRY(hipEventRecord(ev[3], ctx->stream));
     HIP_TRY(hipEventRecord(ev[4], ctx->stream));
     if (spec->n_buckets > 0) {
