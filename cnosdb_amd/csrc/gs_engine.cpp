@@ -398,7 +398,8 @@ __device__ __forceinline__ void gorilla_decode_page(
  * Lanes run in lockstep (one value per iteration), so rings fill together;
  * lanes whose page ended keep cooperating in flushes until all are done. */
 #define GS_RING 16
-#define GS_GOR_BLOCK 256 /* 4 waves (128-thread/8-wave variant measured equal) */
+#define GS_GOR_BLOCK 128 /* 2 waves: halves LDS per block -> 8 waves/SIMD
+                            occupancy (PMC showed 50% wait at 4 waves) */
 
 __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
                           const DevPage *__restrict__ pages, int npages,
@@ -1215,48 +1216,24 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
                 if (budget <= 0) { atomicOr(err, DERR_SHORT); done = true; }
             }
             if (!done) {
-                if (int64_t(hi) >= 0) {
-                    /* ctrl-0 repeat RUN: collapse consecutive zero bits in
-                       one clz — each is one repeated value (float.rs:67-71).
-                       Capped by the window, the ring headroom and the rows
-                       left so the cooperative flush point stays wave-wide. */
-                    int z = __builtin_clzll(hi | 1);
-                    if (z > nb) z = nb;
-                    if (z > GS_RING - rfill) z = GS_RING - rfill;
-                    int64_t left = int64_t(nrows) - r;
-                    if (int64_t(z) > left) z = int(left);
-                    if (z < 1) z = 1; /* row budget exhausted: consume 1 */
-                    consume(unsigned(z));
-                    if (budget < 0) { atomicOr(err, DERR_SHORT); done = true; }
-                    else
-                        for (int k = 0; k < z && r < int64_t(nrows); k++)
-                            stage_row(val);
+                uint32_t top13 = uint32_t(hi >> 51);
+                bool stg = true;
+                if (!(top13 & 0x1000)) {
+                    consume(1);
                 } else {
-                    uint32_t top13 = uint32_t(hi >> 51);
-                    bool stg = true;
-                    unsigned hdr;
-                    if (top13 & 0x0800) { /* ctrl 11: new window */
+                    if (top13 & 0x0800) {
                         uint32_t lead = (top13 >> 6) & 0x1f;
                         meaningful = top13 & 0x3f;
                         if (meaningful > 0) trailing = 64 - lead - meaningful;
                         else { trailing = 0; meaningful = 64; }
-                        hdr = 13;
+                        consume(13);
                     } else {
-                        hdr = 2;
+                        consume(2);
                     }
-                    uint64_t sb;
-                    if (hdr + meaningful <= 64) {
-                        /* whole record inside the 64-bit view: one consume */
-                        sb = (hi >> (64 - hdr - meaningful)) &
-                             ((meaningful == 64) ? ~0ULL
-                                                 : ((1ULL << meaningful) - 1));
-                        consume(hdr + meaningful);
-                    } else {
-                        consume(hdr);
-                        while (nb < int(meaningful)) topup();
-                        sb = (meaningful == 64) ? hi : (hi >> (64 - meaningful));
-                        consume(meaningful);
-                    }
+                    while (nb < int(meaningful)) topup();
+                    uint64_t sb =
+                        (meaningful == 64) ? hi : (hi >> (64 - meaningful));
+                    consume(meaningful);
                     if (budget < 0) {
                         atomicOr(err, DERR_SHORT);
                         done = true;
@@ -1265,8 +1242,8 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
                         val ^= sb << trailing;
                         if (val == GORILLA_SENTINEL) { done = true; stg = false; }
                     }
-                    if (stg && r < int64_t(nrows)) stage_row(val);
                 }
+                if (stg && r < int64_t(nrows)) stage_row(val);
             }
             if (__any(rfill == GS_RING)) flush();
         }
