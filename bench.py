@@ -321,13 +321,12 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     dist = None
+    device = torch.device(f"cuda:{local_rank}")
+    torch.cuda.set_device(device)
     if world > 1:
         import torch.distributed as tdist
         dist = tdist
-        dist.init_process_group("nccl")
-        torch.cuda.set_device(local_rank)
-    device = torch.device(f"cuda:{local_rank}")
-    torch.cuda.set_device(device)
+        dist.init_process_group("nccl", device_id=device)
 
     nseries, npts, page_rows = args.series, args.npts, args.page_rows
     assert npts % page_rows == 0

@@ -20,6 +20,13 @@
 #include "../../include/cnosdb_gs.h"
 #include "gs_internal.h"
 
+/* hipError_t returns are checked with HIP_TRY on every path that can fail
+ * user-visibly; cleanup paths (hipFree/hipEventDestroy in destructors) and
+ * async enqueues deliberately drop the return — every enqueue is followed
+ * by a checked hipStreamSynchronize on the same stream, which surfaces any
+ * deferred error.  Silence the nodiscard noise for those. */
+#pragma clang diagnostic ignored "-Wunused-value"
+
 #include <cstdio>
 #include <cstring>
 #include <cstdlib>
@@ -1922,6 +1929,19 @@ void gs_groups_free(GsGroupSet *set) {
 
 int64_t gs_set_rows(const GsGroupSet *set) { return set ? set->total_rows : -1; }
 int64_t gs_set_series(const GsGroupSet *set) { return set ? set->nsgroups : -1; }
+
+/* Pushed-down COUNT answered from page metadata, never from decoded data
+ * (PushDownAggregateReader, tskv/src/reader/pushdown_agg_reader.rs:39-106:
+ * sums num_values — rows INCLUDING nulls — of the column's pages; SURVEY.md
+ * A.10.12). */
+int64_t gs_count_pushdown(const GsGroupSet *set, uint32_t col) {
+    if (!set || col >= set->ncols) return -1;
+    int64_t n = 0;
+    const SlotPages &sp = set->slots[col];
+    for (int k = 0; k < PC_NCLASS; k++)
+        for (const DevPage &p : sp.host[k]) n += p.nrows;
+    return n;
+}
 int64_t gs_set_groups(const GsGroupSet *set) { return set ? int64_t(set->ngroups) : -1; }
 GsStatus gs_set_row_offsets(const GsGroupSet *set, int64_t *out) {
     if (!set || !out) return fail(GS_ERR, "bad args");

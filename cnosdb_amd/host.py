@@ -113,6 +113,8 @@ class PageLib:
         lib.gs_set_groups.argtypes = [ctypes.c_void_p]
         lib.gs_set_series.restype = ctypes.c_int64
         lib.gs_set_series.argtypes = [ctypes.c_void_p]
+        lib.gs_count_pushdown.restype = ctypes.c_int64
+        lib.gs_count_pushdown.argtypes = [ctypes.c_void_p, ctypes.c_uint32]
         lib.gs_set_row_offsets.restype = ctypes.c_int32
         lib.gs_set_row_offsets.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
         lib.gs_decode.restype = ctypes.c_int32
@@ -250,6 +252,10 @@ class GroupSet:
         self.ngroups = ngroups
         self._keepalive = keepalive  # page byte buffers must outlive upload
 
+    def count_pushdown(self, col=0):
+        """pushed-down COUNT from page metadata (no decode)."""
+        return PageLib().lib.gs_count_pushdown(self._h, col)
+
     def row_offsets(self):
         out = np.zeros(self.ngroups, dtype=np.int64)
         st = PageLib().lib.gs_set_row_offsets(self._h, _np_ptr(out))
@@ -323,7 +329,9 @@ class Engine:
             raise RuntimeError(f"gs_groups_upload failed: {self._pl.err()}")
         h = ctypes.c_void_p(h)
         rows = self.lib.gs_set_rows(h)
-        return GroupSet(self, h, rows, ngroups, (buf, specs, gdescs))
+        # gs_groups_upload synchronizes after staging all page bytes, so the
+        # host buffer need not be retained (matters at 8 ranks x ~17 GB)
+        return GroupSet(self, h, rows, ngroups, None)
 
     def upload(self, groups, validate_crc=True):
         """groups: list of (series_id, [(page_bytes, ctype), ...]);

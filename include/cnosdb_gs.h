@@ -118,6 +118,8 @@ int64_t gs_set_rows(const GsGroupSet *set);
 int64_t gs_set_groups(const GsGroupSet *set);
 /* series-level group count (consecutive same-series groups merged) */
 int64_t gs_set_series(const GsGroupSet *set);
+/* pushed-down COUNT from page metadata (pushdown_agg_reader.rs:39-106) */
+int64_t gs_count_pushdown(const GsGroupSet *set, uint32_t col);
 /* copy the per-group row offsets (ngroups entries) into caller buffer */
 GsStatus gs_set_row_offsets(const GsGroupSet *set, int64_t *out);
 

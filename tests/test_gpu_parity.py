@@ -300,6 +300,32 @@ def test_scan_tombstones(engine):
     gset.free()
 
 
+def test_decode_u64_bitcast(engine):
+    """u64 fields are bit-cast to i64 and use the Delta codec
+    (unsigned.rs:20-45); decode must round-trip the u64 bit patterns."""
+    cases = []
+    for _ in range(6):
+        n = int(rng.integers(1, 3000))
+        cases.append(rng.integers(0, 2**64, n, dtype=np.uint64))
+    groups = []
+    for i, vals in enumerate(cases):
+        ts = np.arange(len(vals), dtype=np.int64) * 1000
+        fpage = gs.page_of(vals.view(np.int64), gs.CT_I64)  # encode as i64
+        groups.append((i, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (fpage, gs.CT_U64)]))
+    gset = engine.upload(groups)
+    out = torch.zeros(gset.rows, dtype=torch.int64, device="cuda")
+    engine.decode(gset, 1, out)
+    offs = gset.row_offsets()
+    host = out.cpu().numpy().view(np.uint64)
+    for i, vals in enumerate(cases):
+        got = host[offs[i]:offs[i] + len(vals)]
+        assert (got == vals).all(), f"u64 case {i}"
+    # COUNT pushdown from metadata
+    assert gset.count_pushdown(1) == sum(len(v) for v in cases)
+    gset.free()
+
+
 def test_scan_filter_compact_irregular_ts(engine):
     """Irregular (simple8b) timestamps force the general scan path (the
     fused path requires RLE ts pages); outputs must still match oracle."""
