@@ -42,7 +42,7 @@ T0 = 1_700_000_000_000_000_000
 BUCKET_NS = 300 * NS  # 5 minutes
 
 
-def build_workload(nseries, npts, page_rows, unique, sub_batches, seed=231):
+def build_workload(nseries, npts, page_rows, unique, sub_batches, nfields=1, seed=231):
     """Generate encoded TSM pages. `unique` distinct value series are
     generated and replicated across series (device copies are distinct, so
     HBM traffic is real); ts pages are identical across series (TSBS
@@ -104,17 +104,17 @@ def build_workload(nseries, npts, page_rows, unique, sub_batches, seed=231):
         sids = np.repeat(np.arange(sb * per_sb, (sb + 1) * per_sb,
                                    dtype=np.uint32), npages)
         for s in range(sb * per_sb, (sb + 1) * per_sb):
-            vp = val_pages[pattern[s]]
             for p in range(npages):
                 parts.append(ts_pages[p])
-                parts.append(vp[p])
+                for f in range(nfields):
+                    parts.append(val_pages[(pattern[s] + f) % unique][p])
         buf = b"".join(parts)
         lens = np.array([len(x) for x in parts], dtype=np.int64)
         offs = np.zeros(lens.size, dtype=np.int64)
         np.cumsum(lens[:-1], out=offs[1:])
         nvals = np.full(lens.size, page_rows, dtype=np.int64)
-        cts = np.tile(np.array([gs.CT_TIME, gs.CT_F64], dtype=np.uint8),
-                      lens.size // 2)
+        cts = np.tile(np.array([gs.CT_TIME] + [gs.CT_F64] * nfields,
+                               dtype=np.uint8), lens.size // (1 + nfields))
         raw_bytes += buf.__len__()
         sub.append((buf, offs, lens, nvals, cts, sids))
     return sub, raw_bytes, npages
@@ -303,6 +303,8 @@ def main():
     ap.add_argument("--page-rows", type=int, default=4000)
     ap.add_argument("--sub-batches", type=int, default=4)
     ap.add_argument("--unique", type=int, default=256)
+    ap.add_argument("--fields", type=int, default=1,
+                    help="f64 field pages per group (8 = TSBS cpu-max-all-8, config #3)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -339,7 +341,7 @@ def main():
     t_setup = time.perf_counter()
     sub, raw_bytes, npages = build_workload(
         nseries, npts, page_rows, min(args.unique, nseries),
-        args.sub_batches, seed=231 + rank)
+        args.sub_batches, nfields=args.fields, seed=231 + rank)
     # two engine contexts = two HIP streams on the one device: sub-batches
     # alternate streams so the ALU-bound Gorilla decode of one overlaps the
     # HBM-bound aggregate/ts phases of the other
@@ -349,10 +351,13 @@ def main():
     for sb, (buf, offs, lens, nvals, cts, sids) in enumerate(sub):
         e = engines[sb % 2]
         sets.append(e.upload_packed(buf, offs, lens, nvals, cts, sids,
-                                    pages_per_group=2, validate_crc=False))
+                                    pages_per_group=1 + args.fields,
+                                    validate_crc=False))
         eng_of.append(e)
         if sb == 0:
-            raw_f64_bytes_sb0 = int(lens[1::2].sum())
+            m = np.ones(lens.size, dtype=bool)
+            m[::1 + args.fields] = False  # drop ts pages
+            raw_f64_bytes_sb0 = int(lens[m].sum()) // args.fields
     del sub
     sb_rows = sets[0].rows
     douts = [(torch.zeros(sb_rows, dtype=torch.int64, device=device),
@@ -374,15 +379,17 @@ def main():
     def step():
         out_rows = 0
         phase_ms = np.zeros(5)
-        for i, st_ in enumerate(sets):
-            e = eng_of[i]
-            e.scan_async(st_, douts[i % 2][0], douts[i % 2][1],
-                         time_range=(lo, hi), agg=aggs[i % 2])
-        for i, st_ in enumerate(sets):
-            r = eng_of[i].scan_wait(st_)
-            out_rows += r.out_rows
-            phase_ms += [r.ms_decode_ts, r.ms_decode_val, r.ms_filter,
-                         r.ms_compact, r.ms_agg]
+        for f in range(args.fields):
+            for i, st_ in enumerate(sets):
+                e = eng_of[i]
+                e.scan_async(st_, douts[i % 2][0], douts[i % 2][1],
+                             time_range=(lo, hi), agg=aggs[i % 2],
+                             field_col=f)
+            for i, st_ in enumerate(sets):
+                r = eng_of[i].scan_wait(st_)
+                out_rows += r.out_rows
+                phase_ms += [r.ms_decode_ts, r.ms_decode_val, r.ms_filter,
+                             r.ms_compact, r.ms_agg]
         # combine the per-stream bucket partials (tiny)
         torch.add(aggs[0]["d_sum"], aggs[1]["d_sum"], out=d_sum)
         torch.add(aggs[0]["d_count"], aggs[1]["d_count"], out=d_cnt)
@@ -421,10 +428,10 @@ def main():
     # scan on a single stream, untimed region (stream-distance HIP events
     # are inflated by cross-stream overlap during the timed steps)
     engines[0].scan_async(sets[0], douts[0][0], douts[0][1],
-                          time_range=(lo, hi), agg=aggs[0])
+                          time_range=(lo, hi), agg=aggs[0], field_col=0)
     r_cal = engines[0].scan_wait(sets[0])
 
-    values_per_step = nseries * npts  # field values decoded+filtered per rank
+    values_per_step = nseries * npts * args.fields  # field values decoded+filtered per rank
     total_values = values_per_step * world * args.steps
     value = total_values / dt
     ms_per_step = dt / args.steps * 1000
@@ -474,6 +481,7 @@ def main():
             "config": {
                 "workload": "tsbs-devops-scan (BASELINE configs[1]: 10k series x 1M pts, delta-ts + Gorilla-f64 decode, ts-range filter 50%, fused 5-min max/sum/count)",
                 "series_per_gpu": nseries,
+                "fields": args.fields,
                 "points_per_series": npts,
                 "page_rows": page_rows,
                 "selectivity": 0.5,

@@ -2036,10 +2036,11 @@ GsStatus gs_apply_tombstone(GsCtx *ctx, GsGroupSet *set, const int64_t *d_ts,
 }
 
 static bool fused_capable(GsGroupSet *set, const GsScanSpec *spec) {
+    uint32_t fc = 1 + uint32_t(spec->field_col);
     return spec->d_out_ts && spec->d_out_val && spec->n_tombstones == 0 &&
-           !set->any_nulls_field && set->ncols == 2 &&
+           !set->any_nulls_field && fc < set->ncols &&
            set->slots[0].n[PC_RLE_TS] == int(set->ngroups) &&
-           set->slots[1].n[PC_GOR] == int(set->ngroups);
+           set->slots[fc].n[PC_GOR] == int(set->ngroups);
 }
 
 static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
@@ -2052,7 +2053,7 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
     hipEvent_t *ev = set->sev;
     int ng = int(set->ngroups);
     const DevPage *ts_pages = set->slots[0].dev[PC_RLE_TS];
-    const DevPage *f_pages = set->slots[1].dev[PC_GOR];
+    const DevPage *f_pages = set->slots[1 + spec->field_col].dev[PC_GOR];
     int nblocks = (ng + SCAN_BLOCK * SCAN_ITEMS - 1) /
                   (SCAN_BLOCK * SCAN_ITEMS);
     if (nblocks > 2048)
@@ -2166,8 +2167,9 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
         return fail(GS_ERR, "bad args to gs_scan");
     if (set->ncols < 2)
         return fail(GS_ERR, "gs_scan needs a time page + one f64 field page");
-    if (set->slots[1].ctype != GS_CT_F64)
-        return fail(GS_ERR, "gs_scan field column must be f64");
+    if (spec->field_col < 0 || 1 + uint32_t(spec->field_col) >= set->ncols ||
+        set->slots[1 + spec->field_col].ctype != GS_CT_F64)
+        return fail(GS_ERR, "gs_scan field column must be an f64 slot");
     HIP_TRY(hipSetDevice(ctx->device));
 
     /* per-phase timing on the engine stream (HIP events) */
@@ -2202,7 +2204,8 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
     GsStatus st = gs_decode(ctx, set, 0, spec->d_ts, nullptr);
     if (st != GS_OK) return st;
     HIP_TRY(hipEventRecord(ev[1], ctx->stream));
-    st = gs_decode(ctx, set, 1, spec->d_val, d_valid);
+    st = gs_decode(ctx, set, 1 + uint32_t(spec->field_col), spec->d_val,
+                   d_valid);
     if (st != GS_OK) return st;
     HIP_TRY(hipEventRecord(ev[2], ctx->stream));
 

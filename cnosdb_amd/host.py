@@ -45,6 +45,7 @@ class GsColumnGroupDesc(ctypes.Structure):
 class GsScanSpec(ctypes.Structure):
     _fields_ = [
         ("range", GsTimeRange),
+        ("field_col", ctypes.c_int32),
         ("tombstones", ctypes.POINTER(GsTimeRange)),
         ("n_tombstones", ctypes.c_size_t),
         ("bucket_ns", ctypes.c_int64),
@@ -421,10 +422,11 @@ class Engine:
         return lens
 
     def _mk_spec(self, d_ts, d_val, time_range, tombstones, d_out_ts,
-                 d_out_val, agg):
+                 d_out_val, agg, field_col=0):
         spec = GsScanSpec()
         lo, hi = time_range if time_range else (-(2**63), 2**63 - 1)
         spec.range = GsTimeRange(lo, hi)
+        spec.field_col = field_col
         if tombstones:
             tarr = (GsTimeRange * len(tombstones))(*[GsTimeRange(a, b) for a, b in tombstones])
             spec.tombstones = tarr
@@ -445,11 +447,12 @@ class Engine:
             spec.d_agg_count = agg["d_count"].data_ptr()
         return spec
 
-    def scan_async(self, gset, d_out_ts, d_out_val, time_range=None, agg=None):
+    def scan_async(self, gset, d_out_ts, d_out_val, time_range=None, agg=None,
+                   field_col=0):
         """Enqueue the fused scan without synchronizing (fused-capable
         shapes only); pair with scan_wait."""
         spec = self._mk_spec(None, None, time_range, None, d_out_ts,
-                             d_out_val, agg)
+                             d_out_val, agg, field_col)
         st = self.lib.gs_scan_async(self._ctx, gset._h, ctypes.byref(spec))
         if st != 0:
             raise RuntimeError(f"gs_scan_async failed ({st}): {self._pl.err()}")
@@ -462,10 +465,11 @@ class Engine:
         return res
 
     def scan(self, gset, d_ts, d_val, time_range=None, tombstones=None,
-             d_out_ts=None, d_out_val=None, agg=None):
+             d_out_ts=None, d_out_val=None, agg=None, field_col=0):
         """Fused scan. agg: dict(bucket_ns, t0, n_buckets, d_max, d_sum,
         d_count).  Returns GsScanResult."""
         spec = GsScanSpec()
+        spec.field_col = field_col
         lo, hi = time_range if time_range else (-(2**63), 2**63 - 1)
         spec.range = GsTimeRange(lo, hi)
         if tombstones:

@@ -153,6 +153,11 @@ GsStatus gs_apply_tombstone(GsCtx *ctx, GsGroupSet *set, const int64_t *d_ts,
  */
 typedef struct {
     GsTimeRange range;     /* closed; use INT64_MIN/MAX for no filter */
+    /* column slot of the f64 field to scan (0 = the first field page,
+       i.e. pages[1]).  A multi-metric query (TSBS cpu-max-all-8) scans
+       the same set once per field, reusing the uploaded pages; spans are
+       recomputed per call (cheap, time-only). */
+    int32_t field_col;
     /* deleted time ranges (tombstones) applied to the field column's
        validity before filter/agg (tsm/reader.rs:529-544) */
     const GsTimeRange *tombstones;

@@ -395,6 +395,41 @@ def test_scan_fused_agg_equals_general(engine):
     gset.free()
 
 
+def test_scan_multi_field(engine):
+    """Multi-metric scan (TSBS cpu-max-all style): one set with several f64
+    field pages, scanned per field via spec.field_col."""
+    r = np.random.default_rng(21)
+    t0 = 1_700_000_000_000_000_000
+    nseries, npts, nfields = 8, 4096, 3
+    groups, truth = [], []
+    for s in range(nseries):
+        ts = t0 + np.arange(npts, dtype=np.int64) * 10**9
+        fields = [np.round(np.clip(np.cumsum(r.normal(0, 0.5, npts)) + 50, 0, 100), 1)
+                  for _ in range(nfields)]
+        pages = [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME)]
+        pages += [(gs.page_of(f, gs.CT_F64), gs.CT_F64) for f in fields]
+        groups.append((s, pages))
+        truth.append((ts, fields))
+    gset = engine.upload(groups)
+    rows = gset.rows
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    lo, hi = t0 + 1000 * 10**9, t0 + 3000 * 10**9
+    for fc in range(nfields):
+        engine.scan_async(gset, d_ots, d_oval, time_range=(lo, hi),
+                          field_col=fc)
+        res = engine.scan_wait(gset)
+        exp_v = []
+        for ts, fields in truth:
+            s0, c = orc.time_span(ts, lo, hi)
+            exp_v.append(fields[fc][s0:s0 + c])
+        exp_v = np.concatenate(exp_v).astype(np.float64)
+        assert res.out_rows == exp_v.size
+        got = d_oval[:res.out_rows].cpu().numpy()
+        assert got.view(np.uint64).tolist() == exp_v.view(np.uint64).tolist(), f"field {fc}"
+    gset.free()
+
+
 def test_crc_validation_rejects_corruption(engine):
     ts = np.arange(100, dtype=np.int64)
     page = bytearray(gs.page_of(ts, gs.CT_TIME))
