@@ -24,12 +24,13 @@ from .host import (
     CT_I64,
     CT_F64,
     CT_BOOL,
+    CT_U64,
 )
 
 __all__ = [
     "Engine", "GroupSet", "PageLib", "lib_path",
     "encode_ts", "encode_i64", "encode_f64", "encode_bool",
     "build_page", "page_of",
-    "CT_TIME", "CT_I64", "CT_F64", "CT_BOOL",
+    "CT_TIME", "CT_I64", "CT_F64", "CT_BOOL", "CT_U64",
 ]
 __version__ = "0.1"
