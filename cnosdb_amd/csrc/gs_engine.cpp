@@ -1151,6 +1151,10 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
                                    double *__restrict__ out,
                                    unsigned *__restrict__ err) {
     __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1];
+    /* per-flush lane descriptors: {dst pointer, staged count} packed so the
+       flush loop reads ONE broadcast ds_read per source lane instead of
+       three cross-lane shuffles */
+    __shared__ uint64_t fdesc[GS_GOR_BLOCK / 64][64][2];
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     auto rslot = ring[wv];
@@ -1196,14 +1200,16 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
             nb -= int(k);
             budget -= int64_t(k);
         };
+        auto fd = fdesc[wv];
         auto flush = [&]() {
+            fd[lane][0] = (uint64_t)(uintptr_t)(o + run0);
+            fd[lane][1] = uint64_t(rfill);
+            __builtin_amdgcn_wave_barrier();
             for (int sl = 0; sl < 64; sl++) {
-                unsigned long long ob =
-                    __shfl((unsigned long long)(uintptr_t)o, sl, 64);
-                int cnt = __shfl(rfill, sl, 64);
-                long long r0 = __shfl((long long)run0, sl, 64);
+                uint64_t ob = fd[sl][0];
+                int cnt = int(fd[sl][1]);
                 if (lane < cnt)
-                    ((double *)(uintptr_t)ob)[r0 + lane] = rslot[lane][sl];
+                    ((double *)(uintptr_t)ob)[lane] = rslot[lane][sl];
             }
             rfill = 0;
         };
