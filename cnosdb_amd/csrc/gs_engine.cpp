@@ -413,6 +413,7 @@ __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
                           uint8_t *__restrict__ valid,
                           unsigned *__restrict__ err) {
     __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1]; /* [wave][slot][lane] */
+    __shared__ uint64_t fdesc[GS_GOR_BLOCK / 64][64][2];
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     auto rslot = ring[wv];
@@ -455,14 +456,16 @@ __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
             nb -= int(k);
             budget -= int64_t(k);
         };
+        auto fd = fdesc[wv];
         auto flush = [&]() {
+            fd[lane][0] = (uint64_t)(uintptr_t)(o + (r - rfill));
+            fd[lane][1] = uint64_t(rfill);
+            __builtin_amdgcn_wave_barrier();
             for (int sl = 0; sl < 64; sl++) {
-                unsigned long long ob =
-                    __shfl((unsigned long long)(uintptr_t)o, sl, 64);
-                int cnt = __shfl(rfill, sl, 64);
-                int row0 = __shfl(r, sl, 64) - cnt;
+                uint64_t ob = fd[sl][0];
+                int cnt = int(fd[sl][1]);
                 if (lane < cnt)
-                    ((double *)(uintptr_t)ob)[row0 + lane] = rslot[lane][sl];
+                    ((double *)(uintptr_t)ob)[lane] = rslot[lane][sl];
             }
             rfill = 0;
         };
