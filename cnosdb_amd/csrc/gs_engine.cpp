@@ -1697,15 +1697,15 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         uint32_t nrows_g = groups[g].pages[0].num_values;
         for (uint32_t c = 0; c < ncols; c++) {
             const GsPageSpec &ps = groups[g].pages[c];
-            if (ps.len < 16) { fail(GS_ERR_FORMAT, "page too short"); delete set; return nullptr; }
+            if (ps.len < 16) { fail(GS_ERR_FORMAT, "page too short"); gs_groups_free(set); return nullptr; }
             uint32_t bl = (uint32_t(ps.bytes[0]) << 24) | (uint32_t(ps.bytes[1]) << 16) |
                           (uint32_t(ps.bytes[2]) << 8) | ps.bytes[3];
             uint64_t dl64 = 0;
             for (int k = 0; k < 8; k++) dl64 = (dl64 << 8) | ps.bytes[4 + k];
-            if (16ull + bl > ps.len) { fail(GS_ERR_FORMAT, "bitset overruns page"); delete set; return nullptr; }
+            if (16ull + bl > ps.len) { fail(GS_ERR_FORMAT, "bitset overruns page"); gs_groups_free(set); return nullptr; }
             if (dl64 != ps.num_values || ps.num_values != nrows_g) {
                 fail(GS_ERR_FORMAT, "row count mismatch in group");
-                delete set; return nullptr;
+                gs_groups_free(set); return nullptr;
             }
             uint64_t data_len = ps.len - 16 - bl;
             const uint8_t *bitset = ps.bytes + 16;
@@ -1715,7 +1715,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                                (uint32_t(ps.bytes[14]) << 8) | ps.bytes[15];
                 if (gs_crc32(data, data_len) != crc) {
                     fail(GS_ERR_CRC, "page crc32 mismatch");
-                    delete set; return nullptr;
+                    gs_groups_free(set); return nullptr;
                 }
             }
             /* all_valid: full bytes 0xff, partial last byte has low bits set */
@@ -1755,7 +1755,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
             hp.dp.all_valid = av ? 1 : 0;
             if (c == 0 && !av) {
                 fail(GS_ERR_FORMAT, "time page must be fully valid");
-                delete set; return nullptr;
+                gs_groups_free(set); return nullptr;
             }
             if (c > 0 && !av) set->any_nulls_field = true;
             staged.push_back(hp);
@@ -1770,7 +1770,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
      * hipMemcpyAsync would cost ~4 us apiece — 246k copies measured) */
     if (hipMalloc(&set->d_blob, blob ? blob : 16) != hipSuccess) {
         fail(GS_ERR, "hipMalloc blob failed (out of HBM?)");
-        delete set; return nullptr;
+        gs_groups_free(set); return nullptr;
     }
     {
         const size_t CHUNK = size_t(256) << 20;
@@ -1779,7 +1779,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         for (int k = 0; k < 2; k++) {
             if (hipHostMalloc(&stage[k], CHUNK) != hipSuccess) {
                 fail(GS_ERR, "hipHostMalloc staging failed");
-                delete set; return nullptr;
+                gs_groups_free(set); return nullptr;
             }
             hipEventCreate(&evs[k]);
             hipEventRecord(evs[k], ctx->stream);
@@ -1807,7 +1807,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
             if (pi == start_pi) { /* single page larger than CHUNK */
                 fail(GS_ERR, "page larger than staging chunk");
                 for (int k = 0; k < 2; k++) { hipHostFree(stage[k]); hipEventDestroy(evs[k]); }
-                delete set; return nullptr;
+                gs_groups_free(set); return nullptr;
             }
             hipMemcpyAsync(set->d_blob + base, stage[cur], fill_end - base,
                            hipMemcpyHostToDevice, ctx->stream);
@@ -1825,7 +1825,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         if (sp.ctype == 255) sp.ctype = hp.dp.ctype;
         if (hp.dp.ctype != sp.ctype) {
             fail(GS_ERR, "mixed ctypes in one column slot");
-            delete set; return nullptr;
+            gs_groups_free(set); return nullptr;
         }
         int cls = PC_SEQ;
         if (hp.dp.all_valid && hp.dp.sub == 2 && hp.dp.data_len > 2) {
@@ -1844,7 +1844,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
             if (sp.n[k]) {
                 if (hipMalloc(&sp.dev[k], sp.n[k] * sizeof(DevPage)) != hipSuccess) {
                     fail(GS_ERR, "hipMalloc page table failed");
-                    delete set; return nullptr;
+                    gs_groups_free(set); return nullptr;
                 }
                 hipMemcpyAsync(sp.dev[k], sp.host[k].data(),
                                sp.n[k] * sizeof(DevPage),
@@ -1876,7 +1876,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
     set->nsgroups = int(hsg.size());
     if (hipMalloc(&set->d_sgroups, hsg.size() * sizeof(DevGroup)) != hipSuccess) {
         fail(GS_ERR, "hipMalloc sgroup table failed");
-        delete set; return nullptr;
+        gs_groups_free(set); return nullptr;
     }
     hipMemcpyAsync(set->d_sgroups, hsg.data(), hsg.size() * sizeof(DevGroup),
                    hipMemcpyHostToDevice, ctx->stream);
@@ -1899,7 +1899,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         hipMalloc(&set->d_sgroup_first, sgfirst.size() * sizeof(int32_t)) != hipSuccess ||
         hipMalloc(&set->d_sgroups_out, hsg.size() * sizeof(DevGroup)) != hipSuccess) {
         fail(GS_ERR, "hipMalloc group tables failed");
-        delete set; return nullptr;
+        gs_groups_free(set); return nullptr;
     }
     hipMemcpyAsync(set->d_sgroup_first, sgfirst.data(),
                    sgfirst.size() * sizeof(int32_t), hipMemcpyHostToDevice,
