@@ -31,6 +31,7 @@
 #include <cstring>
 #include <cstdlib>
 #include <vector>
+#include <algorithm>
 #include <string>
 
 /* ------------------------------------------------------------- error glue */
@@ -1111,8 +1112,8 @@ __global__ void k_spans_rle(const uint8_t *__restrict__ blob,
             en = int64_t(e0);
         }
         if (en < st) en = st;
-        sp_start[p] = st;
-        sp_cnt[p] = en - st;
+        sp_start[pg.grp] = st;
+        sp_cnt[pg.grp] = en - st;
     }
 }
 
@@ -1135,8 +1136,8 @@ __global__ void k_rle_ts_filtered(const uint8_t *__restrict__ blob,
         uint64_t dv; uint32_t nr;
         dev_varint(q + 8, pg.data_len - 10, &dv, &nr);
         int64_t delta = int64_t(dv * scaler);
-        int64_t st = sp_start[p], cnt = sp_cnt[p];
-        int64_t *o = out_ts + out_off[p];
+        int64_t st = sp_start[pg.grp], cnt = sp_cnt[pg.grp];
+        int64_t *o = out_ts + out_off[pg.grp];
         for (int64_t j = threadIdx.x; j < cnt; j += blockDim.x)
             o[j] = int64_t(uint64_t(first) + uint64_t(st + j) * uint64_t(delta));
     }
@@ -1169,9 +1170,9 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
         bool have = p0 < npages;
         DevPage pg = pages[have ? p0 : 0];
         const uint8_t *s = blob + pg.data_off + 1;
-        int64_t sel_lo = have ? sp_start[p0] : 0;
-        int64_t sel_hi = have ? sel_lo + sp_cnt[p0] : 0;
-        double *o = have ? out + out_off[p0] - sel_lo : out; /* o[r] valid for r in span */
+        int64_t sel_lo = have ? sp_start[pg.grp] : 0;
+        int64_t sel_hi = have ? sel_lo + sp_cnt[pg.grp] : 0;
+        double *o = have ? out + out_off[pg.grp] - sel_lo : out; /* o[r] valid for r in span */
         uint32_t nrows = pg.nrows;
         uint32_t slen = pg.data_len - 1;
         uint64_t val = dev_be64(s + 1);
@@ -1753,6 +1754,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                                            hp.dp.enc == GS_ENC_DELTATS))
                             ? (data[1] >> 4) : 0;
             hp.dp.all_valid = av ? 1 : 0;
+            hp.dp.grp = uint32_t(g);
             if (c == 0 && !av) {
                 fail(GS_ERR_FORMAT, "time page must be fully valid");
                 gs_groups_free(set); return nullptr;
@@ -1839,6 +1841,12 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
     }
     for (uint32_t c = 0; c < ncols; c++) {
         SlotPages &sp = set->slots[c];
+        /* longest Gorilla pages first: decode time scales with compressed
+           bits, so launching the slow pages early removes the ragged tail */
+        std::stable_sort(sp.host[PC_GOR].begin(), sp.host[PC_GOR].end(),
+                         [](const DevPage &a, const DevPage &b) {
+                             return a.data_len > b.data_len;
+                         });
         for (int k = 0; k < PC_NCLASS; k++) {
             sp.n[k] = int(sp.host[k].size());
             if (sp.n[k]) {

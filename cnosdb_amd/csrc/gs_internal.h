@@ -17,6 +17,7 @@ struct DevPage {
     uint8_t enc;        /* first byte of data buffer, 0 if empty */
     uint8_t sub;        /* sub-tag high nibble for DELTA/DELTATS, else 0 */
     uint8_t all_valid;  /* host-precomputed: every validity bit set */
+    uint32_t grp;       /* page-group index (class lists may be re-ordered) */
 };
 
 struct DevGroup {
