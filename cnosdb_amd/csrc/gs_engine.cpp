@@ -774,68 +774,12 @@ __global__ void k_agg_partial(const DevGroup *__restrict__ groups, int n,
             }
             __syncthreads();
         }
-        if (use_lds) {
-            /* four buckets per wave: their rows are contiguous, the four
-               accumulator sets are independent, so the four shuffle trees
-               pipeline instead of serializing their cross-lane latency */
-            for (int b0 = wave * 4; b0 < nbuckets; b0 += nwaves * 4) {
-                int nb4 = nbuckets - b0 < 4 ? nbuckets - b0 : 4;
-                int64_t s = soff[b0];
-                int64_t e = soff[b0 + nb4];
-                int64_t q1 = soff[b0 + (nb4 > 1 ? 1 : 0)];
-                int64_t q2 = soff[b0 + (nb4 > 2 ? 2 : 1)];
-                int64_t q3 = soff[b0 + (nb4 > 3 ? 3 : 2)];
-                if (nb4 < 2) q1 = e;
-                if (nb4 < 3) q2 = e;
-                if (nb4 < 4) q3 = e;
-                double mx0 = -__builtin_inf(), mx1 = mx0, mx2 = mx0, mx3 = mx0;
-                double sm0 = 0, sm1 = 0, sm2 = 0, sm3 = 0;
-                long long c0 = 0, c1 = 0, c2 = 0, c3 = 0;
-                for (int64_t r = s + lane; r < e; r += 64) {
-                    bool ok = !vd || vd[r];
-                    double x = ok ? v[r] : 0.0;
-                    int which = (r >= q1) + (r >= q2) + (r >= q3);
-                    bool w0 = ok && which == 0, w1 = ok && which == 1;
-                    bool w2 = ok && which == 2, w3 = ok && which == 3;
-                    if (w0) { if (x > mx0) mx0 = x; sm0 += x; c0++; }
-                    if (w1) { if (x > mx1) mx1 = x; sm1 += x; c1++; }
-                    if (w2) { if (x > mx2) mx2 = x; sm2 += x; c2++; }
-                    if (w3) { if (x > mx3) mx3 = x; sm3 += x; c3++; }
-                }
-                for (int off = 32; off > 0; off >>= 1) {
-                    double o0 = __shfl_down(mx0, off, 64);
-                    double o1 = __shfl_down(mx1, off, 64);
-                    double o2 = __shfl_down(mx2, off, 64);
-                    double o3 = __shfl_down(mx3, off, 64);
-                    double p0 = __shfl_down(sm0, off, 64);
-                    double p1 = __shfl_down(sm1, off, 64);
-                    double p2 = __shfl_down(sm2, off, 64);
-                    double p3 = __shfl_down(sm3, off, 64);
-                    long long q0 = __shfl_down(c0, off, 64);
-                    long long qq1 = __shfl_down(c1, off, 64);
-                    long long qq2 = __shfl_down(c2, off, 64);
-                    long long qq3 = __shfl_down(c3, off, 64);
-                    if (o0 > mx0) mx0 = o0;
-                    if (o1 > mx1) mx1 = o1;
-                    if (o2 > mx2) mx2 = o2;
-                    if (o3 > mx3) mx3 = o3;
-                    sm0 += p0; sm1 += p1; sm2 += p2; sm3 += p3;
-                    c0 += q0; c1 += qq1; c2 += qq2; c3 += qq3;
-                }
-                if (lane == 0) {
-                    size_t idx = size_t(g) * nbuckets + b0;
-                    pmax[idx] = mx0; psum[idx] = sm0; pcnt[idx] = c0;
-                    if (nb4 > 1) { pmax[idx+1] = mx1; psum[idx+1] = sm1; pcnt[idx+1] = c1; }
-                    if (nb4 > 2) { pmax[idx+2] = mx2; psum[idx+2] = sm2; pcnt[idx+2] = c2; }
-                    if (nb4 > 3) { pmax[idx+3] = mx3; psum[idx+3] = sm3; pcnt[idx+3] = c3; }
-                }
-            }
-            __syncthreads();
-            continue;
-        }
         for (int b = wave; b < nbuckets; b += nwaves) {
             int64_t s, e;
-            {
+            if (use_lds) {
+                s = soff[b];
+                e = soff[b + 1];
+            } else {
                 int64_t blo = t0 + int64_t(b) * bucket_ns;
                 int64_t bhi = blo + bucket_ns;
                 if (blo < range_lo) blo = range_lo;
