@@ -42,6 +42,13 @@ T0 = 1_700_000_000_000_000_000
 BUCKET_NS = 300 * NS  # 5 minutes
 
 
+def _host_threads():
+    """OMP threads per rank: divide the host cores among co-located ranks
+    (the round-end 8-GPU run launches 8 ranks on one node)."""
+    lw = int(os.environ.get("LOCAL_WORLD_SIZE", "1"))
+    return max(1, (os.cpu_count() or 8) // max(1, lw))
+
+
 def build_workload(nseries, npts, page_rows, unique, sub_batches, nfields=1, seed=231):
     """Generate encoded TSM pages. `unique` distinct value series are
     generated and replicated across series (device copies are distinct, so
@@ -82,7 +89,7 @@ def build_workload(nseries, npts, page_rows, unique, sub_batches, nfields=1, see
         st = lib.gs_encode_f64_pages_omp(
             flat.ctypes.data_as(ctypes.c_void_p), page_rows, total_pages,
             enc.ctypes.data_as(ctypes.c_void_p), cap,
-            lens.ctypes.data_as(ctypes.c_void_p), os.cpu_count() or 8)
+            lens.ctypes.data_as(ctypes.c_void_p), _host_threads())
         assert st == 0, st
         for k in range(cu):
             row = []
@@ -170,7 +177,7 @@ def cpu_baseline_leg(nseries_sample, npts, page_rows, lo, hi, seed=231):
     rows = total_pages * page_rows
     out_ts = np.zeros(rows, dtype=np.int64)
     out_v = np.zeros(rows, dtype=np.float64)
-    cores = os.cpu_count() or 8
+    cores = _host_threads()
     t = time.perf_counter()
     st = o.orc_decode_pages_omp(descs_t, total_pages,
                                 out_ts.ctypes.data_as(ctypes.c_void_p), cores)
