@@ -2118,7 +2118,7 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                 return fail(GS_ERR, "hipMalloc agg partials failed");
             set->partials_cap = cells;
         }
-        hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 2048 ? 2048 : nsg),
+        hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
                            dim3(256), 0, ctx->stream, set->d_sgroups_out, nsg,
                            spec->d_out_ts, spec->d_out_val, nullptr,
                            INT64_MIN, INT64_MAX, spec->t0, spec->bucket_ns,
@@ -2274,7 +2274,7 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                 return fail(GS_ERR, "hipMalloc agg partials failed");
             set->partials_cap = cells;
         }
-        hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 2048 ? 2048 : nsg),
+        hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
                            dim3(256), 0, ctx->stream, set->d_sgroups, nsg,
                            spec->d_ts, spec->d_val, d_valid,
                            spec->range.min_ts, spec->range.max_ts,
