@@ -750,8 +750,11 @@ __global__ void k_agg_partial(const DevGroup *__restrict__ groups, int n,
     /* bucket-boundary row offsets staged in LDS: one cooperative binary
        search per boundary (256 in flight per block) instead of 2 dependent
        searches per (wave,bucket) — the searches were the latency hot spot */
+    /* dynamic LDS: sized to nbuckets+1 at launch so occupancy is not
+       capped by the worst-case table (32.8 KB static -> 4 blocks/CU even
+       for small bucket counts) */
+    extern __shared__ __attribute__((aligned(16))) uint32_t soff[];
     constexpr int MAXB = 8192;
-    __shared__ uint32_t soff[MAXB + 1];
     const bool use_lds = nbuckets <= MAXB;
     const int64_t hi_open = (range_hi == INT64_MAX) ? INT64_MAX : range_hi + 1;
     for (int g = blockIdx.x; g < n; g += gridDim.x) {
@@ -2118,8 +2121,11 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                 return fail(GS_ERR, "hipMalloc agg partials failed");
             set->partials_cap = cells;
         }
+        size_t agg_shm = spec->n_buckets <= 8192
+                             ? (size_t(spec->n_buckets) + 1) * 4 : 0;
         hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
-                           dim3(256), 0, ctx->stream, set->d_sgroups_out, nsg,
+                           dim3(256), agg_shm, ctx->stream,
+                           set->d_sgroups_out, nsg,
                            spec->d_out_ts, spec->d_out_val, nullptr,
                            INT64_MIN, INT64_MAX, spec->t0, spec->bucket_ns,
                            spec->n_buckets, set->d_pmax, set->d_psum,
@@ -2274,8 +2280,10 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                 return fail(GS_ERR, "hipMalloc agg partials failed");
             set->partials_cap = cells;
         }
+        size_t agg_shm = spec->n_buckets <= 8192
+                             ? (size_t(spec->n_buckets) + 1) * 4 : 0;
         hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
-                           dim3(256), 0, ctx->stream, set->d_sgroups, nsg,
+                           dim3(256), agg_shm, ctx->stream, set->d_sgroups, nsg,
                            spec->d_ts, spec->d_val, d_valid,
                            spec->range.min_ts, spec->range.max_ts,
                            spec->t0, spec->bucket_ns,
