@@ -24,7 +24,7 @@ collective the path needs.
 ranks.  A "value" is one point of the measured f64 field column (1e7*N
 series-points per step... see config).  Roofline + cpu_baseline objects
 per the driver contract; the roofline covers the dominant kernel
-(k_seq_f64, the Gorilla page decoder).
+(k_gor_lds_filtered, the fused Gorilla page decoder).
 """
 import argparse
 import json
