@@ -652,6 +652,133 @@ __global__ void k_rle_par(const uint8_t *__restrict__ blob,
     }
 }
 
+/* Block-parallel simple8b delta decode (all-valid DeltaTs/Delta pages,
+ * sub-tag 1; timestamp.rs:261-299 / integer.rs:216-248): one workgroup
+ * per page.  Words are fixed 8-byte units, so the block scans per-word
+ * (value count, delta sum) pairs with a carry, then each thread unpacks
+ * its word directly to the right output positions — the delta prefix-sum
+ * that is sequential in the reference becomes a block scan. */
+__global__ void k_s8b_par(const uint8_t *__restrict__ blob,
+                          const DevPage *__restrict__ pages, int npages,
+                          int64_t *__restrict__ out,
+                          uint8_t *__restrict__ valid,
+                          unsigned *__restrict__ err) {
+    __shared__ uint32_t scnt[256];
+    __shared__ uint64_t ssum[256];
+    __shared__ uint64_t carry[2]; /* [0]=value offset, [1]=delta prefix */
+    const uint8_t S8B_COUNT[16] = {240, 120, 60, 30, 20, 15, 12, 10,
+                                   8, 7, 6, 5, 4, 3, 2, 1};
+    const uint8_t S8B_WIDTH[16] = {0, 0, 1, 2, 3, 4, 5, 6,
+                                   7, 8, 10, 12, 15, 20, 30, 60};
+    for (int p = blockIdx.x; p < npages; p += gridDim.x) {
+        DevPage pg = pages[p];
+        const uint8_t *s = blob + pg.data_off + 1; /* sub-tag byte */
+        bool is_ts = pg.enc == GS_ENC_DELTATS;
+        uint64_t scaler = 1;
+        if (is_ts) {
+            unsigned s10 = s[0] & 0x0f;
+            for (unsigned k = 0; k < s10; k++) scaler *= 10;
+        }
+        const uint8_t *q = s + 1;
+        uint32_t plen = pg.data_len - 2;
+        if (plen < 8) { if (threadIdx.x == 0) atomicOr(err, DERR_SHORT); continue; }
+        uint64_t first_raw = dev_be64(q);
+        int64_t first = is_ts ? int64_t(first_raw) : dev_zzdec(first_raw);
+        const uint8_t *words = q + 8;
+        uint32_t nwords = (plen - 8) / 8;
+        int64_t *o = out + pg.row_off;
+        uint8_t *vd = valid ? valid + pg.row_off : nullptr;
+        uint32_t n = pg.nrows;
+        if (threadIdx.x == 0) {
+            if (n > 0) o[0] = first;
+            carry[0] = 1; /* first value occupies row 0 */
+            carry[1] = 0;
+        }
+        if (vd)
+            for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) vd[r] = 1;
+        __syncthreads();
+        for (uint32_t tile = 0; tile < nwords; tile += blockDim.x) {
+            uint32_t wi = tile + threadIdx.x;
+            uint64_t w = 0;
+            uint32_t cnt = 0;
+            uint64_t dsum = 0;
+            unsigned sel = 0, bits = 0;
+            uint64_t mask = 0;
+            if (wi < nwords) {
+                w = dev_be64(words + 8ull * wi);
+                sel = unsigned(w >> 60);
+                cnt = S8B_COUNT[sel];
+                bits = S8B_WIDTH[sel];
+                if (sel <= 1) {
+                    /* runs of "1": delta sum = cnt (x scaler for ts,
+                       zigzag-decoded 1 -> -1 for i64) */
+                    dsum = is_ts ? uint64_t(cnt) * scaler
+                                 : uint64_t(int64_t(-1)) * cnt;
+                } else {
+                    mask = (bits == 60) ? 0x0fffffffffffffffULL
+                                        : ((1ULL << bits) - 1);
+                    uint64_t v = w;
+                    for (uint32_t k = 0; k < cnt; k++) {
+                        uint64_t d = v & mask;
+                        dsum += is_ts ? d * scaler : uint64_t(dev_zzdec(d));
+                        v >>= bits;
+                    }
+                }
+            }
+            /* block exclusive scan of (cnt, dsum) */
+            scnt[threadIdx.x] = cnt;
+            ssum[threadIdx.x] = dsum;
+            __syncthreads();
+            for (int off = 1; off < int(blockDim.x); off <<= 1) {
+                uint32_t c2 = threadIdx.x >= unsigned(off)
+                                  ? scnt[threadIdx.x - off] : 0;
+                uint64_t s2 = threadIdx.x >= unsigned(off)
+                                  ? ssum[threadIdx.x - off] : 0;
+                __syncthreads();
+                scnt[threadIdx.x] += c2;
+                ssum[threadIdx.x] += s2;
+                __syncthreads();
+            }
+            uint64_t voff = carry[0] +
+                            (threadIdx.x ? scnt[threadIdx.x - 1] : 0);
+            uint64_t dpre = carry[1] +
+                            (threadIdx.x ? ssum[threadIdx.x - 1] : 0);
+            if (wi < nwords && cnt) {
+                uint64_t acc = uint64_t(first) + dpre;
+                if (sel <= 1) {
+                    uint64_t step = is_ts ? scaler : uint64_t(int64_t(-1));
+                    for (uint32_t k = 0; k < cnt && voff + k < n; k++) {
+                        acc += step;
+                        o[voff + k] = int64_t(acc);
+                    }
+                } else {
+                    uint64_t v = w;
+                    for (uint32_t k = 0; k < cnt; k++) {
+                        uint64_t d = v & mask;
+                        acc += is_ts ? d * scaler : uint64_t(dev_zzdec(d));
+                        if (voff + k < n) o[voff + k] = int64_t(acc);
+                        v >>= bits;
+                    }
+                }
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) {
+                carry[0] += scnt[blockDim.x - 1];
+                carry[1] += ssum[blockDim.x - 1];
+            }
+            __syncthreads();
+            if (carry[0] >= n) break; /* enough values for every row */
+        }
+        __syncthreads();
+        /* iterator exhaustion: rows past the decoded count are 0
+           (k_seq_i64 / Int64Builder skip semantics) */
+        uint64_t total = carry[0];
+        for (uint64_t r = total + threadIdx.x; r < n; r += blockDim.x)
+            o[r] = 0;
+        __syncthreads();
+    }
+}
+
 /* per-group closed-interval span on the sorted decoded ts
  * (TimeRange semantics, domain.rs:36-44) */
 __global__ void k_spans(const DevGroup *__restrict__ groups, int n,
@@ -1833,9 +1960,15 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
             gs_groups_free(set); return nullptr;
         }
         int cls = PC_SEQ;
-        if (hp.dp.all_valid && hp.dp.sub == 2 && hp.dp.data_len > 2) {
+        bool int_ct = hp.dp.ctype == GS_CT_TIME || hp.dp.ctype == GS_CT_I64 ||
+                      hp.dp.ctype == GS_CT_U64;
+        if (hp.dp.all_valid && hp.dp.sub == 2 && int_ct && hp.dp.data_len > 2) {
             if (hp.dp.enc == GS_ENC_DELTATS) cls = PC_RLE_TS;
             else if (hp.dp.enc == GS_ENC_DELTA) cls = PC_RLE_I64;
+        } else if (hp.dp.all_valid && hp.dp.sub == 1 && int_ct &&
+                   (hp.dp.enc == GS_ENC_DELTATS || hp.dp.enc == GS_ENC_DELTA) &&
+                   hp.dp.data_len > 2) {
+            cls = PC_S8B;
         } else if (hp.dp.all_valid && hp.dp.enc == GS_ENC_GORILLA &&
                    hp.dp.ctype == GS_CT_F64 && hp.dp.data_len >= 10) {
             cls = PC_GOR;
@@ -2017,6 +2150,12 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
                            dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
                            sp.dev[PC_GOR], n, (double *)d_out, d_valid,
                            ctx->d_err);
+    }
+    if (sp.n[PC_S8B]) {
+        int n = sp.n[PC_S8B];
+        hipLaunchKernelGGL(k_s8b_par, dim3(n > 65535 ? 65535 : n), dim3(256),
+                           0, ctx->stream, set->d_blob, sp.dev[PC_S8B], n,
+                           (int64_t *)d_out, d_valid, ctx->d_err);
     }
     if (sp.n[PC_RLE_TS])
         hipLaunchKernelGGL(k_rle_par,

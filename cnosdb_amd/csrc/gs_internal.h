@@ -32,7 +32,8 @@ enum PageClass {
     PC_RLE_TS = 1,  /* DeltaTs + RLE + all-valid: closed-form parallel */
     PC_RLE_I64 = 2, /* Delta + RLE + all-valid: closed-form parallel */
     PC_GOR = 3,     /* Gorilla + all-valid: LDS-staged cooperative stores */
-    PC_NCLASS = 4,
+    PC_S8B = 4,     /* DeltaTs/Delta + simple8b + all-valid: block-parallel */
+    PC_NCLASS = 5,
 };
 
 extern "C" {
