@@ -275,6 +275,35 @@ def bench_compact(args):
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     value = total * args.steps / dt
+
+    # CPU baseline: oracle merge_dedup (numpy restatement of the reference
+    # loser-tree + dedup) + host re-encode on a bounded sample
+    cpu_baseline = None
+    if not args.skip_cpu_baseline:
+        from oracle import pyoracle as orc
+        sample_series = min(16, nseries)
+        host_streams = []
+        for f in range(k):
+            ts = stream_ts[f]
+            host_streams.append([
+                (ts, np.round(np.clip(np.cumsum(rng.normal(0, 1, ts.size)) + 50, 0, 100), 2), None)
+                for _ in range(sample_series)])
+        tcb = time.perf_counter()
+        rows_cb = 0
+        for s_ in range(sample_series):
+            mts, mv, _ = orc.merge_dedup([host_streams[f][s_] for f in range(k)])
+            rows_cb += sum(host_streams[f][s_][0].size for f in range(k))
+            r = 0
+            while r < mts.size:
+                nblk = min(block_rows, mts.size - r)
+                gs.encode_ts(mts[r:r + nblk])
+                gs.encode_f64(mv[r:r + nblk])
+                r += nblk
+        dt_cb = time.perf_counter() - tcb
+        cpu_baseline = {"value": rows_cb / dt_cb, "unit": "rows/s",
+                        "cores": 1, "kind": "port",
+                        "sample": f"{sample_series} series x k=8 streams "
+                                  "(numpy merge_dedup + host re-encode, single core)"}
     print(json.dumps({
         "metric": "compaction merge rows/sec (config #5: k=8 overlapping L0 groups, dedup-by-ts, re-encode)",
         "value": value,
@@ -293,6 +322,7 @@ def bench_compact(args):
                    "rows_in": int(total), "rows_out": int(out_rows),
                    "encoded_out_bytes": enc_bytes,
                    "block_rows": block_rows, "setup_s": round(setup_s, 1)},
+        "cpu_baseline": cpu_baseline,
     }))
     for g in gsets:
         g.free()
