@@ -1058,17 +1058,36 @@ __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
     }
 }
 
-/* thread per (set, series): exclusive scan of owner flags */
+/* block per (set, series): exclusive block-scan of owner flags */
 __global__ void k_cm_prefix(CompactArgs a, int nsets, int nseries) {
-    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    __shared__ int32_t sh[256];
+    __shared__ int32_t carry;
     int total = nsets * nseries;
-    for (; idx < total; idx += gridDim.x * blockDim.x) {
+    for (int idx = blockIdx.x; idx < total; idx += gridDim.x) {
         int f = idx / nseries, s = idx % nseries;
         DevGroup g = a.groups[f][s];
         const uint8_t *fl = a.flags[f] + g.row_off;
         int32_t *pf = a.prefix[f] + g.row_off;
-        int32_t acc = 0;
-        for (int64_t j = 0; j < g.nrows; j++) { pf[j] = acc; acc += fl[j]; }
+        if (threadIdx.x == 0) carry = 0;
+        __syncthreads();
+        for (int64_t tile = 0; tile < g.nrows; tile += blockDim.x) {
+            int64_t j = tile + threadIdx.x;
+            int32_t v = (j < g.nrows) ? fl[j] : 0;
+            sh[threadIdx.x] = v;
+            __syncthreads();
+            for (int off = 1; off < int(blockDim.x); off <<= 1) {
+                int32_t u = threadIdx.x >= unsigned(off)
+                                ? sh[threadIdx.x - off] : 0;
+                __syncthreads();
+                sh[threadIdx.x] += u;
+                __syncthreads();
+            }
+            if (j < g.nrows)
+                pf[j] = carry + sh[threadIdx.x] - v; /* exclusive */
+            __syncthreads();
+            if (threadIdx.x == 0) carry += sh[blockDim.x - 1];
+            __syncthreads();
+        }
     }
 }
 
@@ -2561,7 +2580,8 @@ GsStatus gs_compact_merge(GsCtx *ctx, GsGroupSet *const *sets, int32_t nsets,
     hipLaunchKernelGGL(k_cm_flags, dim3(gx, nsets), dim3(256), 0, ctx->stream,
                        a, nsets, nseries, ctx->d_err);
     int total = nsets * nseries;
-    hipLaunchKernelGGL(k_cm_prefix, dim3(grid_for(total, 256)), dim3(256), 0,
+    hipLaunchKernelGGL(k_cm_prefix,
+                       dim3(total > 65535 ? 65535 : total), dim3(256), 0,
                        ctx->stream, a, nsets, nseries);
     std::vector<int64_t> counts(size_t(nsets) * nseries), ooff(nseries + 1);
     HIP_TRY(hipStreamSynchronize(ctx->stream));
