@@ -134,57 +134,6 @@ struct DevS8b {
     }
 };
 
-/* Gorilla bit reader: MSB-first over the BE byte stream, semantically
- * identical bit order to the reference's rotate-left cursor
- * (float.rs:418-606) but held as a 128-bit MSB-aligned window so a field
- * read is one compare + shift pair instead of the reference's split-read
- * masking — the hot-kernel form of the same contract. */
-struct DevBitReader {
-    const uint8_t *src;
-    uint32_t len, pos; /* byte cursor */
-    uint64_t hi, lo;   /* valid bits MSB-aligned across hi then lo */
-    int n;             /* number of valid bits in the window */
-    __device__ void init(const uint8_t *s, uint32_t l, uint32_t p) {
-        src = s; len = l; pos = p; hi = 0; lo = 0; n = 0;
-    }
-    __device__ __forceinline__ void insert_word() {
-        /* pull the next <=8 bytes into the window below the n valid bits
-           (call only with n <= 64) */
-        uint32_t rem = len - pos;
-        uint64_t x;
-        int bits;
-        if (__builtin_expect(rem >= 8, 1)) {
-            x = dev_be64(src + pos);
-            pos += 8;
-            bits = 64;
-        } else if (rem > 0) {
-            x = 0;
-            for (uint32_t k = pos; k < len; k++) x = (x << 8) | src[k];
-            x <<= (8 - rem) * 8;
-            bits = int(rem) * 8;
-            pos = len;
-        } else {
-            return;
-        }
-        if (n == 0) { hi = x; lo = 0; }
-        else { hi |= x >> n; lo = x << (64 - n); }
-        n += bits;
-    }
-    /* read cnt (1..=64) MSB-first bits into the LOW bits of *out.
-       false = stream exhausted (reference: "unexpected end of block") */
-    __device__ __forceinline__ bool read(unsigned cnt, uint64_t *out) {
-        while (n < int(cnt)) {
-            if (pos >= len) return false;
-            insert_word();
-        }
-        uint64_t bits = (cnt == 64) ? hi : (hi >> (64 - cnt));
-        if (cnt == 64) { hi = lo; lo = 0; }
-        else { hi = (hi << cnt) | (lo >> (64 - cnt)); lo <<= cnt; }
-        n -= int(cnt);
-        *out = bits;
-        return true;
-    }
-};
 
 #define GORILLA_SENTINEL 0x7ff8000000000ffULL
 
