@@ -411,8 +411,6 @@ __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
             fd[lane][0] = (uint64_t)(uintptr_t)(o + (r - rfill));
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
-#pragma unroll 8 /* keep several LDS reads in flight: the serial
-                    read->store chain was the kernel's main wait */
             for (int sl = 0; sl < 64; sl++) {
                 uint64_t ob = fd[sl][0];
                 int cnt = int(fd[sl][1]);
@@ -1309,8 +1307,6 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
             fd[lane][0] = (uint64_t)(uintptr_t)(o + run0);
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
-#pragma unroll 8 /* keep several LDS reads in flight: the serial
-                    read->store chain was the kernel's main wait */
             for (int sl = 0; sl < 64; sl++) {
                 uint64_t ob = fd[sl][0];
                 int cnt = int(fd[sl][1]);
