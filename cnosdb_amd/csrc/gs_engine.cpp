@@ -728,6 +728,28 @@ __global__ void k_s8b_par(const uint8_t *__restrict__ blob,
     }
 }
 
+/* Null-encoded (raw BE) all-valid pages: one workgroup per page, fully
+ * coalesced byteswap copy (ts_without_compress_decode_to_array,
+ * timestamp.rs:301-323; uncompressed ts carries NO deltas under Null).
+ * Works for i64/u64/f64 alike (bit pattern passthrough). */
+__global__ void k_raw_par(const uint8_t *__restrict__ blob,
+                          const DevPage *__restrict__ pages, int npages,
+                          int64_t *__restrict__ out,
+                          uint8_t *__restrict__ valid) {
+    for (int p = blockIdx.x; p < npages; p += gridDim.x) {
+        DevPage pg = pages[p];
+        const uint8_t *q = blob + pg.data_off + 1;
+        uint32_t avail = (pg.data_len - 1) / 8;
+        int64_t *o = out + pg.row_off;
+        uint8_t *vd = valid ? valid + pg.row_off : nullptr;
+        uint32_t n = pg.nrows;
+        for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
+            o[r] = (r < avail) ? int64_t(dev_be64(q + 8ull * r)) : 0;
+            if (vd) vd[r] = 1;
+        }
+    }
+}
+
 /* per-group closed-interval span on the sorted decoded ts
  * (TimeRange semantics, domain.rs:36-44) */
 __global__ void k_spans(const DevGroup *__restrict__ groups, int n,
@@ -1940,6 +1962,9 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         } else if (hp.dp.all_valid && hp.dp.enc == GS_ENC_GORILLA &&
                    hp.dp.ctype == GS_CT_F64 && hp.dp.data_len >= 10) {
             cls = PC_GOR;
+        } else if (hp.dp.all_valid && hp.dp.enc == GS_ENC_NULL &&
+                   hp.dp.ctype != GS_CT_BOOL && hp.dp.data_len >= 1) {
+            cls = PC_RAW;
         }
         sp.host[cls].push_back(hp.dp);
     }
@@ -2124,6 +2149,12 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
         hipLaunchKernelGGL(k_s8b_par, dim3(n > 65535 ? 65535 : n), dim3(256),
                            0, ctx->stream, set->d_blob, sp.dev[PC_S8B], n,
                            (int64_t *)d_out, d_valid, ctx->d_err);
+    }
+    if (sp.n[PC_RAW]) {
+        int n = sp.n[PC_RAW];
+        hipLaunchKernelGGL(k_raw_par, dim3(n > 65535 ? 65535 : n), dim3(256),
+                           0, ctx->stream, set->d_blob, sp.dev[PC_RAW], n,
+                           (int64_t *)d_out, d_valid);
     }
     if (sp.n[PC_RLE_TS])
         hipLaunchKernelGGL(k_rle_par,

@@ -33,7 +33,8 @@ enum PageClass {
     PC_RLE_I64 = 2, /* Delta + RLE + all-valid: closed-form parallel */
     PC_GOR = 3,     /* Gorilla + all-valid: LDS-staged cooperative stores */
     PC_S8B = 4,     /* DeltaTs/Delta + simple8b + all-valid: block-parallel */
-    PC_NCLASS = 5,
+    PC_RAW = 5,     /* Null encoding (raw BE) + all-valid: coalesced bswap */
+    PC_NCLASS = 6,
 };
 
 extern "C" {

@@ -395,6 +395,32 @@ def test_scan_fused_agg_equals_general(engine):
     gset.free()
 
 
+def test_decode_null_encoding_raw_pages(engine):
+    """Encoding::Null pages (CODEC(NULL) columns): raw BE values, no
+    compression (timestamp.rs:301-323, float.rs:387-413)."""
+    cases = []
+    for _ in range(5):
+        n = int(rng.integers(1, 3000))
+        cases.append(rng.integers(-2**60, 2**60, n).astype(np.int64))
+    groups = []
+    for i, vals in enumerate(cases):
+        ts = np.arange(len(vals), dtype=np.int64) * 1000
+        data = bytes([1]) + vals.astype(">i8").tobytes()  # ENC_NULL + BE
+        fpage = gs.build_page(data, len(vals))
+        groups.append((i, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (fpage, gs.CT_I64)]))
+    gset = engine.upload(groups)
+    out = torch.zeros(gset.rows, dtype=torch.int64, device="cuda")
+    engine.decode(gset, 1, out)
+    offs = gset.row_offsets()
+    host = out.cpu().numpy()
+    for i, vals in enumerate(cases):
+        got = host[offs[i]:offs[i] + len(vals)]
+        exp = orc.decode_i64(bytes([1]) + vals.astype(">i8").tobytes(), len(vals))
+        assert (got == exp).all() and (got == vals).all(), f"null-enc {i}"
+    gset.free()
+
+
 def test_scan_multi_field(engine):
     """Multi-metric scan (TSBS cpu-max-all style): one set with several f64
     field pages, scanned per field via spec.field_col."""
