@@ -411,11 +411,21 @@ __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
             fd[lane][0] = (uint64_t)(uintptr_t)(o + (r - rfill));
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
-            for (int sl = 0; sl < 64; sl++) {
-                uint64_t ob = fd[sl][0];
-                int cnt = int(fd[sl][1]);
-                if (lane < cnt)
-                    ((double *)(uintptr_t)ob)[lane] = rslot[lane][sl];
+            /* register-batched: issue 8 independent LDS reads, then their
+               8 stores — the serial read->store chain was the kernel's
+               dominant wait */
+            for (int base = 0; base < 64; base += 8) {
+                double vbuf[8];
+                uint64_t ob[8];
+                int cnt[8];
+                for (int j = 0; j < 8; j++) {
+                    vbuf[j] = rslot[lane][base + j];
+                    ob[j] = fd[base + j][0];
+                    cnt[j] = int(fd[base + j][1]);
+                }
+                for (int j = 0; j < 8; j++)
+                    if (lane < cnt[j])
+                        ((double *)(uintptr_t)ob[j])[lane] = vbuf[j];
             }
             rfill = 0;
         };
@@ -1329,11 +1339,21 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
             fd[lane][0] = (uint64_t)(uintptr_t)(o + run0);
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
-            for (int sl = 0; sl < 64; sl++) {
-                uint64_t ob = fd[sl][0];
-                int cnt = int(fd[sl][1]);
-                if (lane < cnt)
-                    ((double *)(uintptr_t)ob)[lane] = rslot[lane][sl];
+            /* register-batched: issue 8 independent LDS reads, then their
+               8 stores — the serial read->store chain was the kernel's
+               dominant wait */
+            for (int base = 0; base < 64; base += 8) {
+                double vbuf[8];
+                uint64_t ob[8];
+                int cnt[8];
+                for (int j = 0; j < 8; j++) {
+                    vbuf[j] = rslot[lane][base + j];
+                    ob[j] = fd[base + j][0];
+                    cnt[j] = int(fd[base + j][1]);
+                }
+                for (int j = 0; j < 8; j++)
+                    if (lane < cnt[j])
+                        ((double *)(uintptr_t)ob[j])[lane] = vbuf[j];
             }
             rfill = 0;
         };
