@@ -804,6 +804,96 @@ __global__ void k_tombstone(const DevGroup *__restrict__ groups, int n,
     }
 }
 
+/* value-predicate mask over the selected span (DataFilter semantics,
+ * reader/filter.rs:91-142): mask = validity AND pred(value); per-group
+ * selected count by block reduction */
+__device__ __forceinline__ bool dev_pred(int op, double a, double b,
+                                         double x) {
+    switch (op) {
+    case GS_PRED_GT: return x > a;
+    case GS_PRED_GE: return x >= a;
+    case GS_PRED_LT: return x < a;
+    case GS_PRED_LE: return x <= a;
+    case GS_PRED_EQ: return x == a;
+    case GS_PRED_NE: return x != a;
+    case GS_PRED_BETWEEN: return x >= a && x <= b;
+    default: return true;
+    }
+}
+
+__global__ void k_vmask(const DevGroup *__restrict__ groups, int n,
+                        const double *__restrict__ val,
+                        const uint8_t *__restrict__ valid,
+                        const int64_t *__restrict__ sp_start,
+                        const int64_t *__restrict__ sp_cnt, int op, double a,
+                        double b, uint8_t *__restrict__ mask,
+                        int64_t *__restrict__ sel_cnt) {
+    __shared__ long long sred[256];
+    for (int g = blockIdx.x; g < n; g += gridDim.x) {
+        int64_t base = groups[g].row_off + sp_start[g];
+        int64_t cnt = sp_cnt[g];
+        long long c = 0;
+        for (int64_t r = threadIdx.x; r < cnt; r += blockDim.x) {
+            bool ok = (!valid || valid[base + r]) &&
+                      dev_pred(op, a, b, val[base + r]);
+            mask[base + r] = ok;
+            c += ok;
+        }
+        sred[threadIdx.x] = c;
+        __syncthreads();
+        for (int w = blockDim.x >> 1; w > 0; w >>= 1) {
+            if (threadIdx.x < unsigned(w)) sred[threadIdx.x] += sred[threadIdx.x + w];
+            __syncthreads();
+        }
+        if (threadIdx.x == 0) sel_cnt[g] = sred[0];
+        __syncthreads();
+    }
+}
+
+/* masked compaction: block-scan of the mask within the span, gather the
+ * selected rows (filter_record_batch's mask-then-gather,
+ * reader/filter.rs:130-142) */
+__global__ void k_compact_masked(const DevGroup *__restrict__ groups, int n,
+                                 const int64_t *__restrict__ ts,
+                                 const double *__restrict__ val,
+                                 const uint8_t *__restrict__ mask,
+                                 const int64_t *__restrict__ sp_start,
+                                 const int64_t *__restrict__ sp_cnt,
+                                 const int64_t *__restrict__ out_off,
+                                 int64_t *__restrict__ out_ts,
+                                 double *__restrict__ out_val) {
+    __shared__ int32_t sh[256];
+    __shared__ int64_t carry;
+    for (int g = blockIdx.x; g < n; g += gridDim.x) {
+        int64_t base = groups[g].row_off + sp_start[g];
+        int64_t cnt = sp_cnt[g];
+        int64_t dst = out_off[g];
+        if (threadIdx.x == 0) carry = 0;
+        __syncthreads();
+        for (int64_t tile = 0; tile < cnt; tile += blockDim.x) {
+            int64_t r = tile + threadIdx.x;
+            int32_t m = (r < cnt) ? mask[base + r] : 0;
+            sh[threadIdx.x] = m;
+            __syncthreads();
+            for (int off = 1; off < int(blockDim.x); off <<= 1) {
+                int32_t u = threadIdx.x >= unsigned(off)
+                                ? sh[threadIdx.x - off] : 0;
+                __syncthreads();
+                sh[threadIdx.x] += u;
+                __syncthreads();
+            }
+            if (m) {
+                int64_t pos = dst + carry + sh[threadIdx.x] - 1;
+                out_ts[pos] = ts[base + r];
+                out_val[pos] = val[base + r];
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) carry += sh[blockDim.x - 1];
+            __syncthreads();
+        }
+    }
+}
+
 /* compacting copy of the selected span (filter_record_batch semantics,
  * reader/filter.rs:130-142: row filter on time; field nulls travel) */
 __global__ void k_compact(const DevGroup *__restrict__ groups, int n,
@@ -1746,6 +1836,8 @@ struct GsGroupSet {
     GsTimeRange *d_ranges = nullptr;
     size_t ranges_cap = 0;
     uint8_t *d_valid = nullptr; /* lazily allocated internal validity bytes */
+    uint8_t *d_mask = nullptr;  /* value-predicate row mask (lazy) */
+    int64_t *d_sel_cnt = nullptr; /* per-group masked selected counts (lazy) */
     bool any_nulls_field = false;
     /* agg partials: ngroups x nbuckets cells, cached across scans */
     double *d_pmax = nullptr;
@@ -2087,6 +2179,8 @@ void gs_groups_free(GsGroupSet *set) {
     hipFree(set->d_out_off);
     if (set->d_ranges) hipFree(set->d_ranges);
     if (set->d_valid) hipFree(set->d_valid);
+    if (set->d_mask) hipFree(set->d_mask);
+    if (set->d_sel_cnt) hipFree(set->d_sel_cnt);
     if (set->d_pmax) hipFree(set->d_pmax);
     if (set->d_psum) hipFree(set->d_psum);
     if (set->d_pcnt) hipFree(set->d_pcnt);
@@ -2216,6 +2310,7 @@ GsStatus gs_apply_tombstone(GsCtx *ctx, GsGroupSet *set, const int64_t *d_ts,
 static bool fused_capable(GsGroupSet *set, const GsScanSpec *spec) {
     uint32_t fc = 1 + uint32_t(spec->field_col);
     return spec->d_out_ts && spec->d_out_val && spec->n_tombstones == 0 &&
+           spec->value_pred.op == GS_PRED_NONE &&
            !set->any_nulls_field && fc < set->ncols &&
            set->slots[0].n[PC_RLE_TS] == int(set->ngroups) &&
            set->slots[fc].n[PC_GOR] == int(set->ngroups);
@@ -2401,25 +2496,56 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                        ctx->stream, set->d_groups, ng, spec->d_ts,
                        spec->range.min_ts, spec->range.max_ts, set->d_sp_start,
                        set->d_sp_cnt);
+
+    /* optional value predicate (DataFilter, reader/filter.rs:91-142) */
+    uint8_t *d_mask = nullptr;
+    const int64_t *d_counts = set->d_sp_cnt;
+    if (spec->value_pred.op != GS_PRED_NONE) {
+        if (!set->d_mask)
+            HIP_TRY(hipMalloc(&set->d_mask, size_t(set->total_rows)));
+        if (!set->d_sel_cnt)
+            HIP_TRY(hipMalloc(&set->d_sel_cnt,
+                              set->ngroups * sizeof(int64_t)));
+        d_mask = set->d_mask;
+        hipLaunchKernelGGL(k_vmask, dim3(ng > 65535 ? 65535 : ng), dim3(256),
+                           0, ctx->stream, set->d_groups, ng, spec->d_val,
+                           d_valid, set->d_sp_start, set->d_sp_cnt,
+                           spec->value_pred.op, spec->value_pred.a,
+                           spec->value_pred.b, d_mask, set->d_sel_cnt);
+        d_counts = set->d_sel_cnt;
+    }
     HIP_TRY(hipEventRecord(ev[3], ctx->stream));
 
     int64_t out_rows = 0;
-    if (spec->d_out_ts && spec->d_out_val) {
+    bool want_rows = spec->d_out_ts && spec->d_out_val;
+    if (want_rows || d_mask) {
         std::vector<int64_t> cnt(set->ngroups), off(set->ngroups);
         HIP_TRY(hipStreamSynchronize(ctx->stream));
-        HIP_TRY(hipMemcpy(cnt.data(), set->d_sp_cnt,
+        HIP_TRY(hipMemcpy(cnt.data(), d_counts,
                           set->ngroups * sizeof(int64_t),
                           hipMemcpyDeviceToHost));
         int64_t acc = 0;
         for (size_t g = 0; g < set->ngroups; g++) { off[g] = acc; acc += cnt[g]; }
         out_rows = acc;
-        HIP_TRY(hipMemcpyAsync(set->d_out_off, off.data(),
-                               set->ngroups * sizeof(int64_t),
-                               hipMemcpyHostToDevice, ctx->stream));
-        hipLaunchKernelGGL(k_compact, dim3(ng > 2048 ? 2048 : ng), dim3(256), 0,
-                           ctx->stream, set->d_groups, ng, spec->d_ts,
-                           spec->d_val, set->d_sp_start, set->d_sp_cnt,
-                           set->d_out_off, spec->d_out_ts, spec->d_out_val);
+        if (want_rows) {
+            HIP_TRY(hipMemcpyAsync(set->d_out_off, off.data(),
+                                   set->ngroups * sizeof(int64_t),
+                                   hipMemcpyHostToDevice, ctx->stream));
+            if (d_mask)
+                hipLaunchKernelGGL(k_compact_masked,
+                                   dim3(ng > 2048 ? 2048 : ng), dim3(256), 0,
+                                   ctx->stream, set->d_groups, ng, spec->d_ts,
+                                   spec->d_val, d_mask, set->d_sp_start,
+                                   set->d_sp_cnt, set->d_out_off,
+                                   spec->d_out_ts, spec->d_out_val);
+            else
+                hipLaunchKernelGGL(k_compact, dim3(ng > 2048 ? 2048 : ng),
+                                   dim3(256), 0, ctx->stream, set->d_groups,
+                                   ng, spec->d_ts, spec->d_val,
+                                   set->d_sp_start, set->d_sp_cnt,
+                                   set->d_out_off, spec->d_out_ts,
+                                   spec->d_out_val);
+        }
     }
     HIP_TRY(hipEventRecord(ev[4], ctx->stream));
 
@@ -2442,7 +2568,8 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                              ? (size_t(spec->n_buckets) + 1) * 4 : 0;
         hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
                            dim3(256), agg_shm, ctx->stream, set->d_sgroups, nsg,
-                           spec->d_ts, spec->d_val, d_valid,
+                           spec->d_ts, spec->d_val,
+                           d_mask ? d_mask : d_valid,
                            spec->range.min_ts, spec->range.max_ts,
                            spec->t0, spec->bucket_ns,
                            spec->n_buckets, set->d_pmax, set->d_psum,

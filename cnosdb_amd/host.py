@@ -42,6 +42,15 @@ class GsColumnGroupDesc(ctypes.Structure):
     ]
 
 
+PRED_OPS = {"gt": 1, "ge": 2, "lt": 3, "le": 4, "eq": 5, "ne": 6,
+            "between": 7}
+
+
+class GsValuePred(ctypes.Structure):
+    _fields_ = [("op", ctypes.c_int32), ("a", ctypes.c_double),
+                ("b", ctypes.c_double)]
+
+
 class GsScanSpec(ctypes.Structure):
     _fields_ = [
         ("range", GsTimeRange),
@@ -58,6 +67,7 @@ class GsScanSpec(ctypes.Structure):
         ("d_out_val", ctypes.c_void_p),
         ("d_ts", ctypes.c_void_p),
         ("d_val", ctypes.c_void_p),
+        ("value_pred", GsValuePred),
     ]
 
 
@@ -465,11 +475,17 @@ class Engine:
         return res
 
     def scan(self, gset, d_ts, d_val, time_range=None, tombstones=None,
-             d_out_ts=None, d_out_val=None, agg=None, field_col=0):
+             d_out_ts=None, d_out_val=None, agg=None, field_col=0,
+             value_pred=None):
         """Fused scan. agg: dict(bucket_ns, t0, n_buckets, d_max, d_sum,
-        d_count).  Returns GsScanResult."""
+        d_count).  value_pred: (op, a) or ("between", a, b) evaluated like
+        DataFilter's pushed expr.  Returns GsScanResult."""
         spec = GsScanSpec()
         spec.field_col = field_col
+        if value_pred:
+            spec.value_pred.op = PRED_OPS[value_pred[0]]
+            spec.value_pred.a = float(value_pred[1])
+            spec.value_pred.b = float(value_pred[2]) if len(value_pred) > 2 else 0.0
         lo, hi = time_range if time_range else (-(2**63), 2**63 - 1)
         spec.range = GsTimeRange(lo, hi)
         if tombstones:

@@ -151,6 +151,28 @@ GsStatus gs_apply_tombstone(GsCtx *ctx, GsGroupSet *set, const int64_t *d_ts,
  * downsampling aggregate that stock DataFusion would run above
  * (SURVEY.md §8a row "downsampling agg").
  */
+/* Value predicate on the scanned f64 field, evaluated like DataFilter's
+ * pushed PhysicalExpr (reader/filter.rs:91-142): rows kept iff the time
+ * range AND the predicate hold; a null field value fails the predicate
+ * (arrow comparison-with-null semantics), unlike the time-only filter
+ * where null field rows survive. */
+enum {
+    GS_PRED_NONE = 0,
+    GS_PRED_GT = 1,
+    GS_PRED_GE = 2,
+    GS_PRED_LT = 3,
+    GS_PRED_LE = 4,
+    GS_PRED_EQ = 5,
+    GS_PRED_NE = 6,
+    GS_PRED_BETWEEN = 7, /* closed [a, b] */
+};
+
+typedef struct {
+    int32_t op; /* GS_PRED_* */
+    double a;
+    double b;   /* BETWEEN upper bound */
+} GsValuePred;
+
 typedef struct {
     GsTimeRange range;     /* closed; use INT64_MIN/MAX for no filter */
     /* column slot of the f64 field to scan (0 = the first field page,
@@ -180,6 +202,8 @@ typedef struct {
        d_val: decoded f64 column (gs_set_rows rows) */
     int64_t *d_ts;
     double *d_val;
+    /* optional value predicate (general path; disables the fused path) */
+    GsValuePred value_pred;
 } GsScanSpec;
 
 typedef struct {

@@ -463,3 +463,46 @@ def test_crc_validation_rejects_corruption(engine):
     with pytest.raises(RuntimeError):
         engine.upload([(0, [(bytes(page), gs.CT_TIME),
                             (gs.page_of(np.zeros(100), gs.CT_F64), gs.CT_F64)])])
+
+
+def test_scan_value_predicate(engine):
+    """DataFilter value predicates (reader/filter.rs:91-142): rows kept iff
+    time range AND pred(value); null field values fail the predicate."""
+    r = np.random.default_rng(31)
+    t0 = 1_700_000_000_000_000_000
+    groups, truth = [], []
+    for s in range(10):
+        n = 4096
+        ts = t0 + np.arange(n, dtype=np.int64) * 10**9
+        vals = np.round(np.clip(np.cumsum(r.normal(0, 0.5, n)) + 50, 0, 100), 1)
+        valid = r.random(n) > 0.2
+        groups.append((s, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (gs.page_of(vals, gs.CT_F64, valid), gs.CT_F64)]))
+        truth.append((ts, vals, valid))
+    gset = engine.upload(groups)
+    rows = gset.rows
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    lo, hi = t0 + 500 * 10**9, t0 + 3500 * 10**9
+    for pred, npred in [(("gt", 50.0), lambda v: v > 50.0),
+                        (("le", 49.0), lambda v: v <= 49.0),
+                        (("between", 40.0, 60.0), lambda v: (v >= 40) & (v <= 60))]:
+        res = engine.scan(gset, d_ts, d_val, time_range=(lo, hi),
+                          d_out_ts=d_ots, d_out_val=d_oval, value_pred=pred)
+        exp_ts, exp_val = [], []
+        for ts, vals, valid in truth:
+            s0, c = orc.time_span(ts, lo, hi)
+            sel = np.zeros(ts.size, bool)
+            sel[s0:s0 + c] = True
+            sel &= valid & npred(np.where(valid, vals, np.nan))
+            exp_ts.append(ts[sel])
+            exp_val.append(vals[sel])
+        exp_ts = np.concatenate(exp_ts)
+        exp_val = np.concatenate(exp_val).astype(np.float64)
+        assert res.out_rows == exp_ts.size, pred
+        assert (d_ots[:res.out_rows].cpu().numpy() == exp_ts).all(), pred
+        got = d_oval[:res.out_rows].cpu().numpy()
+        assert got.view(np.uint64).tolist() == exp_val.view(np.uint64).tolist(), pred
+    gset.free()
