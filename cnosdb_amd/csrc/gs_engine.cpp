@@ -1388,11 +1388,16 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
     for (int rd = 0; rd < rounds; rd++) {
         int p0 = base_id + rd * stride;
         bool have = p0 < npages;
+        /* branch-free prologue: clamp to page 0 so every load issues
+           unconditionally and the compiler batches them under one wait
+           (the predicated version serialized ~5 dependent vmcnt(0)s per
+           page); !have lanes are done immediately and never use these */
         DevPage pg = pages[have ? p0 : 0];
         const uint8_t *s = blob + pg.data_off + 1;
-        int64_t sel_lo = have ? sp_start[pg.grp] : 0;
-        int64_t sel_hi = have ? sel_lo + sp_cnt[pg.grp] : 0;
-        double *o = have ? out + out_off[pg.grp] - sel_lo : out; /* o[r] valid for r in span */
+        int64_t sel_lo = sp_start[pg.grp];
+        int64_t sel_hi = sel_lo + sp_cnt[pg.grp];
+        if (!have) { sel_lo = 0; sel_hi = 0; }
+        double *o = out + out_off[pg.grp] - sel_lo; /* o[r] valid for r in span */
         uint32_t nrows = pg.nrows;
         uint32_t slen = pg.data_len - 1;
         uint64_t val = dev_be64(s + 1);
