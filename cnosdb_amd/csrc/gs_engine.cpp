@@ -760,6 +760,35 @@ __global__ void k_raw_par(const uint8_t *__restrict__ blob,
     }
 }
 
+/* BitPack bool pages, all-valid: parallel bit extract (boolean.rs:79-110;
+ * MSB-first bits after the varint count) */
+__global__ void k_bool_par(const uint8_t *__restrict__ blob,
+                           const DevPage *__restrict__ pages, int npages,
+                           uint8_t *__restrict__ out,
+                           uint8_t *__restrict__ valid,
+                           unsigned *__restrict__ err) {
+    for (int p = blockIdx.x; p < npages; p += gridDim.x) {
+        DevPage pg = pages[p];
+        const uint8_t *data = blob + pg.data_off;
+        uint64_t count;
+        uint32_t nr;
+        if (pg.data_len < 3 || data[1] != (1 << 4) ||
+            !dev_varint(data + 2, pg.data_len - 2, &count, &nr)) {
+            if (threadIdx.x == 0) atomicOr(err, DERR_FORMAT);
+            continue;
+        }
+        const uint8_t *bits = data + 2 + nr;
+        uint8_t *o = out + pg.row_off;
+        uint8_t *vd = valid ? valid + pg.row_off : nullptr;
+        uint32_t n = pg.nrows;
+        if (count < n && threadIdx.x == 0) atomicOr(err, DERR_SHORT);
+        for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
+            o[r] = (r < count) ? ((bits[r >> 3] >> (7 - (r & 7))) & 1) : 0;
+            if (vd) vd[r] = 1;
+        }
+    }
+}
+
 /* per-group closed-interval span on the sorted decoded ts
  * (TimeRange semantics, domain.rs:36-44) */
 __global__ void k_spans(const DevGroup *__restrict__ groups, int n,
@@ -2082,6 +2111,9 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         } else if (hp.dp.all_valid && hp.dp.enc == GS_ENC_NULL &&
                    hp.dp.ctype != GS_CT_BOOL && hp.dp.data_len >= 1) {
             cls = PC_RAW;
+        } else if (hp.dp.all_valid && hp.dp.enc == GS_ENC_BITPACK &&
+                   hp.dp.ctype == GS_CT_BOOL && hp.dp.data_len >= 3) {
+            cls = PC_BOOL;
         }
         sp.host[cls].push_back(hp.dp);
     }
@@ -2268,6 +2300,12 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
         hipLaunchKernelGGL(k_s8b_par, dim3(n > 65535 ? 65535 : n), dim3(256),
                            0, ctx->stream, set->d_blob, sp.dev[PC_S8B], n,
                            (int64_t *)d_out, d_valid, ctx->d_err);
+    }
+    if (sp.n[PC_BOOL]) {
+        int n = sp.n[PC_BOOL];
+        hipLaunchKernelGGL(k_bool_par, dim3(n > 65535 ? 65535 : n), dim3(256),
+                           0, ctx->stream, set->d_blob, sp.dev[PC_BOOL], n,
+                           (uint8_t *)d_out, d_valid, ctx->d_err);
     }
     if (sp.n[PC_RAW]) {
         int n = sp.n[PC_RAW];

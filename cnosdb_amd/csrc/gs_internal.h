@@ -34,7 +34,8 @@ enum PageClass {
     PC_GOR = 3,     /* Gorilla + all-valid: LDS-staged cooperative stores */
     PC_S8B = 4,     /* DeltaTs/Delta + simple8b + all-valid: block-parallel */
     PC_RAW = 5,     /* Null encoding (raw BE) + all-valid: coalesced bswap */
-    PC_NCLASS = 6,
+    PC_BOOL = 6,    /* BitPack + all-valid: parallel bit extract */
+    PC_NCLASS = 7,
 };
 
 extern "C" {
