@@ -157,6 +157,28 @@ def test_decode_bool(engine):
     gset.free()
 
 
+def test_decode_bool_with_nulls(engine):
+    """null-carrying bool pages take the sequential scatter path"""
+    cases, valids = [], []
+    for _ in range(6):
+        n = int(rng.integers(2, 2000))
+        cases.append(rng.integers(0, 2, n).astype(np.uint8))
+        valids.append(rng.random(n) > 0.3)
+    gset = _upload_single_col(engine, cases, gs.CT_BOOL, valids)
+    out, dv = _decode_col(engine, gset, gs.CT_BOOL, with_valid=True)
+    offs = gset.row_offsets()
+    host, hv = out.cpu().numpy(), dv.cpu().numpy()
+    for i, (vals, valid) in enumerate(zip(cases, valids)):
+        n = len(vals)
+        present = vals[valid]
+        data = gs.encode_bool(present) if present.size else b""
+        exp = orc.decode_bool(data, n, valid)
+        got = host[offs[i]:offs[i] + n]
+        assert (got == exp).all(), f"bool-null case {i}"
+        assert (hv[offs[i]:offs[i] + n] == valid.astype(np.uint8)).all()
+    gset.free()
+
+
 def test_decode_with_nulls_scatter(engine):
     cases, valids = [], []
     for _ in range(8):
