@@ -2778,4 +2778,74 @@ GsStatus gs_compact_merge(GsCtx *ctx, GsGroupSet *const *sets, int32_t nsets,
     return GS_OK;
 }
 
+/* ---- Arrow C Data Interface export ---- */
+namespace {
+struct ExportPriv {
+    void *data;
+    uint8_t *bitmap;
+    const void *bufs[2];
+};
+void export_release(GsArrowArray *a) {
+    if (!a || !a->private_data) return;
+    ExportPriv *p = (ExportPriv *)a->private_data;
+    free(p->data);
+    free(p->bitmap);
+    free(p);
+    a->release = nullptr;
+    a->private_data = nullptr;
+}
+} // namespace
+
+GsStatus gs_export_group_column(GsCtx *ctx, GsGroupSet *set, int64_t group,
+                                const void *d_col, int32_t elem_size,
+                                const uint8_t *d_valid, GsArrowArray *out) {
+    if (!ctx || !set || group < 0 || size_t(group) >= set->ngroups || !d_col ||
+        !out || (elem_size != 8 && elem_size != 1))
+        return fail(GS_ERR, "bad args to gs_export_group_column");
+    HIP_TRY(hipSetDevice(ctx->device));
+    int64_t row0 = set->row_offsets[group];
+    int64_t rows = (size_t(group) + 1 < set->ngroups
+                        ? set->row_offsets[group + 1]
+                        : set->total_rows) - row0;
+    ExportPriv *p = (ExportPriv *)calloc(1, sizeof(ExportPriv));
+    if (!p) return fail(GS_ERR, "oom");
+    p->data = malloc(size_t(rows) * elem_size ? size_t(rows) * elem_size : 1);
+    if (!p->data) { free(p); return fail(GS_ERR, "oom"); }
+    if (hipMemcpy(p->data, (const uint8_t *)d_col + row0 * elem_size,
+                  size_t(rows) * elem_size, hipMemcpyDeviceToHost) !=
+        hipSuccess) {
+        free(p->data); free(p);
+        return fail(GS_ERR, "export D2H failed");
+    }
+    int64_t null_count = 0;
+    if (d_valid) {
+        std::vector<uint8_t> vbytes(rows);
+        if (hipMemcpy(vbytes.data(), d_valid + row0, size_t(rows),
+                      hipMemcpyDeviceToHost) != hipSuccess) {
+            free(p->data); free(p);
+            return fail(GS_ERR, "export validity D2H failed");
+        }
+        size_t nb = size_t(rows + 7) / 8;
+        p->bitmap = (uint8_t *)calloc(nb ? nb : 1, 1);
+        if (!p->bitmap) { free(p->data); free(p); return fail(GS_ERR, "oom"); }
+        for (int64_t r = 0; r < rows; r++) {
+            if (vbytes[r]) p->bitmap[r >> 3] |= uint8_t(1u << (r & 7));
+            else null_count++;
+        }
+    }
+    p->bufs[0] = p->bitmap; /* NULL when no validity buffer */
+    p->bufs[1] = p->data;
+    out->length = rows;
+    out->null_count = d_valid ? null_count : 0;
+    out->offset = 0;
+    out->n_buffers = 2;
+    out->n_children = 0;
+    out->buffers = p->bufs;
+    out->children = nullptr;
+    out->dictionary = nullptr;
+    out->release = export_release;
+    out->private_data = p;
+    return GS_OK;
+}
+
 } // extern "C"

@@ -71,6 +71,21 @@ class GsScanSpec(ctypes.Structure):
     ]
 
 
+class GsArrowArray(ctypes.Structure):
+    pass
+
+
+GsArrowArray._fields_ = [
+    ("length", ctypes.c_int64), ("null_count", ctypes.c_int64),
+    ("offset", ctypes.c_int64), ("n_buffers", ctypes.c_int64),
+    ("n_children", ctypes.c_int64),
+    ("buffers", ctypes.POINTER(ctypes.c_void_p)),
+    ("children", ctypes.POINTER(ctypes.POINTER(GsArrowArray))),
+    ("dictionary", ctypes.POINTER(GsArrowArray)),
+    ("release", ctypes.c_void_p), ("private_data", ctypes.c_void_p),
+]
+
+
 class GsScanResult(ctypes.Structure):
     _fields_ = [
         ("out_rows", ctypes.c_int64),
@@ -151,6 +166,10 @@ class PageLib:
             ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p,
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p,
             ctypes.c_int64, ctypes.c_void_p]
+        lib.gs_export_group_column.restype = ctypes.c_int32
+        lib.gs_export_group_column.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p,
+            ctypes.c_int32, ctypes.c_void_p, ctypes.POINTER(GsArrowArray)]
         lib.gs_compact_merge.restype = ctypes.c_int32
         lib.gs_compact_merge.argtypes = [
             ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
@@ -389,6 +408,32 @@ class Engine:
                                          arr, len(ranges))
         if st != 0:
             raise RuntimeError(f"gs_apply_tombstone failed: {self._pl.err()}")
+
+    def export_group_column(self, gset, group, d_col, elem_size=8,
+                            d_valid=None):
+        """Export one group's rows as an Arrow-C-data-interface array
+        (host copy).  Returns (GsArrowArray, np data view, np bitmap or
+        None); call arr-release via free_arrow when done."""
+        arr = GsArrowArray()
+        st = self.lib.gs_export_group_column(
+            self._ctx, gset._h, group, ctypes.c_void_p(d_col.data_ptr()),
+            elem_size,
+            ctypes.c_void_p(d_valid.data_ptr()) if d_valid is not None else None,
+            ctypes.byref(arr))
+        if st != 0:
+            raise RuntimeError(f"gs_export_group_column failed: {self._pl.err()}")
+        n = arr.length
+        data = np.ctypeslib.as_array(
+            ctypes.cast(arr.buffers[1], ctypes.POINTER(ctypes.c_uint8)),
+            shape=(n * elem_size,)).copy()
+        bitmap = None
+        if arr.buffers[0]:
+            bitmap = np.ctypeslib.as_array(
+                ctypes.cast(arr.buffers[0], ctypes.POINTER(ctypes.c_uint8)),
+                shape=((n + 7) // 8,)).copy()
+        rel = ctypes.CFUNCTYPE(None, ctypes.POINTER(GsArrowArray))(arr.release)
+        rel(ctypes.byref(arr))
+        return arr, data, bitmap
 
     def compact_merge(self, gsets, d_ts_list, d_val_list, d_valid_list,
                       d_out_ts, d_out_val, d_out_valid=None):

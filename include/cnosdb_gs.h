@@ -261,6 +261,34 @@ GsStatus gs_compact_merge(GsCtx *ctx, GsGroupSet *const *sets, int32_t nsets,
                           double *d_out_val, uint8_t *d_out_valid,
                           int64_t *h_out_offsets, int64_t *out_rows);
 
+/* ---- Arrow C Data Interface export (decode output as
+ * ArrowArray so the Rust shim reconstitutes RecordBatches zero-copy from
+ * the shim's side).  GsArrowArray is layout-identical to `struct
+ * ArrowArray` of the Arrow C data interface; for primitive columns
+ * buffers = {validity bitmap or NULL, data}. */
+typedef struct GsArrowArray {
+    int64_t length;
+    int64_t null_count;
+    int64_t offset;
+    int64_t n_buffers;
+    int64_t n_children;
+    const void **buffers;
+    struct GsArrowArray **children;
+    struct GsArrowArray *dictionary;
+    void (*release)(struct GsArrowArray *);
+    void *private_data;
+} GsArrowArray;
+
+/* Copy one group's rows of a decoded device column into a freshly
+ * allocated host ArrowArray (validity bytes packed LSB-first into the
+ * arrow bitmap; d_valid NULL => no validity buffer, null_count 0).
+ * elem_size: 8 for i64/f64/u64, 1 for bool (exported as byte values —
+ * the shim repacks to arrow bool bits if needed).  Caller releases via
+ * out->release(out). */
+GsStatus gs_export_group_column(GsCtx *ctx, GsGroupSet *set, int64_t group,
+                                const void *d_col, int32_t elem_size,
+                                const uint8_t *d_valid, GsArrowArray *out);
+
 #ifdef __cplusplus
 }
 #endif

@@ -528,3 +528,25 @@ def test_scan_value_predicate(engine):
         got = d_oval[:res.out_rows].cpu().numpy()
         assert got.view(np.uint64).tolist() == exp_val.view(np.uint64).tolist(), pred
     gset.free()
+
+
+def test_arrow_export(engine):
+    """Arrow C data interface export: group rows + packed validity bitmap
+    (the SURVEY §8b output form for the Rust shim)."""
+    cases, valids = [], []
+    for _ in range(4):
+        n = int(rng.integers(2, 1500))
+        cases.append(rng.integers(-2**40, 2**40, n).astype(np.int64))
+        valids.append(rng.random(n) > 0.3)
+    gset = _upload_single_col(engine, cases, gs.CT_I64, valids)
+    out, dv = _decode_col(engine, gset, gs.CT_I64, with_valid=True)
+    for g, (vals, valid) in enumerate(zip(cases, valids)):
+        arr, data, bitmap = engine.export_group_column(gset, g, out, 8, dv)
+        assert arr.length == len(vals)
+        assert arr.null_count == int((~valid).sum())
+        got = data.view(np.int64)
+        exp = np.where(valid, vals, 0)
+        assert (got == exp).all()
+        bits = np.unpackbits(bitmap, bitorder="little")[:len(vals)]
+        assert (bits.astype(bool) == valid).all()
+    gset.free()
