@@ -355,7 +355,7 @@ __device__ __forceinline__ void gorilla_decode_page(
  * Lanes run in lockstep (one value per iteration), so rings fill together;
  * lanes whose page ended keep cooperating in flushes until all are done. */
 #define GS_RING 16
-#define GS_GOR_BLOCK 256 /* 4 waves (2-wave/8-wave-SIMD variant measured equal) */
+#define GS_GOR_BLOCK 256 /* 4 waves; RING 8/16/32 x block 128/256 swept: 16/256 optimal */
 
 __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
                           const DevPage *__restrict__ pages, int npages,
