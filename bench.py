@@ -333,7 +333,9 @@ def bench_compact(args):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--mode", choices=["scan", "compact"], default="scan")
-    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--gpus", type=int, default=1,
+                    help="driver contract flag; the actual world size comes "
+                         "from the torchrun environment (WORLD_SIZE)")
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--series", type=int, default=10000)
