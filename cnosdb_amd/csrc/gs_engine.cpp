@@ -1004,12 +1004,7 @@ __global__ void k_agg_partial(const DevGroup *__restrict__ groups, int n,
             }
             __syncthreads();
         }
-        /* contiguous bucket chunk per wave: consecutive buckets' rows are
-           adjacent, so each wave streams one contiguous region */
-        int per = (nbuckets + nwaves - 1) / nwaves;
-        int b_lo = wave * per;
-        int b_hi = b_lo + per < nbuckets ? b_lo + per : nbuckets;
-        for (int b = b_lo; b < b_hi; b++) {
+        for (int b = wave; b < nbuckets; b += nwaves) {
             int64_t s, e;
             if (use_lds) {
                 s = soff[b];
