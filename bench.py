@@ -363,12 +363,20 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     dist = None
-    device = torch.device(f"cuda:{local_rank}")
+    # GS_BENCH_BACKEND=gloo lets the multi-rank path be validated with
+    # several ranks sharing one GPU (NCCL requires a device per rank)
+    backend = os.environ.get("GS_BENCH_BACKEND", "nccl")
+    ndev = torch.cuda.device_count()
+    dev_idx = local_rank % max(1, ndev)
+    device = torch.device(f"cuda:{dev_idx}")
     torch.cuda.set_device(device)
     if world > 1:
         import torch.distributed as tdist
         dist = tdist
-        dist.init_process_group("nccl", device_id=device)
+        if backend == "nccl":
+            dist.init_process_group("nccl", device_id=device)
+        else:
+            dist.init_process_group(backend)
 
     nseries, npts, page_rows = args.series, args.npts, args.page_rows
     assert npts % page_rows == 0
@@ -385,7 +393,7 @@ def main():
     # two engine contexts = two HIP streams on the one device: sub-batches
     # alternate streams so the ALU-bound Gorilla decode of one overlaps the
     # HBM-bound aggregate/ts phases of the other
-    engines = [gs.Engine(local_rank), gs.Engine(local_rank)]
+    engines = [gs.Engine(dev_idx), gs.Engine(dev_idx)]
     sets, eng_of = [], []
     raw_f64_bytes_sb0 = None
     for sb, (buf, offs, lens, nvals, cts, sids) in enumerate(sub):
