@@ -1207,30 +1207,32 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
         for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
             if (!fl[j]) continue;
             int64_t x = t[j];
-            /* merged position: owners with smaller ts across all streams */
+            /* merged position (owners with smaller ts across all streams)
+               and, in the same pass, where x occurs in OLDER streams —
+               the dedup value walk reuses those hits instead of searching
+               again (batch_builder.rs:139-151 semantics: this element is
+               the owner, so no NEWER stream contains x; the newest
+               non-null among {f, older hits} wins, all-null -> null) */
             int64_t pos = base + pf[j];
+            int64_t hit[GS_MAX_STREAMS];
             for (int f2 = 0; f2 < nsets; f2++) {
-                if (f2 == f) continue;
+                if (f2 == f) { hit[f2] = g.row_off + j; continue; }
                 DevGroup g2 = a.groups[f2][s];
                 int64_t ins;
-                dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &ins);
-                /* owners in f2 below insertion point */
+                bool found =
+                    dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &ins);
+                hit[f2] = (found && f2 < f) ? g2.row_off + ins : -1;
                 pos += (ins > 0) ? a.prefix[f2][g2.row_off + ins - 1] +
                                        a.flags[f2][g2.row_off + ins - 1]
                                  : 0;
             }
-            /* value: newest stream containing x with a non-null value
-               (batch_builder.rs:139-151); all-null -> null, slot 0 */
             double v = 0.0;
             uint8_t ok = 0;
-            for (int f2 = nsets - 1; f2 >= 0; f2--) {
-                DevGroup g2 = a.groups[f2][s];
-                int64_t ins;
-                if (!dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &ins))
-                    continue;
+            for (int f2 = f; f2 >= 0; f2--) {
+                if (hit[f2] < 0) continue;
                 const uint8_t *vd2 = a.valid[f2];
-                if (!vd2 || vd2[g2.row_off + ins]) {
-                    v = a.val[f2][g2.row_off + ins];
+                if (!vd2 || vd2[hit[f2]]) {
+                    v = a.val[f2][hit[f2]];
                     ok = 1;
                     break;
                 }
