@@ -242,7 +242,7 @@ def bench_compact(args):
         total += gset.rows
     d_ots = torch.zeros(total, dtype=torch.int64, device="cuda")
     d_oval = torch.zeros(total, dtype=torch.float64, device="cuda")
-    block_rows = 1000  # re-encode block size: plenty of pages for the
+    block_rows = 500  # re-encode block size: plenty of pages for the
     # thread-per-page encoder (reference caps blocks by bytes, not rows)
     cap = block_rows * 12 + 128
     setup_s = time.perf_counter() - t_setup
