@@ -242,7 +242,7 @@ def bench_compact(args):
         total += gset.rows
     d_ots = torch.zeros(total, dtype=torch.int64, device="cuda")
     d_oval = torch.zeros(total, dtype=torch.float64, device="cuda")
-    block_rows = 500  # re-encode block size: plenty of pages for the
+    block_rows = 1000  # re-encode block size: plenty of pages for the
     # thread-per-page encoder (reference caps blocks by bytes, not rows)
     cap = block_rows * 12 + 128
     setup_s = time.perf_counter() - t_setup
@@ -250,14 +250,15 @@ def bench_compact(args):
     def step():
         out_rows, offs = eng.compact_merge(gsets, tss, vls, [None] * k,
                                            d_ots, d_oval)
-        row_off, rows_arr = [], []
-        for s in range(nseries):
-            r = offs[s]
-            while r < offs[s + 1]:
-                n = min(block_rows, offs[s + 1] - r)
-                row_off.append(r)
-                rows_arr.append(n)
-                r += n
+        # vectorized per-series block split (a python loop here cost more
+        # than the kernels at small block sizes)
+        counts = offs[1:] - offs[:-1]
+        npg_per = -(-counts // block_rows)
+        series_base = np.repeat(offs[:-1], npg_per)
+        series_cnt = np.repeat(counts, npg_per)
+        intra = np.concatenate([np.arange(n) for n in npg_per]) * block_rows
+        row_off = series_base + intra
+        rows_arr = np.minimum(block_rows, series_cnt - intra)
         npg = len(row_off)
         d_enc_ts = torch.zeros(npg * cap, dtype=torch.uint8, device="cuda")
         d_enc_v = torch.zeros(npg * cap, dtype=torch.uint8, device="cuda")
