@@ -342,7 +342,7 @@ def main():
     ap.add_argument("--series", type=int, default=10000)
     ap.add_argument("--npts", type=int, default=1_000_000)
     ap.add_argument("--page-rows", type=int, default=4000)
-    ap.add_argument("--sub-batches", type=int, default=4)
+    ap.add_argument("--sub-batches", type=int, default=2)
     ap.add_argument("--unique", type=int, default=256)
     ap.add_argument("--fields", type=int, default=1,
                     help="f64 field pages per group (8 = TSBS cpu-max-all-8, config #3)")
