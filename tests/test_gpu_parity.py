@@ -550,3 +550,41 @@ def test_arrow_export(engine):
         bits = np.unpackbits(bitmap, bitorder="little")[:len(vals)]
         assert (bits.astype(bool) == valid).all()
     gset.free()
+
+
+def test_scan_tombstones_plus_value_pred(engine):
+    """Tombstone masking composes with the value predicate: deleted rows
+    are nulls, and nulls fail the predicate."""
+    r = np.random.default_rng(41)
+    t0 = 1_700_000_000_000_000_000
+    groups, truth = [], []
+    for s in range(6):
+        n = 3000
+        ts = t0 + np.arange(n, dtype=np.int64) * 10**9
+        vals = np.round(np.clip(np.cumsum(r.normal(0, 0.5, n)) + 50, 0, 100), 1)
+        groups.append((s, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (gs.page_of(vals, gs.CT_F64), gs.CT_F64)]))
+        truth.append((ts, vals))
+    gset = engine.upload(groups)
+    rows = gset.rows
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    dead = [(t0 + 200 * 10**9, t0 + 400 * 10**9)]
+    lo, hi = t0 + 100 * 10**9, t0 + 2500 * 10**9
+    res = engine.scan(gset, d_ts, d_val, time_range=(lo, hi), tombstones=dead,
+                      d_out_ts=d_ots, d_out_val=d_oval, value_pred=("gt", 50.0))
+    exp_ts, exp_val = [], []
+    for ts, vals in truth:
+        valid = orc.update_nullbits(ts, dead, np.ones(ts.size, bool))
+        sel = (ts >= lo) & (ts <= hi) & valid & (vals > 50.0)
+        exp_ts.append(ts[sel])
+        exp_val.append(vals[sel])
+    exp_ts = np.concatenate(exp_ts)
+    exp_val = np.concatenate(exp_val).astype(np.float64)
+    assert res.out_rows == exp_ts.size
+    assert (d_ots[:res.out_rows].cpu().numpy() == exp_ts).all()
+    got = d_oval[:res.out_rows].cpu().numpy()
+    assert got.view(np.uint64).tolist() == exp_val.view(np.uint64).tolist()
+    gset.free()
