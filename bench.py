@@ -278,34 +278,67 @@ def bench_compact(args):
     dt = time.perf_counter() - t0
     value = total * args.steps / dt
 
-    # CPU baseline: oracle merge_dedup (numpy restatement of the reference
-    # loser-tree + dedup) + host re-encode on a bounded sample
+    # CPU baseline: oracle C k-way merge (OpenMP over series) + host
+    # re-encode on a bounded sample
     cpu_baseline = None
     if not args.skip_cpu_baseline:
+        import ctypes
         from oracle import pyoracle as orc
-        sample_series = min(16, nseries)
-        host_streams = []
-        for f in range(k):
-            ts = stream_ts[f]
-            host_streams.append([
-                (ts, np.round(np.clip(np.cumsum(rng.normal(0, 1, ts.size)) + 50, 0, 100), 2), None)
-                for _ in range(sample_series)])
-        tcb = time.perf_counter()
-        rows_cb = 0
+
+        class OrcStream(ctypes.Structure):
+            _fields_ = [("ts", ctypes.c_void_p), ("val", ctypes.c_void_p),
+                        ("valid", ctypes.c_void_p), ("n", ctypes.c_int64)]
+
+        o = orc.Oracle().lib
+        o.orc_merge_dedup_many.restype = ctypes.c_int32
+        sample_series = min(64, nseries)
+        host = []  # [series][k] -> (ts, vals)
+        total_in = 0
         for s_ in range(sample_series):
-            mts, mv, _ = orc.merge_dedup([host_streams[f][s_] for f in range(k)])
-            rows_cb += sum(host_streams[f][s_][0].size for f in range(k))
-            r = 0
-            while r < mts.size:
-                nblk = min(block_rows, mts.size - r)
-                gs.encode_ts(mts[r:r + nblk])
-                gs.encode_f64(mv[r:r + nblk])
+            per = []
+            for f in range(k):
+                ts = stream_ts[f]
+                vals = np.round(np.clip(
+                    np.cumsum(rng.normal(0, 1, ts.size)) + 50, 0, 100), 2)
+                per.append((ts, vals))
+                total_in += ts.size
+            host.append(per)
+        streams = (OrcStream * (sample_series * k))()
+        caps = np.zeros(sample_series + 1, dtype=np.int64)
+        for s_ in range(sample_series):
+            cap = 0
+            for f in range(k):
+                ts, vals = host[s_][f]
+                streams[s_ * k + f] = OrcStream(
+                    ts.ctypes.data, vals.ctypes.data, None, ts.size)
+                cap += ts.size
+            caps[s_ + 1] = caps[s_] + cap
+        out_ts_h = np.zeros(caps[-1], dtype=np.int64)
+        out_v_h = np.zeros(caps[-1], dtype=np.float64)
+        cnts = np.zeros(sample_series, dtype=np.int64)
+        cores = _host_threads()
+        tcb = time.perf_counter()
+        st_ = o.orc_merge_dedup_many(
+            streams, k, sample_series,
+            caps[:-1].ctypes.data_as(ctypes.c_void_p),
+            out_ts_h.ctypes.data_as(ctypes.c_void_p),
+            out_v_h.ctypes.data_as(ctypes.c_void_p),
+            cnts.ctypes.data_as(ctypes.c_void_p), cores)
+        assert st_ == 0
+        for s_ in range(sample_series):  # re-encode (serial C per call)
+            a, b = caps[s_], caps[s_] + cnts[s_]
+            r = a
+            while r < b:
+                nblk = min(block_rows, b - r)
+                gs.encode_ts(out_ts_h[r:r + nblk])
+                gs.encode_f64(out_v_h[r:r + nblk])
                 r += nblk
         dt_cb = time.perf_counter() - tcb
-        cpu_baseline = {"value": rows_cb / dt_cb, "unit": "rows/s",
-                        "cores": 1, "kind": "port",
-                        "sample": f"{sample_series} series x k=8 streams "
-                                  "(numpy merge_dedup + host re-encode, single core)"}
+        cpu_baseline = {"value": total_in / dt_cb, "unit": "rows/s",
+                        "cores": cores, "kind": "port",
+                        "sample": f"{sample_series} series x k={k} streams "
+                                  "(oracle C k-way merge, OpenMP; host "
+                                  "re-encode single-thread)"}
     print(json.dumps({
         "metric": "compaction merge rows/sec (config #5: k=8 overlapping L0 groups, dedup-by-ts, re-encode)",
         "value": value,

@@ -826,6 +826,78 @@ ORC_EXPORT int orc_decode_pages_omp(const OrcPageDesc *pages, size_t npages,
     return err;
 }
 
+/* ------------------------------------------------- k-way merge (baseline)
+ * CPU restatement of the compaction merge for the cpu_baseline leg of
+ * bench.py --mode compact: k time-sorted streams (oldest..newest, file_id
+ * order), equal-ts rows collapse, the newest non-null value wins
+ * (reader/sort_merge.rs:152-343 + reader/batch_builder.rs:106-155).
+ * Simple k-pointer merge per series, OpenMP over series. */
+typedef struct {
+    const int64_t *ts;
+    const double *val;
+    const uint8_t *valid; /* may be NULL */
+    int64_t n;
+} OrcStream;
+
+/* merge one series' k streams; returns merged row count */
+ORC_EXPORT int64_t orc_merge_dedup(const OrcStream *streams, int k,
+                                   int64_t *out_ts, double *out_val,
+                                   uint8_t *out_valid) {
+    int64_t pos[16];
+    if (k > 16) return -1;
+    for (int f = 0; f < k; f++) pos[f] = 0;
+    int64_t w = 0;
+    for (;;) {
+        int64_t mn = INT64_MAX;
+        int any = 0;
+        for (int f = 0; f < k; f++)
+            if (pos[f] < streams[f].n) {
+                any = 1;
+                if (streams[f].ts[pos[f]] < mn) mn = streams[f].ts[pos[f]];
+            }
+        if (!any) break;
+        double v = 0.0;
+        uint8_t ok = 0;
+        for (int f = k - 1; f >= 0; f--) { /* newest stream first */
+            if (pos[f] < streams[f].n && streams[f].ts[pos[f]] == mn) {
+                if (!ok && (!streams[f].valid || streams[f].valid[pos[f]])) {
+                    v = streams[f].val[pos[f]];
+                    ok = 1;
+                }
+                pos[f]++;
+            }
+        }
+        out_ts[w] = mn;
+        out_val[w] = ok ? v : 0.0;
+        if (out_valid) out_valid[w] = ok;
+        w++;
+    }
+    return w;
+}
+
+/* OpenMP over series: streams laid out [series][k]; outputs at
+ * out_offsets[series] (caller-provided, e.g. upper bounds) */
+ORC_EXPORT int orc_merge_dedup_many(const OrcStream *streams, int k,
+                                    int64_t nseries,
+                                    const int64_t *out_offsets,
+                                    int64_t *out_ts, double *out_val,
+                                    int64_t *out_counts, int nthreads) {
+    int err = 0;
+#pragma omp parallel for schedule(dynamic, 1) num_threads(nthreads)
+    for (int64_t s = 0; s < nseries; s++) {
+        int64_t w = orc_merge_dedup(streams + s * k, k,
+                                    out_ts + out_offsets[s],
+                                    out_val + out_offsets[s], NULL);
+        if (w < 0) {
+#pragma omp atomic write
+            err = -1;
+        } else {
+            out_counts[s] = w;
+        }
+    }
+    return err;
+}
+
 /* time-range filter count: closed interval [mn,mx] over sorted ts
  * (reader/filter.rs semantics for a pure time-range PhysicalExpr;
  * TimeRange is a closed interval, common/models/src/predicate/domain.rs:36-44) */
