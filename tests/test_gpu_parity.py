@@ -588,3 +588,19 @@ def test_scan_tombstones_plus_value_pred(engine):
     got = d_oval[:res.out_rows].cpu().numpy()
     assert got.view(np.uint64).tolist() == exp_val.view(np.uint64).tolist()
     gset.free()
+
+
+def test_truncated_gorilla_stream_errors(engine):
+    """A stream cut before its sentinel must fail decode (the reference's
+    "unexpected end of block", float.rs:462)."""
+    vals = np.cumsum(rng.normal(0, 1, 2000))
+    data = gs.encode_f64(vals)
+    cut = data[:len(data) // 2]
+    page = gs.build_page(cut, 2000)
+    ts = np.arange(2000, dtype=np.int64) * 1000
+    gset = engine.upload([(0, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                               (page, gs.CT_F64)])])
+    out = torch.zeros(gset.rows, dtype=torch.float64, device="cuda")
+    with pytest.raises(RuntimeError):
+        engine.decode(gset, 1, out)
+    gset.free()
