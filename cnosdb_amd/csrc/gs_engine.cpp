@@ -1087,6 +1087,103 @@ __global__ void k_agg_merge(int ngroups, int nbuckets,
     }
 }
 
+/* Fused-path aggregate partial with CLOSED-FORM bucket boundaries.  The
+ * fused scan requires every ts page to be RLE (fused_capable), so the
+ * compacted output of each page-group is an affine ts sequence
+ * (t0sel + i*delta) recorded by k_spans_rle.  Instead of 1 global binary
+ * search over out_ts per bucket boundary (the latency + traffic hot spot
+ * of k_agg_partial: ~log2(rows) dependent DRAM loads each), boundaries
+ * are found by a search over the sgroup's page-group table staged in LDS
+ * followed by one division — no out_ts reads at all.  The reduction and
+ * partial-cell layout are identical to k_agg_partial (deterministic,
+ * every cell written once). */
+__global__ void k_agg_partial_rle(const DevGroup *__restrict__ sg, int nsg,
+                                  const int32_t *__restrict__ sgfirst,
+                                  const int64_t *__restrict__ g_t0sel,
+                                  const int64_t *__restrict__ g_delta,
+                                  const int64_t *__restrict__ out_off,
+                                  const double *__restrict__ val,
+                                  int64_t t0, int64_t bucket_ns, int nbuckets,
+                                  int max_span,
+                                  double *__restrict__ pmax,
+                                  double *__restrict__ psum,
+                                  long long *__restrict__ pcnt) {
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int nwaves = blockDim.x >> 6;
+    extern __shared__ __attribute__((aligned(16))) uint8_t smem[];
+    int64_t *t0s = (int64_t *)smem;              /* [max_span] */
+    int64_t *dlt = t0s + max_span;               /* [max_span] */
+    uint32_t *rel = (uint32_t *)(dlt + max_span); /* [max_span+1] */
+    uint32_t *soff = rel + max_span + 1;         /* [nbuckets+1] */
+    for (int g = blockIdx.x; g < nsg; g += gridDim.x) {
+        const int64_t base = sg[g].row_off;
+        const int32_t nrows = sg[g].nrows;
+        const int gf = sgfirst[g];
+        const int span = sgfirst[g + 1] - gf;
+        for (int j = threadIdx.x; j < span; j += blockDim.x) {
+            rel[j] = uint32_t(out_off[gf + j] - base);
+            t0s[j] = g_t0sel[gf + j];
+            dlt[j] = g_delta[gf + j];
+        }
+        if (threadIdx.x == 0) rel[span] = uint32_t(nrows);
+        __syncthreads();
+        for (int bi = threadIdx.x; bi <= nbuckets; bi += blockDim.x) {
+            const int64_t bound = t0 + int64_t(bi) * bucket_ns;
+            /* count of selected rows with ts < bound; page-groups of one
+               series are time-ordered and non-overlapping */
+            int lo = 0, hi = span;
+            while (lo < hi) {
+                int m = (lo + hi) >> 1;
+                if (t0s[m] < bound) lo = m + 1; else hi = m;
+            }
+            uint32_t off = 0;
+            if (lo > 0) {
+                const int k = lo - 1;
+                const uint32_t cnt = rel[k + 1] - rel[k];
+                const int64_t d = dlt[k], b = bound - t0s[k]; /* b > 0 */
+                uint32_t w;
+                if (d <= 0) {
+                    w = cnt; /* constant ts < bound: all rows qualify */
+                } else {
+                    __int128 q = ((__int128)b + d - 1) / d; /* #i: i*d < b */
+                    w = q >= cnt ? cnt : uint32_t(q);
+                }
+                off = rel[k] + w;
+            }
+            soff[bi] = off;
+        }
+        __syncthreads();
+        const double *v = val + base;
+        for (int b = wave; b < nbuckets; b += nwaves) {
+            const int64_t s = soff[b], e = soff[b + 1];
+            double mx = -__builtin_inf(), sm = 0.0;
+            long long c = 0;
+            for (int64_t r = s + lane; r < e; r += 64) {
+                double x = v[r];
+                if (x > mx) mx = x;
+                sm += x;
+                c++;
+            }
+            for (int off2 = 32; off2 > 0; off2 >>= 1) {
+                double omx = __shfl_down(mx, off2, 64);
+                double osm = __shfl_down(sm, off2, 64);
+                long long oc = __shfl_down(c, off2, 64);
+                if (omx > mx) mx = omx;
+                sm += osm;
+                c += oc;
+            }
+            if (lane == 0) {
+                size_t idx = size_t(g) * nbuckets + b;
+                pmax[idx] = mx;
+                psum[idx] = sm;
+                pcnt[idx] = c;
+            }
+        }
+        __syncthreads();
+    }
+}
+
 /* ------------------------------------------------- compaction merge (k-way)
  * Config #5: k overlapping L0 column groups per series -> one merged,
  * deduped stream (compact.rs:271-404, comapcting_block_meta_group.rs:
@@ -1332,6 +1429,8 @@ __global__ void k_spans_rle(const uint8_t *__restrict__ blob,
                             int64_t lo, int64_t hi,
                             int64_t *__restrict__ sp_start,
                             int64_t *__restrict__ sp_cnt,
+                            int64_t *__restrict__ g_t0sel,
+                            int64_t *__restrict__ g_delta,
                             unsigned *__restrict__ err) {
     for (int p = blockIdx.x * blockDim.x + threadIdx.x; p < npages;
          p += gridDim.x * blockDim.x) {
@@ -1365,6 +1464,10 @@ __global__ void k_spans_rle(const uint8_t *__restrict__ blob,
         if (en < st) en = st;
         sp_start[pg.grp] = st;
         sp_cnt[pg.grp] = en - st;
+        /* RLE grid of the SELECTED rows, for the closed-form aggregate
+           boundary computation (k_agg_partial_rle) */
+        g_t0sel[pg.grp] = first + st * delta;
+        g_delta[pg.grp] = delta;
     }
 }
 
@@ -1869,6 +1972,10 @@ struct GsGroupSet {
     int64_t *d_sp_start = nullptr;
     int64_t *d_sp_cnt = nullptr;
     int64_t *d_out_off = nullptr;
+    int64_t *d_g_t0sel = nullptr; /* per page-group: ts of first selected
+                                     row / RLE delta (k_spans_rle out) */
+    int64_t *d_g_delta = nullptr;
+    int max_span = 0; /* max page-groups per series-group */
     GsTimeRange *d_ranges = nullptr;
     size_t ranges_cap = 0;
     uint8_t *d_valid = nullptr; /* lazily allocated internal validity bytes */
@@ -2185,10 +2292,14 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         hipMalloc(&set->d_out_off, (ngroups + 1) * sizeof(int64_t)) != hipSuccess ||
         hipMalloc(&set->d_blocksums, (nblocks + 1) * sizeof(int64_t)) != hipSuccess ||
         hipMalloc(&set->d_sgroup_first, sgfirst.size() * sizeof(int32_t)) != hipSuccess ||
-        hipMalloc(&set->d_sgroups_out, hsg.size() * sizeof(DevGroup)) != hipSuccess) {
+        hipMalloc(&set->d_sgroups_out, hsg.size() * sizeof(DevGroup)) != hipSuccess ||
+        hipMalloc(&set->d_g_t0sel, ngroups * sizeof(int64_t)) != hipSuccess ||
+        hipMalloc(&set->d_g_delta, ngroups * sizeof(int64_t)) != hipSuccess) {
         fail(GS_ERR, "hipMalloc group tables failed");
         gs_groups_free(set); return nullptr;
     }
+    for (int32_t sp : set->sgroup_span)
+        if (sp > set->max_span) set->max_span = sp;
     hipMemcpyAsync(set->d_sgroup_first, sgfirst.data(),
                    sgfirst.size() * sizeof(int32_t), hipMemcpyHostToDevice,
                    ctx->stream);
@@ -2216,6 +2327,8 @@ void gs_groups_free(GsGroupSet *set) {
     hipFree(set->d_sp_start);
     hipFree(set->d_sp_cnt);
     hipFree(set->d_out_off);
+    if (set->d_g_t0sel) hipFree(set->d_g_t0sel);
+    if (set->d_g_delta) hipFree(set->d_g_delta);
     if (set->d_ranges) hipFree(set->d_ranges);
     if (set->d_valid) hipFree(set->d_valid);
     if (set->d_mask) hipFree(set->d_mask);
@@ -2380,7 +2493,8 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
     hipLaunchKernelGGL(k_spans_rle, dim3(grid_for(ng, 256)), dim3(256), 0,
                        ctx->stream, set->d_blob, ts_pages, ng,
                        spec->range.min_ts, spec->range.max_ts,
-                       set->d_sp_start, set->d_sp_cnt, ctx->d_err);
+                       set->d_sp_start, set->d_sp_cnt, set->d_g_t0sel,
+                       set->d_g_delta, ctx->d_err);
     /* device exclusive scan of span counts -> output offsets (+total) */
     hipLaunchKernelGGL(k_scan_partials, dim3(nblocks), dim3(SCAN_BLOCK), 0,
                        ctx->stream, set->d_sp_cnt, ng, set->d_out_off,
@@ -2419,15 +2533,32 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                 return fail(GS_ERR, "hipMalloc agg partials failed");
             set->partials_cap = cells;
         }
-        size_t agg_shm = spec->n_buckets <= 8192
-                             ? (size_t(spec->n_buckets) + 1) * 4 : 0;
-        hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
-                           dim3(256), agg_shm, ctx->stream,
-                           set->d_sgroups_out, nsg,
-                           spec->d_out_ts, spec->d_out_val, nullptr,
-                           INT64_MIN, INT64_MAX, spec->t0, spec->bucket_ns,
-                           spec->n_buckets, set->d_pmax, set->d_psum,
-                           set->d_pcnt);
+        /* closed-form boundaries via the RLE group table when it fits in
+           LDS (it always does for realistic pages-per-series); otherwise
+           the out_ts binary-search kernel */
+        size_t shm_rle = size_t(set->max_span) * 16 +
+                         (size_t(set->max_span) + size_t(spec->n_buckets) + 2)
+                             * 4;
+        if (spec->n_buckets <= 8192 && shm_rle <= 64 * 1024) {
+            hipLaunchKernelGGL(k_agg_partial_rle,
+                               dim3(nsg > 65535 ? 65535 : nsg), dim3(256),
+                               shm_rle, ctx->stream, set->d_sgroups_out, nsg,
+                               set->d_sgroup_first, set->d_g_t0sel,
+                               set->d_g_delta, set->d_out_off,
+                               spec->d_out_val, spec->t0, spec->bucket_ns,
+                               spec->n_buckets, set->max_span, set->d_pmax,
+                               set->d_psum, set->d_pcnt);
+        } else {
+            size_t agg_shm = spec->n_buckets <= 8192
+                                 ? (size_t(spec->n_buckets) + 1) * 4 : 0;
+            hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
+                               dim3(256), agg_shm, ctx->stream,
+                               set->d_sgroups_out, nsg,
+                               spec->d_out_ts, spec->d_out_val, nullptr,
+                               INT64_MIN, INT64_MAX, spec->t0, spec->bucket_ns,
+                               spec->n_buckets, set->d_pmax, set->d_psum,
+                               set->d_pcnt);
+        }
         int mb = (spec->n_buckets + 3) / 4;
         hipLaunchKernelGGL(k_agg_merge, dim3(mb > 2048 ? 2048 : mb), dim3(256),
                            0, ctx->stream, nsg, spec->n_buckets, set->d_pmax,
