@@ -604,3 +604,68 @@ def test_truncated_gorilla_stream_errors(engine):
     with pytest.raises(RuntimeError):
         engine.decode(gset, 1, out)
     gset.free()
+
+
+def test_scan_fused_agg_multigroup(engine):
+    """Fused aggregate with several page-groups per series (merged into
+    one series-group): per-group RLE deltas differ, groups are separated
+    by time gaps, the scan range cuts mid-group and leaves whole groups
+    empty — exercises the closed-form boundary table of
+    k_agg_partial_rle (page-group search + per-group division)."""
+    r = np.random.default_rng(53)
+    t0 = 1_700_000_000_000_000_000
+    ns = 10**9
+    groups, truth = [], []
+    for s in range(7):
+        cur = t0 if s < 6 else t0 - 50_000 * ns  # series 6: fully below lo
+        sts, svals = [], []
+        for _ in range(7):
+            n = int(r.integers(100, 900))
+            step = int(r.choice([1, 2, 5, 10])) * ns
+            ts = cur + np.arange(n, dtype=np.int64) * step
+            vals = np.round(r.normal(50, 10, n), 2)
+            groups.append((s, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                               (gs.page_of(vals, gs.CT_F64), gs.CT_F64)]))
+            sts.append(ts)
+            svals.append(vals)
+            cur = ts[-1] + int(r.integers(1, 1000)) * ns  # gap between groups
+        truth.append((np.concatenate(sts), np.concatenate(svals)))
+    gset = engine.upload(groups)
+    rows = gset.rows
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    lo, hi = t0 + 700 * ns, t0 + 9000 * ns
+    bucket_ns = 300 * ns
+    nb = 40
+    d_max = torch.full((nb,), -np.inf, dtype=torch.float64, device="cuda")
+    d_sum = torch.zeros(nb, dtype=torch.float64, device="cuda")
+    d_cnt = torch.zeros(nb, dtype=torch.int64, device="cuda")
+    res = engine.scan(gset, d_ts, d_val, time_range=(lo, hi),
+                      d_out_ts=d_ots, d_out_val=d_oval,
+                      agg=dict(bucket_ns=bucket_ns, t0=t0, n_buckets=nb,
+                               d_max=d_max, d_sum=d_sum, d_count=d_cnt))
+    exp_ts, exp_val = [], []
+    emx = np.full(nb, -np.inf)
+    esm = np.zeros(nb)
+    ect = np.zeros(nb, dtype=np.int64)
+    for ts, vals in truth:
+        sel = (ts >= lo) & (ts <= hi)
+        exp_ts.append(ts[sel])
+        exp_val.append(vals[sel])
+        m, su, c = orc.bucket_agg(ts[sel], vals[sel], None, t0, bucket_ns, nb)
+        emx = np.maximum(emx, m)
+        esm += su
+        ect += c
+    exp_ts_c = np.concatenate(exp_ts)
+    exp_val_c = np.concatenate(exp_val).astype(np.float64)
+    assert res.out_rows == exp_ts_c.size
+    assert (d_ots[:res.out_rows].cpu().numpy() == exp_ts_c).all()
+    got = d_oval[:res.out_rows].cpu().numpy()
+    assert got.view(np.uint64).tolist() == exp_val_c.view(np.uint64).tolist()
+    assert (d_cnt.cpu().numpy() == ect).all()
+    gmx = d_max.cpu().numpy()
+    assert (gmx[ect > 0] == emx[ect > 0]).all()
+    assert np.allclose(d_sum.cpu().numpy(), esm, rtol=1e-12)
+    gset.free()
