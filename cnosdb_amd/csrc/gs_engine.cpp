@@ -1155,11 +1155,53 @@ __global__ void k_agg_partial_rle(const DevGroup *__restrict__ sg, int nsg,
         }
         __syncthreads();
         const double *v = val + base;
-        for (int b = wave; b < nbuckets; b += nwaves) {
+        /* CONTIGUOUS bucket range per wave, cut at ROW quantiles via the
+           soff table: each wave sweeps one monotone row stream of ~equal
+           length (HW prefetch friendly + load-balanced even when the
+           time range empties the outer buckets).  Empty buckets skip the
+           shuffle tree. */
+        int b0, b1;
+        {
+            const uint32_t q0 = uint32_t(int64_t(nrows) * wave / nwaves);
+            const uint32_t q1 = uint32_t(int64_t(nrows) * (wave + 1) / nwaves);
+            int lo2 = 0, hi2 = nbuckets + 1;
+            while (lo2 < hi2) {
+                int m = (lo2 + hi2) >> 1;
+                if (soff[m] < q0) lo2 = m + 1; else hi2 = m;
+            }
+            b0 = lo2 < nbuckets ? lo2 : nbuckets;
+            hi2 = nbuckets + 1;
+            while (lo2 < hi2) {
+                int m = (lo2 + hi2) >> 1;
+                if (soff[m] < q1) lo2 = m + 1; else hi2 = m;
+            }
+            b1 = lo2 < nbuckets ? lo2 : nbuckets;
+            /* trailing empty buckets (soff hits nrows early) still need
+               their partial cells written: stretch the last wave */
+            if (wave == nwaves - 1) b1 = nbuckets;
+        }
+        for (int b = b0; b < b1; b++) {
             const int64_t s = soff[b], e = soff[b + 1];
+            size_t idx = size_t(g) * nbuckets + b;
+            if (e <= s) {
+                if (lane == 0) {
+                    pmax[idx] = -__builtin_inf();
+                    psum[idx] = 0.0;
+                    pcnt[idx] = 0;
+                }
+                continue;
+            }
             double mx = -__builtin_inf(), sm = 0.0;
             long long c = 0;
-            for (int64_t r = s + lane; r < e; r += 64) {
+            int64_t r = s + lane;
+            for (; r + 64 < e; r += 128) { /* 2 loads in flight per iter */
+                double x0 = v[r], x1 = v[r + 64];
+                if (x0 > mx) mx = x0;
+                if (x1 > mx) mx = x1;
+                sm += x0 + x1;
+                c += 2;
+            }
+            if (r < e) {
                 double x = v[r];
                 if (x > mx) mx = x;
                 sm += x;
@@ -1174,7 +1216,6 @@ __global__ void k_agg_partial_rle(const DevGroup *__restrict__ sg, int nsg,
                 c += oc;
             }
             if (lane == 0) {
-                size_t idx = size_t(g) * nbuckets + b;
                 pmax[idx] = mx;
                 psum[idx] = sm;
                 pcnt[idx] = c;
