@@ -198,3 +198,46 @@ def bucket_agg(ts, vals, valid, t0, bucket_ns, n_buckets):
     np.add.at(sm, b, v)
     np.add.at(ct, b, 1)
     return mx, sm, ct
+
+
+def encode_str(strings):
+    """string.rs:32-88 — Snappy string block [7][0x10][snappy raw]."""
+    lib = Oracle().lib
+    lib.orc_str_encode.restype = ctypes.c_int64
+    concat = b"".join(strings)
+    lens = np.array([len(s) for s in strings], dtype=np.uint64)
+    src = np.frombuffer(concat, dtype=np.uint8) if concat else np.zeros(1, np.uint8)
+    cap = 2 + 32 + len(concat) + len(concat) // 6 + lens.size * 10 + 64
+    dst = np.zeros(cap, dtype=np.uint8)
+    n = lib.orc_str_encode(_ptr(src), _ptr(lens), len(strings), _ptr(dst),
+                           dst.size)
+    if n < 0:
+        raise RuntimeError("orc_str_encode failed")
+    return dst[:n].tobytes()
+
+
+def decode_str(data, nrows, valid=None):
+    """str_snappy_decode_to_array semantics (string.rs:226-276): returns
+    (list of bytes-or-None per row).  Handles Snappy=7 and Null=1."""
+    lib = Oracle().lib
+    lib.orc_str_decode.restype = ctypes.c_int64
+    src = np.frombuffer(data, dtype=np.uint8) if data else np.zeros(1, np.uint8)
+    if valid is not None:
+        bs = np.packbits(np.asarray(valid, dtype=bool), bitorder="little")
+        bsp = _ptr(bs)
+    else:
+        bsp = None
+    bytes_out = np.zeros(len(data) * 64 + (1 << 20), dtype=np.uint8)
+    lens = np.zeros(max(1, nrows), dtype=np.int64)
+    n = lib.orc_str_decode(_ptr(src), len(data), bsp, nrows, _ptr(bytes_out),
+                           bytes_out.size, _ptr(lens))
+    if n < 0:
+        raise RuntimeError("orc_str_decode failed")
+    out, off = [], 0
+    for r in range(nrows):
+        if lens[r] < 0:
+            out.append(None)
+        else:
+            out.append(bytes_out[off:off + lens[r]].tobytes())
+            off += lens[r]
+    return out

@@ -911,3 +911,304 @@ ORC_EXPORT void orc_ts_span(const int64_t *ts, int64_t nrows, int64_t mn, int64_
     *start = s;
     *count = lo - s;
 }
+
+/* ======================= string codec (Snappy) =========================
+ * Restatement of the reference string block codec
+ * (tskv/src/tsm/codec/string.rs:32-88 encode, :185-276 decode).  Block
+ * layout: [Encoding::Snappy=7][STRING_COMPRESSED_SNAPPY<<4 = 0x10]
+ * [snappy RAW stream of payload], payload = per string
+ * [LEB128 varint len][bytes].  The Snappy arithmetic itself lives in the
+ * un-vendored `snap` crate v1.1.1 (Cargo.lock) — a port of Google's
+ * published snappy reference algorithm — so the compressor below
+ * restates that published algorithm (64 KiB fragments, 256..16384-entry
+ * hash table of 4-byte loads, hash mul 0x1e35a7bd, skip-32 search
+ * acceleration, 15-byte input margin) and is pinned byte-exactly by the
+ * reference's own golden vectors (string.rs:529-566 encode_single/
+ * multi_compressed/unicode/invalid_utf8, transcribed in tests/golden/).
+ * The uncompressed variant (Encoding::Null=1, [u64 BE len][bytes] per
+ * string) follows string.rs:169-183. */
+
+#define SNAP_BLOCK 65536
+#define SNAP_MARGIN 15
+#define SNAP_MAX_TABLE 16384
+
+static uint32_t snap_load32(const uint8_t *p) {
+    uint32_t v; memcpy(&v, p, 4); return v; /* little-endian hosts only */
+}
+
+static uint8_t *snap_emit_literal(uint8_t *op, const uint8_t *lit, size_t n) {
+    size_t n1 = n - 1;
+    if (n1 < 60) {
+        *op++ = (uint8_t)(n1 << 2);
+    } else {
+        uint8_t *tag = op++;
+        int c = 0;
+        size_t v = n1;
+        while (v > 0) { *op++ = v & 0xff; v >>= 8; c++; }
+        *tag = (uint8_t)((59 + c) << 2);
+    }
+    memcpy(op, lit, n);
+    return op + n;
+}
+
+static uint8_t *snap_emit_copy_upto64(uint8_t *op, size_t offset, size_t len) {
+    if (len < 12 && offset < 2048) {
+        *op++ = (uint8_t)(1 | ((len - 4) << 2) | ((offset >> 8) << 5));
+        *op++ = (uint8_t)(offset & 0xff);
+    } else {
+        *op++ = (uint8_t)(2 | ((len - 1) << 2));
+        *op++ = (uint8_t)(offset & 0xff);
+        *op++ = (uint8_t)((offset >> 8) & 0xff);
+    }
+    return op;
+}
+
+static uint8_t *snap_emit_copy(uint8_t *op, size_t offset, size_t len) {
+    while (len >= 68) { op = snap_emit_copy_upto64(op, offset, 64); len -= 64; }
+    if (len > 64) { op = snap_emit_copy_upto64(op, offset, 60); len -= 60; }
+    return snap_emit_copy_upto64(op, offset, len);
+}
+
+static uint8_t *snap_compress_fragment(const uint8_t *input, size_t n,
+                                       uint8_t *op, uint16_t *table) {
+    size_t table_size = 256;
+    while (table_size < SNAP_MAX_TABLE && table_size < n) table_size <<= 1;
+    int shift = 32;
+    for (size_t t = table_size; t > 1; t >>= 1) shift--;
+    memset(table, 0, table_size * sizeof(uint16_t));
+    const uint8_t *ip = input, *ip_end = input + n;
+    const uint8_t *next_emit = ip;
+    if (n >= SNAP_MARGIN) {
+        const uint8_t *ip_limit = ip_end - SNAP_MARGIN;
+        ip++;
+        uint32_t next_hash = (snap_load32(ip) * 0x1e35a7bdu) >> shift;
+        for (;;) {
+            uint32_t skip = 32;
+            const uint8_t *next_ip = ip;
+            const uint8_t *candidate;
+            do {
+                ip = next_ip;
+                uint32_t h = next_hash;
+                uint32_t adv = skip >> 5;
+                skip += adv;
+                next_ip = ip + adv;
+                if (next_ip > ip_limit) goto emit_remainder;
+                next_hash = (snap_load32(next_ip) * 0x1e35a7bdu) >> shift;
+                candidate = input + table[h];
+                table[h] = (uint16_t)(ip - input);
+            } while (snap_load32(ip) != snap_load32(candidate));
+            op = snap_emit_literal(op, next_emit, (size_t)(ip - next_emit));
+            uint32_t cand32;
+            do {
+                const uint8_t *base = ip;
+                size_t matched = 4;
+                {
+                    const uint8_t *s1 = candidate + 4, *s2 = ip + 4;
+                    while (s2 < ip_end && *s1 == *s2) { s1++; s2++; matched++; }
+                }
+                ip += matched;
+                op = snap_emit_copy(op, (size_t)(base - candidate), matched);
+                next_emit = ip;
+                if (ip >= ip_limit) goto emit_remainder;
+                uint32_t ph = (snap_load32(ip - 1) * 0x1e35a7bdu) >> shift;
+                table[ph] = (uint16_t)(ip - 1 - input);
+                uint32_t ch = (snap_load32(ip) * 0x1e35a7bdu) >> shift;
+                candidate = input + table[ch];
+                cand32 = snap_load32(candidate);
+                table[ch] = (uint16_t)(ip - input);
+            } while (snap_load32(ip) == cand32);
+            ip++;
+            next_hash = (snap_load32(ip) * 0x1e35a7bdu) >> shift;
+        }
+    }
+emit_remainder:
+    if (next_emit < ip_end)
+        op = snap_emit_literal(op, next_emit, (size_t)(ip_end - next_emit));
+    return op;
+}
+
+ORC_EXPORT int64_t orc_snappy_max_compress_len(int64_t n) {
+    return 32 + n + n / 6; /* snap::raw::max_compress_len formula */
+}
+
+ORC_EXPORT int64_t orc_snappy_compress(const uint8_t *src, size_t n,
+                                       uint8_t *dst, size_t cap) {
+    if (cap < (size_t)orc_snappy_max_compress_len((int64_t)n)) return -1;
+    uint8_t *op = dst;
+    size_t v = n; /* preamble: varint uncompressed length */
+    while (v >= 0x80) { *op++ = (uint8_t)(v | 0x80); v >>= 7; }
+    *op++ = (uint8_t)v;
+    uint16_t table[SNAP_MAX_TABLE];
+    for (size_t off = 0; off < n; off += SNAP_BLOCK) {
+        size_t frag = n - off < SNAP_BLOCK ? n - off : SNAP_BLOCK;
+        op = snap_compress_fragment(src + off, frag, op, table);
+    }
+    if (n == 0) { /* empty input: preamble only (snappy of "" is [0]) */ }
+    return op - dst;
+}
+
+ORC_EXPORT int64_t orc_snappy_uncompressed_len(const uint8_t *src, size_t len,
+                                               int *consumed) {
+    uint64_t v = 0;
+    int shift_ = 0, i = 0;
+    while ((size_t)i < len && i < 10) {
+        uint8_t b = src[i++];
+        v |= (uint64_t)(b & 0x7f) << shift_;
+        if (!(b & 0x80)) { if (consumed) *consumed = i; return (int64_t)v; }
+        shift_ += 7;
+    }
+    return -1;
+}
+
+ORC_EXPORT int64_t orc_snappy_decompress(const uint8_t *src, size_t len,
+                                         uint8_t *dst, size_t cap) {
+    int hdr = 0;
+    int64_t ulen = orc_snappy_uncompressed_len(src, len, &hdr);
+    if (ulen < 0 || (size_t)ulen > cap) return -1;
+    const uint8_t *ip = src + hdr, *end = src + len;
+    uint8_t *op = dst, *op_end = dst + ulen;
+    while (ip < end) {
+        uint8_t tag = *ip++;
+        size_t l, off;
+        switch (tag & 3) {
+        case 0:
+            l = tag >> 2;
+            if (l >= 60) {
+                int c = (int)l - 59;
+                if (ip + c > end) return -1;
+                l = 0;
+                for (int i = 0; i < c; i++) l |= (size_t)ip[i] << (8 * i);
+                ip += c;
+            }
+            l += 1;
+            if (ip + l > end || op + l > op_end) return -1;
+            memcpy(op, ip, l);
+            ip += l;
+            op += l;
+            continue;
+        case 1:
+            if (ip >= end) return -1;
+            l = ((tag >> 2) & 7) + 4;
+            off = ((size_t)(tag >> 5) << 8) | *ip++;
+            break;
+        case 2:
+            if (ip + 2 > end) return -1;
+            l = (tag >> 2) + 1;
+            off = (size_t)ip[0] | ((size_t)ip[1] << 8);
+            ip += 2;
+            break;
+        default:
+            if (ip + 4 > end) return -1;
+            l = (tag >> 2) + 1;
+            off = (size_t)ip[0] | ((size_t)ip[1] << 8) |
+                  ((size_t)ip[2] << 16) | ((size_t)ip[3] << 24);
+            ip += 4;
+            break;
+        }
+        if (off == 0 || (size_t)(op - dst) < off || op + l > op_end) return -1;
+        const uint8_t *cp = op - off; /* may overlap: byte-by-byte */
+        for (size_t i = 0; i < l; i++) op[i] = cp[i];
+        op += l;
+    }
+    return (op == op_end && ip == end) ? ulen : -1;
+}
+
+/* strings -> block bytes (string.rs:32-88).  src = concatenated string
+ * bytes; lens[i] = byte length of string i.  Empty input -> 0 bytes
+ * (string.rs:33-35). */
+ORC_EXPORT int64_t orc_str_encode(const uint8_t *src, const uint64_t *lens,
+                                  int64_t nstr, uint8_t *dst, size_t cap) {
+    if (nstr == 0) return 0;
+    size_t payload = 0;
+    for (int64_t i = 0; i < nstr; i++) {
+        uint64_t v = lens[i];
+        payload += lens[i];
+        do { payload++; v >>= 7; } while (v > 0);
+    }
+    uint8_t *buf = (uint8_t *)malloc(payload);
+    if (!buf) return -1;
+    size_t p = 0, s = 0;
+    for (int64_t i = 0; i < nstr; i++) {
+        uint64_t v = lens[i];
+        while (v >= 0x80) { buf[p++] = (uint8_t)(v | 0x80); v >>= 7; }
+        buf[p++] = (uint8_t)v;
+        memcpy(buf + p, src + s, lens[i]);
+        p += lens[i];
+        s += lens[i];
+    }
+    if (cap < 2 + (size_t)orc_snappy_max_compress_len((int64_t)payload)) {
+        free(buf);
+        return -1;
+    }
+    dst[0] = 7;    /* Encoding::Snappy, models/src/codec.rs:48 */
+    dst[1] = 0x10; /* STRING_COMPRESSED_SNAPPY << 4, string.rs:21,79 */
+    int64_t c = orc_snappy_compress(buf, payload, dst + 2, cap - 2);
+    free(buf);
+    return c < 0 ? -1 : c + 2;
+}
+
+/* block bytes + bitset -> concatenated strings + per-row lengths
+ * (str_snappy_decode_to_array, string.rs:226-276; nulls consume nothing
+ * from the payload; empty src -> all rows null; Encoding::Null blocks
+ * use [u64 BE len][bytes] per string, string.rs:169-183,289-300).
+ * lens_out[row] = -1 for null rows.  Returns total payload bytes. */
+ORC_EXPORT int64_t orc_str_decode(const uint8_t *src, size_t len,
+                                  const uint8_t *bitset, int64_t nrows,
+                                  uint8_t *bytes_out, size_t bytes_cap,
+                                  int64_t *lens_out) {
+    for (int64_t r = 0; r < nrows; r++) lens_out[r] = -1;
+    if (len == 0) return 0;
+    uint8_t *payload = NULL;
+    const uint8_t *pl;
+    size_t pn;
+    int be_lens = 0;
+    if (src[0] == 7) { /* Snappy */
+        if (len < 2) return -1;
+        int64_t ulen = orc_snappy_uncompressed_len(src + 2, len - 2, NULL);
+        if (ulen < 0) return -1;
+        payload = (uint8_t *)malloc(ulen ? (size_t)ulen : 1);
+        if (!payload) return -1;
+        if (orc_snappy_decompress(src + 2, len - 2, payload, (size_t)ulen) != ulen) {
+            free(payload);
+            return -1;
+        }
+        pl = payload;
+        pn = (size_t)ulen;
+    } else if (src[0] == 1) { /* Encoding::Null: uncompressed */
+        pl = src + 1;
+        pn = len - 1;
+        be_lens = 1;
+    } else {
+        return -1;
+    }
+    size_t i = 0, w = 0;
+    int64_t rc = 0;
+    for (int64_t r = 0; r < nrows; r++) {
+        if (bitset && !((bitset[r >> 3] >> (r & 7)) & 1)) continue;
+        if (i >= pn) break; /* reference stops silently at payload end */
+        uint64_t slen;
+        if (be_lens) {
+            if (i + 8 > pn) { rc = -1; break; }
+            slen = 0;
+            for (int k = 0; k < 8; k++) slen = (slen << 8) | pl[i + k];
+            i += 8;
+        } else {
+            slen = 0;
+            int sh = 0, ok = 0;
+            while (i < pn) {
+                uint8_t b = pl[i++];
+                slen |= (uint64_t)(b & 0x7f) << sh;
+                sh += 7;
+                if (!(b & 0x80)) { ok = 1; break; }
+            }
+            if (!ok) { rc = -1; break; }
+        }
+        if (i + slen > pn || w + slen > bytes_cap) { rc = -1; break; }
+        memcpy(bytes_out + w, pl + i, slen);
+        lens_out[r] = (int64_t)slen;
+        i += slen;
+        w += slen;
+    }
+    if (payload) free(payload);
+    return rc < 0 ? -1 : (int64_t)w;
+}

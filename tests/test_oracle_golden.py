@@ -158,3 +158,57 @@ def test_simple8b_run_of_ones_selectors():
     enc = orc.encode_i64(base)
     assert (enc[1] >> 4) == 1
     assert orc.decode_i64(enc, base.size).tolist() == base.tolist()
+
+
+def test_str_snappy_golden_vectors():
+    """string.rs:529-566: the snap-1.1.1 compressor output is pinned
+    byte-exactly by the reference's own encode tests."""
+    g = GOLD["str_snappy"]
+    for name in ("single", "multi_compressed", "unicode"):
+        strs = [s.encode() for s in g[name]["strings"]]
+        blk = orc.encode_str(strs)
+        assert blk[0] == 7, name  # Encoding::Snappy
+        assert list(blk[1:]) == g[name]["bytes"], name
+        got = orc.decode_str(blk, len(strs))
+        assert got == strs, name
+    blk = orc.encode_str([b"\xC0"])
+    assert list(blk[1:]) == g["invalid_utf8"]["bytes"]
+    assert orc.decode_str(blk, 1) == [b"\xC0"]
+
+
+def test_str_snappy_roundtrip_corpus():
+    """ALLSTR round-trip (string.rs:680-700) + randomized corpora with
+    nulls, empty strings, and a >64 KiB payload (multi-fragment)."""
+    allstr = [s.encode() for s in GOLD["str_snappy"]["allstr_roundtrip"]["strings"]]
+    assert orc.decode_str(orc.encode_str(allstr), len(allstr)) == allstr
+    r = np.random.default_rng(11)
+    tags = [b"hostname=server_%03d" % i for i in range(32)]
+    for trial in range(5):
+        n = int(r.integers(1, 2000))
+        strs = []
+        for _ in range(n):
+            k = r.integers(0, 4)
+            if k == 0:
+                strs.append(tags[int(r.integers(0, 32))])
+            elif k == 1:
+                strs.append(b"")
+            else:
+                strs.append(bytes(r.integers(0, 256, int(r.integers(0, 80))).astype(np.uint8)))
+        valid = r.random(n) > 0.2
+        present = [s for s, v in zip(strs, valid) if v]
+        blk = orc.encode_str(present)
+        got = orc.decode_str(blk, n, valid)
+        exp = [s if v else None for s, v in zip(strs, valid)]
+        assert got == exp, trial
+    big = [bytes(r.integers(97, 123, 50).astype(np.uint8)) for _ in range(3000)]
+    assert orc.decode_str(orc.encode_str(big), len(big)) == big  # >64KB payload
+
+
+def test_str_uncompressed_and_empty():
+    """Encoding::Null string blocks ([u64 BE len][bytes], string.rs:
+    169-183) and the empty-source -> all-null rule (string.rs:231-236)."""
+    strs = [b"abc", b"", b"zz"]
+    blk = bytes([1]) + b"".join(
+        len(s).to_bytes(8, "big") + s for s in strs)
+    assert orc.decode_str(blk, 3) == strs
+    assert orc.decode_str(b"", 4) == [None] * 4
