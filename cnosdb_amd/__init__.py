@@ -18,6 +18,8 @@ from .host import (
     encode_i64,
     encode_f64,
     encode_bool,
+    encode_str,
+    str_page_of,
     build_page,
     page_of,
     CT_TIME,
@@ -25,12 +27,13 @@ from .host import (
     CT_F64,
     CT_BOOL,
     CT_U64,
+    CT_STR,
 )
 
 __all__ = [
     "Engine", "GroupSet", "PageLib", "lib_path",
-    "encode_ts", "encode_i64", "encode_f64", "encode_bool",
-    "build_page", "page_of",
-    "CT_TIME", "CT_I64", "CT_F64", "CT_BOOL", "CT_U64",
+    "encode_ts", "encode_i64", "encode_f64", "encode_bool", "encode_str",
+    "build_page", "page_of", "str_page_of",
+    "CT_TIME", "CT_I64", "CT_F64", "CT_BOOL", "CT_U64", "CT_STR",
 ]
 __version__ = "0.1"

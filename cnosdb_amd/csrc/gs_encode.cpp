@@ -287,6 +287,141 @@ uint32_t gs_crc32(const uint8_t *data, size_t len) {
     return c ^ 0xFFFFFFFFu;
 }
 
+/* ---------------- string block encoder (codec/string.rs:32-88) ----------
+ * [Encoding::Snappy=7][0x10][snappy raw stream] over a payload of
+ * [LEB128 varint len][bytes] per string.  The snappy compressor restates
+ * the published reference algorithm used by the un-vendored `snap` crate
+ * v1.1.1 (64 KiB fragments, power-of-two hash table in [256,16384] of
+ * 4-byte little-endian loads hashed by *0x1e35a7bd>>shift, skip-32
+ * acceleration, 15-byte tail margin); its output is pinned byte-exactly
+ * by the reference's own golden vectors (string.rs:529-566), checked in
+ * tests/test_encoders.py.  Independent of oracle/tsm_oracle.c. */
+
+namespace {
+
+inline uint32_t sload32(const uint8_t *p) {
+    uint32_t v;
+    memcpy(&v, p, 4);
+    return v;
+}
+
+inline uint8_t *slit(uint8_t *op, const uint8_t *p, size_t n) {
+    if (n - 1 < 60) {
+        *op++ = uint8_t((n - 1) << 2);
+    } else {
+        uint8_t *tag = op++;
+        size_t v = n - 1;
+        int nb = 0;
+        for (; v; v >>= 8) *op++ = uint8_t(v), nb++;
+        *tag = uint8_t((59 + nb) << 2);
+    }
+    memcpy(op, p, n);
+    return op + n;
+}
+
+inline uint8_t *scopy1(uint8_t *op, size_t off, size_t len) {
+    if (len < 12 && off < 2048) {
+        *op++ = uint8_t(1 | ((len - 4) << 2) | ((off >> 8) << 5));
+        *op++ = uint8_t(off);
+    } else {
+        *op++ = uint8_t(2 | ((len - 1) << 2));
+        *op++ = uint8_t(off);
+        *op++ = uint8_t(off >> 8);
+    }
+    return op;
+}
+
+inline uint8_t *scopy(uint8_t *op, size_t off, size_t len) {
+    for (; len >= 68; len -= 64) op = scopy1(op, off, 64);
+    if (len > 64) op = scopy1(op, off, 60), len -= 60;
+    return scopy1(op, off, len);
+}
+
+uint8_t *snappy_fragment(const uint8_t *in, size_t n, uint8_t *op,
+                         uint16_t *tab) {
+    size_t tsz = 256;
+    while (tsz < 16384 && tsz < n) tsz <<= 1;
+    int shift = 32 - __builtin_ctzll(tsz);
+    memset(tab, 0, tsz * sizeof(uint16_t));
+    const uint8_t *ip = in, *iend = in + n, *nmit = in;
+    if (n >= 15) {
+        const uint8_t *ilim = iend - 15;
+        uint32_t nh = (sload32(++ip) * 0x1e35a7bdu) >> shift;
+        for (;;) {
+            uint32_t skip = 32;
+            const uint8_t *nip = ip, *cand;
+            do {
+                ip = nip;
+                uint32_t h = nh, adv = skip >> 5;
+                skip += adv;
+                nip = ip + adv;
+                if (nip > ilim) goto tail;
+                nh = (sload32(nip) * 0x1e35a7bdu) >> shift;
+                cand = in + tab[h];
+                tab[h] = uint16_t(ip - in);
+            } while (sload32(ip) != sload32(cand));
+            op = slit(op, nmit, size_t(ip - nmit));
+            uint32_t c32;
+            do {
+                const uint8_t *b = ip;
+                const uint8_t *m1 = cand + 4, *m2 = ip + 4;
+                while (m2 < iend && *m1 == *m2) m1++, m2++;
+                size_t mlen = size_t(m2 - ip);
+                ip += mlen;
+                op = scopy(op, size_t(b - cand), mlen);
+                nmit = ip;
+                if (ip >= ilim) goto tail;
+                uint32_t hp = (sload32(ip - 1) * 0x1e35a7bdu) >> shift;
+                tab[hp] = uint16_t(ip - 1 - in);
+                uint32_t hc = (sload32(ip) * 0x1e35a7bdu) >> shift;
+                cand = in + tab[hc];
+                c32 = sload32(cand);
+                tab[hc] = uint16_t(ip - in);
+            } while (sload32(ip) == c32);
+            nh = (sload32(++ip) * 0x1e35a7bdu) >> shift;
+        }
+    }
+tail:
+    if (nmit < iend) op = slit(op, nmit, size_t(iend - nmit));
+    return op;
+}
+
+} // namespace
+
+int64_t gs_encode_str(const uint8_t *src, const uint64_t *lens, int64_t nstr,
+                      uint8_t *dst, size_t cap) {
+    if (nstr == 0) return 0;
+    size_t payload = 0;
+    for (int64_t i = 0; i < nstr; i++) {
+        uint64_t v = lens[i];
+        payload += size_t(lens[i]);
+        do payload++; while ((v >>= 7));
+    }
+    std::vector<uint8_t> buf(payload);
+    size_t p = 0, s = 0;
+    for (int64_t i = 0; i < nstr; i++) {
+        uint64_t v = lens[i];
+        for (; v >= 0x80; v >>= 7) buf[p++] = uint8_t(v | 0x80);
+        buf[p++] = uint8_t(v);
+        memcpy(buf.data() + p, src + s, size_t(lens[i]));
+        p += size_t(lens[i]);
+        s += size_t(lens[i]);
+    }
+    if (cap < 2 + 32 + payload + payload / 6) return -5;
+    dst[0] = 7;    /* Encoding::Snappy */
+    dst[1] = 0x10; /* STRING_COMPRESSED_SNAPPY << 4 */
+    uint8_t *op = dst + 2;
+    size_t v = payload;
+    for (; v >= 0x80; v >>= 7) *op++ = uint8_t(v | 0x80);
+    *op++ = uint8_t(v);
+    std::vector<uint16_t> tab(16384);
+    for (size_t off = 0; off < payload; off += 65536) {
+        size_t frag = payload - off < 65536 ? payload - off : 65536;
+        op = snappy_fragment(buf.data() + off, frag, op, tab.data());
+    }
+    return op - dst;
+}
+
 /* page assembly, tsm/page.rs:488-497 */
 int64_t gs_build_page(const uint8_t *bitset, int64_t nrows, const uint8_t *data,
                       size_t data_len, uint8_t *dst, size_t cap) {

@@ -1225,6 +1225,164 @@ __global__ void k_agg_partial_rle(const DevGroup *__restrict__ sg, int nsg,
     }
 }
 
+/* --------------------- string column decode (PC_STR) ---------------------
+ * codec/string.rs snappy blocks ([7][0x10][snappy raw of varint-prefixed
+ * strings]) and uncompressed blocks ([1][u64 BE len][bytes]...).  The
+ * snappy bitstream is strictly sequential, so (like Gorilla) parallelism
+ * is pages in flight: one thread decompresses one page into its scratch
+ * slab, then walks the payload against the validity bitset recording
+ * (pos,len) per row; a device scan of the lengths builds the Arrow
+ * offsets and a gather packs the bytes. */
+
+__device__ int64_t dev_snappy_decompress(const uint8_t *__restrict__ src,
+                                         uint32_t len,
+                                         uint8_t *__restrict__ dst) {
+    uint64_t ulen = 0;
+    uint32_t i = 0;
+    int sh = 0, ok = 0;
+    while (i < len && i < 10) {
+        uint8_t b = src[i++];
+        ulen |= uint64_t(b & 0x7f) << sh;
+        sh += 7;
+        if (!(b & 0x80)) { ok = 1; break; }
+    }
+    if (!ok) return -1;
+    const uint8_t *ip = src + i, *end = src + len;
+    uint8_t *op = dst, *op_end = dst + ulen;
+    while (ip < end) {
+        uint8_t tag = *ip++;
+        uint32_t l, off;
+        switch (tag & 3) {
+        case 0: {
+            l = tag >> 2;
+            if (l >= 60) {
+                int c = int(l) - 59;
+                if (ip + c > end) return -1;
+                l = 0;
+                for (int k = 0; k < c; k++) l |= uint32_t(ip[k]) << (8 * k);
+                ip += c;
+            }
+            l += 1;
+            if (ip + l > end || op + l > op_end) return -1;
+            for (uint32_t k = 0; k < l; k++) op[k] = ip[k];
+            ip += l;
+            op += l;
+            continue;
+        }
+        case 1:
+            if (ip >= end) return -1;
+            l = ((tag >> 2) & 7) + 4;
+            off = (uint32_t(tag >> 5) << 8) | *ip++;
+            break;
+        case 2:
+            if (ip + 2 > end) return -1;
+            l = (tag >> 2) + 1;
+            off = uint32_t(ip[0]) | (uint32_t(ip[1]) << 8);
+            ip += 2;
+            break;
+        default:
+            if (ip + 4 > end) return -1;
+            l = (tag >> 2) + 1;
+            off = uint32_t(ip[0]) | (uint32_t(ip[1]) << 8) |
+                  (uint32_t(ip[2]) << 16) | (uint32_t(ip[3]) << 24);
+            ip += 4;
+            break;
+        }
+        if (off == 0 || uint64_t(op - dst) < off || op + l > op_end) return -1;
+        const uint8_t *cp = op - off; /* may overlap: strictly in order */
+        for (uint32_t k = 0; k < l; k++) op[k] = cp[k];
+        op += l;
+    }
+    return (op == op_end && ip == end) ? int64_t(ulen) : -1;
+}
+
+__global__ void k_str_decode(const uint8_t *__restrict__ blob,
+                             const DevPage *__restrict__ pages, int n,
+                             const int64_t *__restrict__ scr_off,
+                             uint8_t *__restrict__ scratch,
+                             int64_t *__restrict__ sz,
+                             int64_t *__restrict__ pos,
+                             uint8_t *__restrict__ valid,
+                             unsigned *__restrict__ err) {
+    for (int p = blockIdx.x * blockDim.x + threadIdx.x; p < n;
+         p += gridDim.x * blockDim.x) {
+        DevPage pg = pages[p];
+        const uint8_t *s = blob + pg.data_off;
+        const uint8_t *bs = blob + pg.bitset_off;
+        const int64_t base = pg.row_off;
+        uint8_t *dst = scratch + scr_off[p];
+        for (uint32_t r = 0; r < pg.nrows; r++) {
+            sz[base + r] = 0;
+            pos[base + r] = 0;
+            if (valid) valid[base + r] = 0;
+        }
+        if (pg.data_len == 0) continue; /* empty src -> all null,
+                                           string.rs:231-236 */
+        int64_t pn;
+        const uint8_t *pl = dst;
+        bool be = false;
+        if (pg.enc == GS_ENC_SNAPPY) {
+            if (pg.data_len < 2) { atomicOr(err, DERR_FORMAT); continue; }
+            pn = dev_snappy_decompress(s + 2, pg.data_len - 2, dst);
+            if (pn < 0) { atomicOr(err, DERR_FORMAT); continue; }
+        } else if (pg.enc == GS_ENC_NULL) { /* string.rs:169-183 */
+            pn = int64_t(pg.data_len) - 1;
+            pl = s + 1;
+            be = true;
+        } else {
+            atomicOr(err, DERR_FORMAT);
+            continue;
+        }
+        uint64_t i = 0, w = 0;
+        for (uint32_t r = 0; r < pg.nrows; r++) {
+            int v = pg.all_valid ? 1 : dev_bit(bs, r);
+            if (!v) continue;
+            if (i >= uint64_t(pn)) break; /* payload exhausted: remaining
+                                             valid rows decode null */
+            uint64_t slen = 0;
+            if (be) {
+                if (i + 8 > uint64_t(pn)) { atomicOr(err, DERR_FORMAT); break; }
+                for (int k = 0; k < 8; k++) slen = (slen << 8) | pl[i + k];
+                i += 8;
+            } else {
+                int sh = 0, ok = 0;
+                while (i < uint64_t(pn)) {
+                    uint8_t b = pl[i++];
+                    slen |= uint64_t(b & 0x7f) << sh;
+                    sh += 7;
+                    if (!(b & 0x80)) { ok = 1; break; }
+                }
+                if (!ok) { atomicOr(err, DERR_FORMAT); break; }
+            }
+            if (i + slen > uint64_t(pn)) { atomicOr(err, DERR_FORMAT); break; }
+            if (be) { /* pack bytes into scratch so gather has one source */
+                for (uint64_t k = 0; k < slen; k++) dst[w + k] = pl[i + k];
+                pos[base + r] = scr_off[p] + int64_t(w);
+                w += slen;
+            } else {
+                pos[base + r] = scr_off[p] + int64_t(i);
+            }
+            sz[base + r] = int64_t(slen);
+            if (valid) valid[base + r] = 1;
+            i += slen;
+        }
+    }
+}
+
+__global__ void k_str_gather(const uint8_t *__restrict__ scratch,
+                             const int64_t *__restrict__ pos,
+                             const int64_t *__restrict__ sz,
+                             const int64_t *__restrict__ off, int64_t rows,
+                             uint8_t *__restrict__ out) {
+    for (int64_t r = blockIdx.x * int64_t(blockDim.x) + threadIdx.x; r < rows;
+         r += int64_t(gridDim.x) * blockDim.x) {
+        const int64_t n2 = sz[r];
+        const uint8_t *s = scratch + pos[r];
+        uint8_t *d = out + off[r];
+        for (int64_t k = 0; k < n2; k++) d[k] = s[k];
+    }
+}
+
 /* ------------------------------------------------- compaction merge (k-way)
  * Config #5: k overlapping L0 column groups per series -> one merged,
  * deduped stream (compact.rs:271-404, comapcting_block_meta_group.rs:
@@ -1987,6 +2145,12 @@ struct SlotPages {
     std::vector<DevPage> host[PC_NCLASS];
     DevPage *dev[PC_NCLASS] = {nullptr, nullptr, nullptr};
     int n[PC_NCLASS] = {0, 0, 0};
+    /* string pages only: per-PC_STR-page scratch offsets (host order of
+       host[PC_STR]) + total; caps parsed from the block headers at
+       upload (snappy preamble varint / uncompressed payload length) */
+    std::vector<int64_t> str_scr;
+    int64_t str_total = 0;
+    int64_t *d_str_scr = nullptr;
 };
 
 struct GsGroupSet {
@@ -2017,6 +2181,13 @@ struct GsGroupSet {
                                      row / RLE delta (k_spans_rle out) */
     int64_t *d_g_delta = nullptr;
     int max_span = 0; /* max page-groups per series-group */
+    /* string decode scratch (lazy, cached across gs_decode_str calls) */
+    uint8_t *d_str_scratch = nullptr;
+    size_t str_scratch_cap = 0;
+    int64_t *d_str_sz = nullptr;  /* per-row string byte length (0=null) */
+    int64_t *d_str_pos = nullptr; /* per-row start index into scratch */
+    int64_t *d_str_bsums = nullptr; /* offset-scan block sums over rows */
+    size_t str_bsums_cap = 0;
     GsTimeRange *d_ranges = nullptr;
     size_t ranges_cap = 0;
     uint8_t *d_valid = nullptr; /* lazily allocated internal validity bytes */
@@ -2248,7 +2419,27 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         int cls = PC_SEQ;
         bool int_ct = hp.dp.ctype == GS_CT_TIME || hp.dp.ctype == GS_CT_I64 ||
                       hp.dp.ctype == GS_CT_U64;
-        if (hp.dp.all_valid && hp.dp.sub == 2 && int_ct && hp.dp.data_len > 2) {
+        if (hp.dp.ctype == GS_CT_STR) {
+            cls = PC_STR;
+            /* scratch cap: decoded payload length (snappy preamble
+               varint, string.rs:197) or the raw payload (Null enc) */
+            int64_t cap = 0;
+            if (hp.dp.data_len >= 3 && hp.dp.enc == GS_ENC_SNAPPY) {
+                uint64_t v = 0;
+                int sh = 0;
+                for (uint32_t i = 2; i < hp.dp.data_len && i < 12; i++) {
+                    uint8_t b = hp.data_src[i];
+                    v |= uint64_t(b & 0x7f) << sh;
+                    sh += 7;
+                    if (!(b & 0x80)) break;
+                }
+                cap = int64_t(v);
+            } else if (hp.dp.data_len >= 1 && hp.dp.enc == GS_ENC_NULL) {
+                cap = int64_t(hp.dp.data_len) - 1;
+            }
+            sp.str_scr.push_back(sp.str_total);
+            sp.str_total += (cap + 15) & ~int64_t(15);
+        } else if (hp.dp.all_valid && hp.dp.sub == 2 && int_ct && hp.dp.data_len > 2) {
             if (hp.dp.enc == GS_ENC_DELTATS) cls = PC_RLE_TS;
             else if (hp.dp.enc == GS_ENC_DELTA) cls = PC_RLE_I64;
         } else if (hp.dp.all_valid && hp.dp.sub == 1 && int_ct &&
@@ -2286,6 +2477,15 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                                sp.n[k] * sizeof(DevPage),
                                hipMemcpyHostToDevice, ctx->stream);
             }
+        }
+        if (sp.n[PC_STR]) {
+            if (hipMalloc(&sp.d_str_scr, sp.n[PC_STR] * sizeof(int64_t)) != hipSuccess) {
+                fail(GS_ERR, "hipMalloc str scratch table failed");
+                gs_groups_free(set); return nullptr;
+            }
+            hipMemcpyAsync(sp.d_str_scr, sp.str_scr.data(),
+                           sp.n[PC_STR] * sizeof(int64_t),
+                           hipMemcpyHostToDevice, ctx->stream);
         }
     }
 
@@ -2354,9 +2554,15 @@ void gs_groups_free(GsGroupSet *set) {
     if (!set) return;
     hipSetDevice(set->ctx->device);
     hipStreamSynchronize(set->ctx->stream);
-    for (auto &sp : set->slots)
+    for (auto &sp : set->slots) {
         for (int k = 0; k < PC_NCLASS; k++)
             if (sp.dev[k]) hipFree(sp.dev[k]);
+        if (sp.d_str_scr) hipFree(sp.d_str_scr);
+    }
+    if (set->d_str_scratch) hipFree(set->d_str_scratch);
+    if (set->d_str_sz) hipFree(set->d_str_sz);
+    if (set->d_str_pos) hipFree(set->d_str_pos);
+    if (set->d_str_bsums) hipFree(set->d_str_bsums);
     hipFree(set->d_blob);
     hipFree(set->d_groups);
     if (set->d_sgroups) hipFree(set->d_sgroups);
@@ -2427,6 +2633,8 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
     HIP_TRY(hipSetDevice(ctx->device));
     SlotPages &sp = set->slots[col];
     uint8_t ct = sp.ctype;
+    if (sp.n[PC_STR])
+        return fail(GS_ERR, "string column: use gs_decode_str");
 
     if (sp.n[PC_SEQ]) {
         int n = sp.n[PC_SEQ];
@@ -2482,6 +2690,70 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
                            sp.dev[PC_RLE_I64], sp.n[PC_RLE_I64], (int64_t *)d_out,
                            d_valid, 0, ctx->d_err);
     HIP_TRY(hipStreamSynchronize(ctx->stream));
+    return check_dev_err(ctx);
+}
+
+GsStatus gs_decode_str(GsCtx *ctx, GsGroupSet *set, uint32_t col,
+                       int64_t *d_offsets, uint8_t *d_bytes,
+                       int64_t bytes_cap, uint8_t *d_valid,
+                       int64_t *total_bytes) {
+    if (!ctx || !set || col >= set->ncols || !d_offsets || !d_bytes ||
+        !total_bytes)
+        return fail(GS_ERR, "bad args to gs_decode_str");
+    SlotPages &sp = set->slots[col];
+    if (sp.ctype != GS_CT_STR || sp.n[PC_STR] != int(sp.host[PC_STR].size()))
+        return fail(GS_ERR, "column is not a string column");
+    if (set->total_rows > int64_t(SCAN_BLOCK) * SCAN_ITEMS * 2048)
+        return fail(GS_ERR, "string decode row limit exceeded (4.19M)");
+    HIP_TRY(hipSetDevice(ctx->device));
+    int n = sp.n[PC_STR];
+    if (!set->d_str_scratch || set->str_scratch_cap < size_t(sp.str_total)) {
+        if (set->d_str_scratch) hipFree(set->d_str_scratch);
+        size_t cap = sp.str_total > 0 ? size_t(sp.str_total) : 16;
+        if (hipMalloc(&set->d_str_scratch, cap) != hipSuccess)
+            return fail(GS_ERR, "hipMalloc string scratch failed");
+        set->str_scratch_cap = cap;
+    }
+    if (!set->d_str_sz) {
+        if (hipMalloc(&set->d_str_sz, set->total_rows * 8) != hipSuccess ||
+            hipMalloc(&set->d_str_pos, set->total_rows * 8) != hipSuccess)
+            return fail(GS_ERR, "hipMalloc string row tables failed");
+    }
+    int nblocks = int((set->total_rows + SCAN_BLOCK * SCAN_ITEMS - 1) /
+                      (SCAN_BLOCK * SCAN_ITEMS));
+    if (set->str_bsums_cap < size_t(nblocks) + 1) {
+        if (set->d_str_bsums) hipFree(set->d_str_bsums);
+        if (hipMalloc(&set->d_str_bsums, (size_t(nblocks) + 1) * 8) != hipSuccess)
+            return fail(GS_ERR, "hipMalloc string scan sums failed");
+        set->str_bsums_cap = size_t(nblocks) + 1;
+    }
+    hipLaunchKernelGGL(k_str_decode, dim3(grid_for(n, 256)), dim3(256), 0,
+                       ctx->stream, set->d_blob, sp.dev[PC_STR], n,
+                       sp.d_str_scr, set->d_str_scratch, set->d_str_sz,
+                       set->d_str_pos, d_valid, ctx->d_err);
+    hipLaunchKernelGGL(k_scan_partials, dim3(nblocks), dim3(SCAN_BLOCK), 0,
+                       ctx->stream, set->d_str_sz, int(set->total_rows),
+                       d_offsets, set->d_str_bsums);
+    hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(1), 0, ctx->stream,
+                       int(set->total_rows), d_offsets, set->d_str_bsums,
+                       nblocks);
+    hipLaunchKernelGGL(k_scan_add, dim3(grid_for(int(set->total_rows), 256)),
+                       dim3(256), 0, ctx->stream, int(set->total_rows),
+                       d_offsets, set->d_str_bsums);
+    int64_t total = 0;
+    HIP_TRY(hipMemcpyAsync(&total, d_offsets + set->total_rows, 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    GsStatus st = check_dev_err(ctx);
+    if (st != GS_OK) return st;
+    if (total > bytes_cap)
+        return fail(GS_ERR, "string bytes buffer too small");
+    hipLaunchKernelGGL(k_str_gather, dim3(grid_for(int(set->total_rows), 256)),
+                       dim3(256), 0, ctx->stream, set->d_str_scratch,
+                       set->d_str_pos, set->d_str_sz, d_offsets,
+                       set->total_rows, d_bytes);
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    *total_bytes = total;
     return check_dev_err(ctx);
 }
 

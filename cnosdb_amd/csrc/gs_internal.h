@@ -35,7 +35,9 @@ enum PageClass {
     PC_S8B = 4,     /* DeltaTs/Delta + simple8b + all-valid: block-parallel */
     PC_RAW = 5,     /* Null encoding (raw BE) + all-valid: coalesced bswap */
     PC_BOOL = 6,    /* BitPack + all-valid: parallel bit extract */
-    PC_NCLASS = 7,
+    PC_STR = 7,     /* string blocks (snappy / uncompressed): sequential
+                       thread-per-page decompress, see gs_decode_str */
+    PC_NCLASS = 8,
 };
 
 extern "C" {

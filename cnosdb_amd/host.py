@@ -11,7 +11,7 @@ import os
 
 import numpy as np
 
-CT_TIME, CT_I64, CT_F64, CT_BOOL, CT_U64 = 0, 1, 2, 3, 4
+CT_TIME, CT_I64, CT_F64, CT_BOOL, CT_U64, CT_STR = 0, 1, 2, 3, 4, 5
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 
@@ -119,8 +119,13 @@ class PageLib:
         lib = ctypes.CDLL(path)
         self.lib = lib
         for nm in ("gs_encode_ts", "gs_encode_i64", "gs_encode_f64",
-                   "gs_encode_bool", "gs_build_page"):
+                   "gs_encode_bool", "gs_encode_str", "gs_build_page"):
             getattr(lib, nm).restype = ctypes.c_int64
+        lib.gs_decode_str.restype = ctypes.c_int32
+        lib.gs_decode_str.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint32,
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_int64)]
         lib.gs_crc32.restype = ctypes.c_uint32
         lib.gs_version.restype = ctypes.c_char_p
         lib.gs_last_error.restype = ctypes.c_char_p
@@ -225,6 +230,37 @@ def encode_bool(values):
     if n < 0:
         raise RuntimeError(f"gs_encode_bool failed: {n}")
     return buf[:n].tobytes()
+
+
+def encode_str(strings):
+    """Snappy string block (codec/string.rs:32-88): strings = list of
+    bytes (non-null values only)."""
+    lib = PageLib().lib
+    concat = b"".join(strings)
+    lens = np.array([len(s) for s in strings], dtype=np.uint64)
+    src = np.frombuffer(concat, dtype=np.uint8) if concat else np.zeros(1, np.uint8)
+    cap = 2 + 32 + len(concat) + len(concat) // 6 + len(strings) * 10 + 64
+    buf = np.zeros(cap, dtype=np.uint8)
+    n = lib.gs_encode_str(_np_ptr(src), _np_ptr(lens), len(strings),
+                          _np_ptr(buf), buf.size)
+    if n < 0:
+        raise RuntimeError(f"gs_encode_str failed: {n}")
+    return buf[:n].tobytes()
+
+
+def str_page_of(strings, valid=None):
+    """Build a string column page. strings = list of bytes per row;
+    valid = bool array or None; null rows' strings are not encoded
+    (page.rs/A.2 semantics)."""
+    nrows = len(strings)
+    if valid is None:
+        present = strings
+        bitset = None
+    else:
+        valid = np.asarray(valid, dtype=bool)
+        present = [s for s, v in zip(strings, valid) if v]
+        bitset = np.packbits(valid, bitorder="little")
+    return build_page(encode_str(present), nrows, bitset)
 
 
 def build_page(data, nrows, bitset=None):
@@ -399,6 +435,22 @@ class Engine:
                                 ctypes.c_void_p(d_valid.data_ptr()) if d_valid is not None else None)
         if st != 0:
             raise RuntimeError(f"gs_decode failed ({st}): {self._pl.err()}")
+
+    def decode_str(self, gset, col, d_offsets, d_bytes, d_valid=None):
+        """String column decode to Arrow varbinary layout
+        (str_snappy_decode_to_array, string.rs:226-276): d_offsets int64
+        CUDA tensor of rows+1, d_bytes uint8 CUDA tensor (capacity);
+        returns total payload bytes written."""
+        total = ctypes.c_int64(0)
+        st = self.lib.gs_decode_str(
+            self._ctx, gset._h, col,
+            ctypes.c_void_p(d_offsets.data_ptr()),
+            ctypes.c_void_p(d_bytes.data_ptr()), d_bytes.numel(),
+            ctypes.c_void_p(d_valid.data_ptr()) if d_valid is not None else None,
+            ctypes.byref(total))
+        if st != 0:
+            raise RuntimeError(f"gs_decode_str failed ({st}): {self._pl.err()}")
+        return total.value
 
     def apply_tombstone(self, gset, d_ts, d_valid, ranges):
         arr = (GsTimeRange * len(ranges))(*[GsTimeRange(a, b) for a, b in ranges])

@@ -45,6 +45,7 @@ enum {
     GS_ENC_NULL = 1,
     GS_ENC_DELTA = 2,    /* i64: zigzag delta + simple8b/RLE  (codec/integer.rs) */
     GS_ENC_GORILLA = 6,  /* f64 XOR                           (codec/float.rs) */
+    GS_ENC_SNAPPY = 7,   /* strings: snappy block             (codec/string.rs) */
     GS_ENC_BITPACK = 10, /* bool                              (codec/boolean.rs) */
     GS_ENC_DELTATS = 11, /* ts: delta + scaled simple8b/RLE   (codec/timestamp.rs) */
 };
@@ -56,6 +57,7 @@ enum {
     GS_CT_F64 = 2,
     GS_CT_BOOL = 3,
     GS_CT_U64 = 4, /* bit-cast to i64, unsigned.rs:20-45 */
+    GS_CT_STR = 5, /* snappy string blocks, codec/string.rs */
 };
 
 /* Closed time interval (common/models/src/predicate/domain.rs:36-44) */
@@ -100,6 +102,14 @@ int64_t gs_encode_ts(const int64_t *src, size_t n, uint8_t *dst, size_t cap);
 int64_t gs_encode_i64(const int64_t *src, size_t n, uint8_t *dst, size_t cap);
 int64_t gs_encode_f64(const double *src, size_t n, uint8_t *dst, size_t cap);
 int64_t gs_encode_bool(const uint8_t *src, size_t n, uint8_t *dst, size_t cap);
+/* string block: src = concatenated string bytes, lens[i] = byte length of
+ * string i (non-null values only).  Produces [7][0x10][snappy raw] per
+ * str_snappy_encode (codec/string.rs:32-88); empty input -> 0 bytes.
+ * The snappy arithmetic restates the published algorithm of the
+ * un-vendored `snap` crate v1.1.1, pinned by the reference's golden
+ * vectors (string.rs:529-566). */
+int64_t gs_encode_str(const uint8_t *src, const uint64_t *lens, int64_t nstr,
+                      uint8_t *dst, size_t cap);
 /* assemble a full page (header + crc + bitset + data), tsm/page.rs:488-497 */
 int64_t gs_build_page(const uint8_t *bitset, int64_t nrows, const uint8_t *data,
                       size_t data_len, uint8_t *dst, size_t cap);
@@ -131,6 +141,24 @@ GsStatus gs_set_row_offsets(const GsGroupSet *set, int64_t *out);
  * 1 for valid rows, 0 for null rows. */
 GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
                    uint8_t *d_valid);
+
+/* ---- string column decode (str_snappy_decode_to_array,
+ * codec/string.rs:226-276 via data_buf_to_arrow_array,
+ * tsm/reader.rs:658-731) ----
+ * Decodes string column slot `col` of every group into Arrow
+ * varbinary layout at the set's row offsets: d_offsets (device,
+ * int64[rows+1], exclusive prefix of byte lengths; null rows
+ * contribute 0) and d_bytes (device, capacity bytes_cap).  d_valid
+ * (device, 1 B/row, may be NULL) gets the validity bytes.  On return
+ * *total_bytes = d_offsets[rows].  Rows whose page payload ends early
+ * decode as null (the reference's builder stops appending).  Handles
+ * GS_ENC_SNAPPY and GS_ENC_NULL blocks; empty data region -> all rows
+ * null.  Current limit: rows <= 4,194,304 per set for the device
+ * offset scan. */
+GsStatus gs_decode_str(GsCtx *ctx, GsGroupSet *set, uint32_t col,
+                       int64_t *d_offsets, uint8_t *d_bytes,
+                       int64_t bytes_cap, uint8_t *d_valid,
+                       int64_t *total_bytes);
 
 /* ---- tombstone masking (tsm/reader.rs:634-656) ----
  * Clears validity (in d_valid) for rows whose decoded timestamp (d_ts,

@@ -122,3 +122,28 @@ def test_no_gpu_fails_loudly():
         pytest.skip("GPU present")
     with pytest.raises(RuntimeError):
         gs.Engine(0)
+
+
+def test_str_encoder_matches_oracle_and_golden():
+    """gs_encode_str must reproduce the snap-1.1.1 output pinned by the
+    reference's golden vectors (string.rs:529-566) and stay byte-equal
+    to the oracle restatement on randomized corpora."""
+    import json
+    import os
+    g = json.load(open(os.path.join(os.path.dirname(__file__), "golden",
+                                    "golden_vectors.json")))["str_snappy"]
+    for name in ("single", "multi_compressed", "unicode"):
+        strs = [s.encode() for s in g[name]["strings"]]
+        blk = gs.encode_str(strs)
+        assert blk[0] == 7 and list(blk[1:]) == g[name]["bytes"], name
+    assert list(gs.encode_str([b"\xC0"])[1:]) == g["invalid_utf8"]["bytes"]
+    assert gs.encode_str([]) == b""
+    r = np.random.default_rng(17)
+    tags = [b"region=cn-%02d" % i for i in range(16)]
+    for t in range(12):
+        n = int(r.integers(1, 4000))
+        strs = [tags[int(r.integers(0, 16))] if r.random() < 0.5
+                else bytes(r.integers(0, 256, int(r.integers(0, 70))).astype(np.uint8))
+                for _ in range(n)]
+        assert gs.encode_str(strs) == orc.encode_str(strs), t
+        assert orc.decode_str(gs.encode_str(strs), n) == strs, t
