@@ -364,9 +364,118 @@ def bench_compact(args):
     eng.close()
 
 
+def bench_strings(args):
+    """string/tag column decode (snappy blocks ->
+    Arrow varbinary layout).  Secondary benchmark line; the driver's
+    headline is the scan mode."""
+    import torch
+    import cnosdb_amd as gs
+    rng = np.random.default_rng(231)
+    page_rows = args.page_rows
+    npages = args.series if args.series != 10000 else 4000
+    per_set = min(1000, npages)
+    uniq = min(args.unique, 256)
+    tagpool = [b"hostname=host_%04d,region=region_%02d,rack=%02d"
+               % (i, i % 16, i % 64) for i in range(100)]
+    eng = gs.Engine(0)
+    t_setup = time.perf_counter()
+    tpage = gs.page_of(T0 + np.arange(page_rows, dtype=np.int64) * NS,
+                       gs.CT_TIME)
+    blocks, str_pages, payloads = [], [], []
+    for u in range(uniq):
+        idx = rng.integers(0, len(tagpool), page_rows)
+        strs = [tagpool[i] for i in idx]
+        blk = gs.encode_str(strs)
+        blocks.append(blk)
+        payloads.append(sum(len(s) for s in strs))
+        str_pages.append(gs.build_page(blk, page_rows))
+    sets = []
+    total_rows = 0
+    total_payload = 0
+    nsets = (npages + per_set - 1) // per_set
+    for s in range(nsets):
+        cnt = min(per_set, npages - s * per_set)
+        groups = [(g, [(tpage, gs.CT_TIME),
+                       (str_pages[(s * per_set + g) % uniq], gs.CT_STR)])
+                  for g in range(cnt)]
+        gset = eng.upload(groups, validate_crc=False)
+        rows = gset.rows
+        total_rows += rows
+        total_payload += sum(payloads[(s * per_set + g) % uniq]
+                             for g in range(cnt))
+        sets.append((gset,
+                     torch.zeros(rows + 1, dtype=torch.int64, device="cuda"),
+                     torch.zeros(rows * 64, dtype=torch.uint8, device="cuda")))
+    setup_s = time.perf_counter() - t_setup
+
+    def step():
+        t = 0
+        for gset, d_off, d_bytes in sets:
+            t += eng.decode_str(gset, 1, d_off, d_bytes)
+        return t
+
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        got = step()
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    assert got == total_payload
+    value = total_rows * args.steps / dt
+
+    cpu_baseline = None
+    if not args.skip_cpu_baseline:
+        import ctypes
+        from oracle import pyoracle as orc
+        o = orc.Oracle().lib
+        o.orc_str_decode_pages_omp.restype = ctypes.c_int64
+        cores = _host_threads()
+        sample_pages = min(2048, npages)
+        bufs = [np.frombuffer(blocks[p % uniq], dtype=np.uint8)
+                for p in range(sample_pages)]
+        ptrs = (ctypes.c_void_p * sample_pages)(*[b.ctypes.data for b in bufs])
+        lens_c = (ctypes.c_size_t * sample_pages)(
+            *[b.size for b in bufs])
+        tcb = time.perf_counter()
+        w = o.orc_str_decode_pages_omp(ptrs, lens_c, sample_pages, page_rows,
+                                       max(payloads) + 4096, cores)
+        dt_cb = time.perf_counter() - tcb
+        assert w > 0
+        cpu_baseline = {"value": sample_pages * page_rows / dt_cb,
+                        "unit": "strings/s", "cores": cores, "kind": "port",
+                        "sample": f"{sample_pages} pages x {page_rows} rows "
+                                  "(oracle C snappy decode, OpenMP)"}
+    print(json.dumps({
+        "metric": "string/tag column decode strings/sec (snappy blocks -> Arrow varbinary)",
+        "value": value,
+        "unit": "strings/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1000,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "u8",
+        "data": "synthetic",
+        "config": {"workload": "tag-column decode (SURVEY §8f row 1)",
+                   "pages": npages, "page_rows": page_rows,
+                   "tag_cardinality": len(tagpool),
+                   "payload_bytes_per_step": int(total_payload),
+                   "payload_GBps": total_payload * args.steps / dt / 1e9,
+                   "setup_s": round(setup_s, 1)},
+        "cpu_baseline": cpu_baseline,
+    }))
+    for gset, _, _ in sets:
+        gset.free()
+    eng.close()
+
+
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--mode", choices=["scan", "compact"], default="scan")
+    ap.add_argument("--mode", choices=["scan", "compact", "strings"], default="scan")
     ap.add_argument("--gpus", type=int, default=1,
                     help="driver contract flag; the actual world size comes "
                          "from the torchrun environment (WORLD_SIZE)")
@@ -388,6 +497,9 @@ def main():
         if args.npts == 1_000_000:
             args.npts = 100_000
         bench_compact(args)
+        return
+    if args.mode == "strings":
+        bench_strings(args)
         return
 
     import torch

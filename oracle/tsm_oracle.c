@@ -1212,3 +1212,36 @@ ORC_EXPORT int64_t orc_str_decode(const uint8_t *src, size_t len,
     if (payload) free(payload);
     return rc < 0 ? -1 : (int64_t)w;
 }
+
+/* cpu_baseline leg for the strings bench: decode many string blocks in
+ * parallel (mirrors decode_pages fan-out over ColumnGroups,
+ * tskv/src/reader/column_group/mod.rs:202-243).  blocks[i]/lens[i] =
+ * page data region; all pages all-valid with nrows rows.  Returns total
+ * payload bytes or -1. */
+ORC_EXPORT int64_t orc_str_decode_pages_omp(const uint8_t **blocks,
+                                            const size_t *lens, int npages,
+                                            int64_t nrows, int64_t cap_per_page,
+                                            int nthreads) {
+    int64_t total = 0;
+    int err = 0;
+#pragma omp parallel num_threads(nthreads)
+    {
+        uint8_t *bytes_buf = (uint8_t *)malloc((size_t)cap_per_page);
+        int64_t *lens_buf = (int64_t *)malloc((size_t)nrows * 8);
+        int64_t mine = 0;
+#pragma omp for schedule(dynamic, 4)
+        for (int p = 0; p < npages; p++) {
+            if (!bytes_buf || !lens_buf) { err = 1; continue; }
+            int64_t w = orc_str_decode(blocks[p], lens[p], NULL, nrows,
+                                       bytes_buf, (size_t)cap_per_page,
+                                       lens_buf);
+            if (w < 0) err = 1;
+            else mine += w;
+        }
+#pragma omp atomic
+        total += mine;
+        free(bytes_buf);
+        free(lens_buf);
+    }
+    return err ? -1 : total;
+}
