@@ -1318,22 +1318,23 @@ __global__ void k_str_decode(const uint8_t *__restrict__ blob,
         }
         if (pg.data_len == 0) continue; /* empty src -> all null,
                                            string.rs:231-236 */
+        /* normalize: the payload lands in this page's scratch slab for
+           both encodings, so the walk and the gather read one source */
         int64_t pn;
-        const uint8_t *pl = dst;
-        bool be = false;
+        int be = 0;
         if (pg.enc == GS_ENC_SNAPPY) {
             if (pg.data_len < 2) { atomicOr(err, DERR_FORMAT); continue; }
             pn = dev_snappy_decompress(s + 2, pg.data_len - 2, dst);
             if (pn < 0) { atomicOr(err, DERR_FORMAT); continue; }
         } else if (pg.enc == GS_ENC_NULL) { /* string.rs:169-183 */
             pn = int64_t(pg.data_len) - 1;
-            pl = s + 1;
-            be = true;
+            for (int64_t k = 0; k < pn; k++) dst[k] = s[1 + k];
+            be = 1;
         } else {
             atomicOr(err, DERR_FORMAT);
             continue;
         }
-        uint64_t i = 0, w = 0;
+        uint64_t i = 0;
         for (uint32_t r = 0; r < pg.nrows; r++) {
             int v = pg.all_valid ? 1 : dev_bit(bs, r);
             if (!v) continue;
@@ -1342,12 +1343,12 @@ __global__ void k_str_decode(const uint8_t *__restrict__ blob,
             uint64_t slen = 0;
             if (be) {
                 if (i + 8 > uint64_t(pn)) { atomicOr(err, DERR_FORMAT); break; }
-                for (int k = 0; k < 8; k++) slen = (slen << 8) | pl[i + k];
+                for (int k = 0; k < 8; k++) slen = (slen << 8) | dst[i + k];
                 i += 8;
             } else {
                 int sh = 0, ok = 0;
                 while (i < uint64_t(pn)) {
-                    uint8_t b = pl[i++];
+                    uint8_t b = dst[i++];
                     slen |= uint64_t(b & 0x7f) << sh;
                     sh += 7;
                     if (!(b & 0x80)) { ok = 1; break; }
@@ -1355,13 +1356,7 @@ __global__ void k_str_decode(const uint8_t *__restrict__ blob,
                 if (!ok) { atomicOr(err, DERR_FORMAT); break; }
             }
             if (i + slen > uint64_t(pn)) { atomicOr(err, DERR_FORMAT); break; }
-            if (be) { /* pack bytes into scratch so gather has one source */
-                for (uint64_t k = 0; k < slen; k++) dst[w + k] = pl[i + k];
-                pos[base + r] = scr_off[p] + int64_t(w);
-                w += slen;
-            } else {
-                pos[base + r] = scr_off[p] + int64_t(i);
-            }
+            pos[base + r] = scr_off[p] + int64_t(i);
             sz[base + r] = int64_t(slen);
             if (valid) valid[base + r] = 1;
             i += slen;
