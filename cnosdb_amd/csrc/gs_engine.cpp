@@ -1234,16 +1234,9 @@ __global__ void k_agg_partial_rle(const DevGroup *__restrict__ sg, int nsg,
  * (pos,len) per row; a device scan of the lengths builds the Arrow
  * offsets and a gather packs the bytes. */
 
-/* wave-cooperative snappy decompress: all 64 lanes parse the tag stream
- * in lockstep (uniform loads, no divergence — every value derives from
- * the same bytes), and each op's data movement is done by all lanes in
- * parallel.  Overlapping copies (offset < length) use the period trick:
- * byte k of the copy equals the committed byte at offset (k mod off) - off
- * relative to the op start, so lanes never read what this op writes. */
-__device__ int64_t wave_snappy_decompress(const uint8_t *__restrict__ src,
-                                          uint32_t len,
-                                          uint8_t *__restrict__ dst,
-                                          int lane) {
+__device__ int64_t dev_snappy_decompress(const uint8_t *__restrict__ src,
+                                         uint32_t len,
+                                         uint8_t *__restrict__ dst) {
     uint64_t ulen = 0;
     uint32_t i = 0;
     int sh = 0, ok = 0;
@@ -1271,7 +1264,7 @@ __device__ int64_t wave_snappy_decompress(const uint8_t *__restrict__ src,
             }
             l += 1;
             if (ip + l > end || op + l > op_end) return -1;
-            for (uint32_t k = lane; k < l; k += 64) op[k] = ip[k];
+            for (uint32_t k = 0; k < l; k++) op[k] = ip[k];
             ip += l;
             op += l;
             continue;
@@ -1296,17 +1289,13 @@ __device__ int64_t wave_snappy_decompress(const uint8_t *__restrict__ src,
             break;
         }
         if (off == 0 || uint64_t(op - dst) < off || op + l > op_end) return -1;
-        for (uint32_t k = lane; k < l; k += 64)
-            op[k] = op[int64_t(k % off) - int64_t(off)];
+        const uint8_t *cp = op - off; /* may overlap: strictly in order */
+        for (uint32_t k = 0; k < l; k++) op[k] = cp[k];
         op += l;
     }
     return (op == op_end && ip == end) ? int64_t(ulen) : -1;
 }
 
-/* One WAVE per page: the snappy/payload byte movement is done by all 64
- * lanes cooperatively (wave_snappy_decompress); the tag stream and the
- * per-row length walk are parsed by all lanes redundantly in lockstep
- * (uniform control flow), with row-table stores predicated to lane 0. */
 __global__ void k_str_decode(const uint8_t *__restrict__ blob,
                              const DevPage *__restrict__ pages, int n,
                              const int64_t *__restrict__ scr_off,
@@ -1315,16 +1304,14 @@ __global__ void k_str_decode(const uint8_t *__restrict__ blob,
                              int64_t *__restrict__ pos,
                              uint8_t *__restrict__ valid,
                              unsigned *__restrict__ err) {
-    const int lane = threadIdx.x & 63;
-    const int wid = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-    const int nw = (gridDim.x * blockDim.x) >> 6;
-    for (int p = wid; p < n; p += nw) {
+    for (int p = blockIdx.x * blockDim.x + threadIdx.x; p < n;
+         p += gridDim.x * blockDim.x) {
         DevPage pg = pages[p];
         const uint8_t *s = blob + pg.data_off;
         const uint8_t *bs = blob + pg.bitset_off;
         const int64_t base = pg.row_off;
         uint8_t *dst = scratch + scr_off[p];
-        for (uint32_t r = lane; r < pg.nrows; r += 64) {
+        for (uint32_t r = 0; r < pg.nrows; r++) {
             sz[base + r] = 0;
             pos[base + r] = 0;
             if (valid) valid[base + r] = 0;
@@ -1336,21 +1323,15 @@ __global__ void k_str_decode(const uint8_t *__restrict__ blob,
         int64_t pn;
         int be = 0;
         if (pg.enc == GS_ENC_SNAPPY) {
-            if (pg.data_len < 2) {
-                if (lane == 0) atomicOr(err, DERR_FORMAT);
-                continue;
-            }
-            pn = wave_snappy_decompress(s + 2, pg.data_len - 2, dst, lane);
-            if (pn < 0) {
-                if (lane == 0) atomicOr(err, DERR_FORMAT);
-                continue;
-            }
+            if (pg.data_len < 2) { atomicOr(err, DERR_FORMAT); continue; }
+            pn = dev_snappy_decompress(s + 2, pg.data_len - 2, dst);
+            if (pn < 0) { atomicOr(err, DERR_FORMAT); continue; }
         } else if (pg.enc == GS_ENC_NULL) { /* string.rs:169-183 */
             pn = int64_t(pg.data_len) - 1;
-            for (int64_t k = lane; k < pn; k += 64) dst[k] = s[1 + k];
+            for (int64_t k = 0; k < pn; k++) dst[k] = s[1 + k];
             be = 1;
         } else {
-            if (lane == 0) atomicOr(err, DERR_FORMAT);
+            atomicOr(err, DERR_FORMAT);
             continue;
         }
         uint64_t i = 0;
@@ -1361,10 +1342,7 @@ __global__ void k_str_decode(const uint8_t *__restrict__ blob,
                                              valid rows decode null */
             uint64_t slen = 0;
             if (be) {
-                if (i + 8 > uint64_t(pn)) {
-                    if (lane == 0) atomicOr(err, DERR_FORMAT);
-                    break;
-                }
+                if (i + 8 > uint64_t(pn)) { atomicOr(err, DERR_FORMAT); break; }
                 for (int k = 0; k < 8; k++) slen = (slen << 8) | dst[i + k];
                 i += 8;
             } else {
@@ -1375,20 +1353,12 @@ __global__ void k_str_decode(const uint8_t *__restrict__ blob,
                     sh += 7;
                     if (!(b & 0x80)) { ok = 1; break; }
                 }
-                if (!ok) {
-                    if (lane == 0) atomicOr(err, DERR_FORMAT);
-                    break;
-                }
+                if (!ok) { atomicOr(err, DERR_FORMAT); break; }
             }
-            if (i + slen > uint64_t(pn)) {
-                if (lane == 0) atomicOr(err, DERR_FORMAT);
-                break;
-            }
-            if (lane == 0) {
-                pos[base + r] = scr_off[p] + int64_t(i);
-                sz[base + r] = int64_t(slen);
-                if (valid) valid[base + r] = 1;
-            }
+            if (i + slen > uint64_t(pn)) { atomicOr(err, DERR_FORMAT); break; }
+            pos[base + r] = scr_off[p] + int64_t(i);
+            sz[base + r] = int64_t(slen);
+            if (valid) valid[base + r] = 1;
             i += slen;
         }
     }
