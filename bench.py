@@ -372,8 +372,8 @@ def bench_strings(args):
     import cnosdb_amd as gs
     rng = np.random.default_rng(231)
     page_rows = args.page_rows
-    npages = args.series if args.series != 10000 else 4000
-    per_set = min(1000, npages)
+    npages = args.series if args.series != 10000 else 24000
+    per_set = min(npages, 30000)  # 134M-row scan limit at 4000 rows/page
     uniq = min(args.unique, 256)
     tagpool = [b"hostname=host_%04d,region=region_%02d,rack=%02d"
                % (i, i % 16, i % 64) for i in range(100)]
@@ -389,20 +389,35 @@ def bench_strings(args):
         blocks.append(blk)
         payloads.append(sum(len(s) for s in strs))
         str_pages.append(gs.build_page(blk, page_rows))
+    # one contiguous buffer: [ts page][uniq str pages]; page offsets repeat
+    buf = np.frombuffer(b"".join([tpage] + str_pages), dtype=np.uint8)
+    str_off = np.zeros(uniq, dtype=np.int64)
+    str_len = np.array([len(p) for p in str_pages], dtype=np.int64)
+    str_off[0] = len(tpage)
+    np.cumsum(str_len[:-1], out=str_off[1:])
+    str_off[1:] += len(tpage)
     sets = []
     total_rows = 0
     total_payload = 0
     nsets = (npages + per_set - 1) // per_set
+    pay = np.array(payloads, dtype=np.int64)
     for s in range(nsets):
         cnt = min(per_set, npages - s * per_set)
-        groups = [(g, [(tpage, gs.CT_TIME),
-                       (str_pages[(s * per_set + g) % uniq], gs.CT_STR)])
-                  for g in range(cnt)]
-        gset = eng.upload(groups, validate_crc=False)
+        u_idx = (np.arange(cnt, dtype=np.int64) + s * per_set) % uniq
+        page_off = np.empty(cnt * 2, dtype=np.int64)
+        page_len = np.empty(cnt * 2, dtype=np.int64)
+        page_off[0::2] = 0
+        page_len[0::2] = len(tpage)
+        page_off[1::2] = str_off[u_idx]
+        page_len[1::2] = str_len[u_idx]
+        nv = np.full(cnt * 2, page_rows, dtype=np.int64)
+        ct = np.tile(np.array([gs.CT_TIME, gs.CT_STR], dtype=np.uint8), cnt)
+        gset = eng.upload_packed(buf, page_off, page_len, nv, ct,
+                                 np.arange(cnt, dtype=np.int64), 2,
+                                 validate_crc=False)
         rows = gset.rows
         total_rows += rows
-        total_payload += sum(payloads[(s * per_set + g) % uniq]
-                             for g in range(cnt))
+        total_payload += int(pay[u_idx].sum())
         sets.append((gset,
                      torch.zeros(rows + 1, dtype=torch.int64, device="cuda"),
                      torch.zeros(rows * 64, dtype=torch.uint8, device="cuda")))

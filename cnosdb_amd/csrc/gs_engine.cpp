@@ -1581,16 +1581,25 @@ __global__ void k_scan_partials(const int64_t *__restrict__ in, int n,
 
 __global__ void k_scan_fixup(int n, int64_t *__restrict__ out,
                              int64_t *__restrict__ blocksums, int nblocks) {
-    /* single thread: scan the (<=2048) block sums in place + write total */
-    if (blockIdx.x == 0 && threadIdx.x == 0) {
-        int64_t acc = 0;
-        for (int i = 0; i < nblocks; i++) {
-            int64_t v = blocksums[i];
-            blocksums[i] = acc;
-            acc += v;
+    /* one wave: shuffle-scan the block sums in 64-wide tiles with a
+       running carry (exact integer prefix — order fixed).  Handles the
+       large-row string decode scans (tens of thousands of blocks)
+       without a serial single-thread walk. */
+    if (blockIdx.x != 0 || threadIdx.x >= 64) return;
+    const int lane = threadIdx.x;
+    int64_t carry = 0;
+    for (int t = 0; t < nblocks; t += 64) {
+        int i = t + lane;
+        int64_t v = (i < nblocks) ? blocksums[i] : 0;
+        int64_t x = v;
+        for (int off = 1; off < 64; off <<= 1) {
+            int64_t u = __shfl_up(x, off, 64);
+            if (lane >= off) x += u;
         }
-        out[n] = acc; /* total */
+        if (i < nblocks) blocksums[i] = carry + x - v; /* exclusive */
+        carry += __shfl(x, 63, 64);
     }
+    if (lane == 0) out[n] = carry; /* total */
 }
 
 __global__ void k_scan_add(int n, int64_t *__restrict__ out,
@@ -2698,8 +2707,8 @@ GsStatus gs_decode_str(GsCtx *ctx, GsGroupSet *set, uint32_t col,
     SlotPages &sp = set->slots[col];
     if (sp.ctype != GS_CT_STR || sp.n[PC_STR] != int(sp.host[PC_STR].size()))
         return fail(GS_ERR, "column is not a string column");
-    if (set->total_rows > int64_t(SCAN_BLOCK) * SCAN_ITEMS * 2048)
-        return fail(GS_ERR, "string decode row limit exceeded (4.19M)");
+    if (set->total_rows > int64_t(SCAN_BLOCK) * SCAN_ITEMS * 65535)
+        return fail(GS_ERR, "string decode row limit exceeded (134M)");
     HIP_TRY(hipSetDevice(ctx->device));
     int n = sp.n[PC_STR];
     if (!set->d_str_scratch || set->str_scratch_cap < size_t(sp.str_total)) {
@@ -2729,7 +2738,7 @@ GsStatus gs_decode_str(GsCtx *ctx, GsGroupSet *set, uint32_t col,
     hipLaunchKernelGGL(k_scan_partials, dim3(nblocks), dim3(SCAN_BLOCK), 0,
                        ctx->stream, set->d_str_sz, int(set->total_rows),
                        d_offsets, set->d_str_bsums);
-    hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(1), 0, ctx->stream,
+    hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(64), 0, ctx->stream,
                        int(set->total_rows), d_offsets, set->d_str_bsums,
                        nblocks);
     hipLaunchKernelGGL(k_scan_add, dim3(grid_for(int(set->total_rows), 256)),
@@ -2807,7 +2816,7 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
     hipLaunchKernelGGL(k_scan_partials, dim3(nblocks), dim3(SCAN_BLOCK), 0,
                        ctx->stream, set->d_sp_cnt, ng, set->d_out_off,
                        set->d_blocksums);
-    hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(1), 0, ctx->stream, ng,
+    hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(64), 0, ctx->stream, ng,
                        set->d_out_off, set->d_blocksums, nblocks);
     hipLaunchKernelGGL(k_scan_add, dim3(grid_for(ng, 256)), dim3(256), 0,
                        ctx->stream, ng, set->d_out_off, set->d_blocksums);

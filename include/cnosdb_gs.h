@@ -153,8 +153,8 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
  * *total_bytes = d_offsets[rows].  Rows whose page payload ends early
  * decode as null (the reference's builder stops appending).  Handles
  * GS_ENC_SNAPPY and GS_ENC_NULL blocks; empty data region -> all rows
- * null.  Current limit: rows <= 4,194,304 per set for the device
- * offset scan. */
+ * null.  Current limit: rows <= 134M per set for the device offset
+ * scan. */
 GsStatus gs_decode_str(GsCtx *ctx, GsGroupSet *set, uint32_t col,
                        int64_t *d_offsets, uint8_t *d_bytes,
                        int64_t bytes_cap, uint8_t *d_valid,
