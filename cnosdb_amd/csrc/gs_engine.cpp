@@ -2930,15 +2930,88 @@ GsStatus gs_scan_wait(GsCtx *ctx, GsGroupSet *set, GsScanResult *result) {
     return scan_fused_wait(ctx, set, result);
 }
 
+/* Raw row set: series structure over caller-resident DEVICE arrays with
+ * no pages — the memcache leg (MemCacheReader rows,
+ * mem_cache/series_data.rs:15-27).  gs_scan over such a set skips the
+ * decode phases and filters/aggregates spec->d_ts / spec->d_val
+ * directly; rows are taken as non-null (pre-filter nulls when building
+ * the arrays).  counts[i] = rows of series i (time-sorted). */
+GsGroupSet *gs_raw_set(GsCtx *ctx, const int64_t *counts, int64_t nseries) {
+    if (!ctx || !counts || nseries <= 0) {
+        fail(GS_ERR, "bad args to gs_raw_set");
+        return nullptr;
+    }
+    if (hipSetDevice(ctx->device) != hipSuccess) {
+        fail(GS_ERR, "hipSetDevice failed");
+        return nullptr;
+    }
+    GsGroupSet *set = new GsGroupSet();
+    set->ctx = ctx;
+    set->ngroups = size_t(nseries);
+    set->ncols = 0;
+    set->row_offsets.resize(nseries);
+    std::vector<DevGroup> hg(nseries);
+    int64_t rows = 0;
+    for (int64_t i = 0; i < nseries; i++) {
+        if (counts[i] < 0 || counts[i] > INT32_MAX) {
+            fail(GS_ERR, "raw set series row count out of range");
+            gs_groups_free(set);
+            return nullptr;
+        }
+        set->row_offsets[i] = rows;
+        hg[i].row_off = rows;
+        hg[i].nrows = int32_t(counts[i]);
+        hg[i].pad = 0;
+        rows += counts[i];
+    }
+    set->total_rows = rows;
+    set->nsgroups = int(nseries);
+    set->sgroup_span.assign(nseries, 1);
+    set->max_span = 1;
+    std::vector<int32_t> sgfirst(nseries + 1);
+    for (int64_t i = 0; i <= nseries; i++) sgfirst[i] = int32_t(i);
+    size_t nblocks = (size_t(nseries) + SCAN_BLOCK * SCAN_ITEMS - 1) /
+                     (SCAN_BLOCK * SCAN_ITEMS);
+    if (hipMalloc(&set->d_groups, nseries * sizeof(DevGroup)) != hipSuccess ||
+        hipMalloc(&set->d_sgroups, nseries * sizeof(DevGroup)) != hipSuccess ||
+        hipMalloc(&set->d_sgroups_out, nseries * sizeof(DevGroup)) != hipSuccess ||
+        hipMalloc(&set->d_sgroup_first, sgfirst.size() * sizeof(int32_t)) != hipSuccess ||
+        hipMalloc(&set->d_sp_start, nseries * sizeof(int64_t)) != hipSuccess ||
+        hipMalloc(&set->d_sp_cnt, nseries * sizeof(int64_t)) != hipSuccess ||
+        hipMalloc(&set->d_out_off, (nseries + 1) * sizeof(int64_t)) != hipSuccess ||
+        hipMalloc(&set->d_blocksums, (nblocks + 1) * sizeof(int64_t)) != hipSuccess) {
+        fail(GS_ERR, "hipMalloc raw set tables failed");
+        gs_groups_free(set);
+        return nullptr;
+    }
+    hipMemcpyAsync(set->d_groups, hg.data(), nseries * sizeof(DevGroup),
+                   hipMemcpyHostToDevice, ctx->stream);
+    hipMemcpyAsync(set->d_sgroups, hg.data(), nseries * sizeof(DevGroup),
+                   hipMemcpyHostToDevice, ctx->stream);
+    hipMemcpyAsync(set->d_sgroup_first, sgfirst.data(),
+                   sgfirst.size() * sizeof(int32_t), hipMemcpyHostToDevice,
+                   ctx->stream);
+    if (hipStreamSynchronize(ctx->stream) != hipSuccess) {
+        fail(GS_ERR, "raw set upload failed");
+        gs_groups_free(set);
+        return nullptr;
+    }
+    return set;
+}
+
 GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                  GsScanResult *result) {
     if (!ctx || !set || !spec || !result || !spec->d_ts || !spec->d_val)
         return fail(GS_ERR, "bad args to gs_scan");
-    if (set->ncols < 2)
-        return fail(GS_ERR, "gs_scan needs a time page + one f64 field page");
-    if (spec->field_col < 0 || 1 + uint32_t(spec->field_col) >= set->ncols ||
-        set->slots[1 + spec->field_col].ctype != GS_CT_F64)
-        return fail(GS_ERR, "gs_scan field column must be an f64 slot");
+    const bool raw = set->ncols == 0; /* gs_raw_set: memcache rows already
+                                         resident in spec->d_ts/d_val */
+    if (!raw) {
+        if (set->ncols < 2)
+            return fail(GS_ERR, "gs_scan needs a time page + one f64 field page");
+        if (spec->field_col < 0 || 1 + uint32_t(spec->field_col) >= set->ncols ||
+            set->slots[1 + spec->field_col].ctype != GS_CT_F64)
+            return fail(GS_ERR, "gs_scan field column must be an f64 slot");
+    }
     HIP_TRY(hipSetDevice(ctx->device));
 
     /* per-phase timing on the engine stream (HIP events) */
@@ -2970,12 +3043,21 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
     }
 
     HIP_TRY(hipEventRecord(ev[0], ctx->stream));
-    GsStatus st = gs_decode(ctx, set, 0, spec->d_ts, nullptr);
-    if (st != GS_OK) return st;
+    GsStatus st;
+    if (!raw) {
+        st = gs_decode(ctx, set, 0, spec->d_ts, nullptr);
+        if (st != GS_OK) return st;
+    }
     HIP_TRY(hipEventRecord(ev[1], ctx->stream));
-    st = gs_decode(ctx, set, 1 + uint32_t(spec->field_col), spec->d_val,
-                   d_valid);
-    if (st != GS_OK) return st;
+    if (!raw) {
+        st = gs_decode(ctx, set, 1 + uint32_t(spec->field_col), spec->d_val,
+                       d_valid);
+        if (st != GS_OK) return st;
+    } else if (d_valid) {
+        /* raw rows are non-null; tombstones clear from an all-valid base */
+        HIP_TRY(hipMemsetAsync(d_valid, 1, size_t(set->total_rows),
+                               ctx->stream));
+    }
     HIP_TRY(hipEventRecord(ev[2], ctx->stream));
 
     if (spec->n_tombstones > 0) {

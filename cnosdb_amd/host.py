@@ -436,6 +436,19 @@ class Engine:
         if st != 0:
             raise RuntimeError(f"gs_decode failed ({st}): {self._pl.err()}")
 
+    def raw_set(self, counts):
+        """Series structure over caller-resident device arrays (memcache
+        rows, MemCacheReader): counts[i] = rows of series i.  The
+        returned set is scanned with Engine.scan where d_ts/d_val hold
+        the rows themselves, and composes with compact_merge."""
+        counts = np.ascontiguousarray(counts, dtype=np.int64)
+        self.lib.gs_raw_set.restype = ctypes.c_void_p
+        h = self.lib.gs_raw_set(self._ctx, _np_ptr(counts), counts.size)
+        if not h:
+            raise RuntimeError(f"gs_raw_set failed: {self._pl.err()}")
+        return GroupSet(self, ctypes.c_void_p(h), int(counts.sum()),
+                        counts.size, None)
+
     def decode_str(self, gset, col, d_offsets, d_bytes, d_valid=None):
         """String column decode to Arrow varbinary layout
         (str_snappy_decode_to_array, string.rs:226-276): d_offsets int64

@@ -142,6 +142,18 @@ GsStatus gs_set_row_offsets(const GsGroupSet *set, int64_t *out);
 GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
                    uint8_t *d_valid);
 
+/* ---- memcache rows (MemCacheReader, tskv/src/reader/memcache_reader.rs;
+ * rows from mem_cache/series_data.rs:15-27) ----
+ * A raw row set gives the unflushed in-memory rows the same series
+ * structure as a page set, with NO pages: counts[i] = rows of series i
+ * (time-sorted, non-null — pre-filter nulls when building the arrays).
+ * gs_scan over a raw set skips the decode phases and filters/aggregates
+ * the caller's device-resident spec->d_ts / spec->d_val directly
+ * (tombstones and value predicates apply as usual; the fused path never
+ * runs).  Raw sets also feed gs_compact_merge, so hot (memcache) and
+ * cold (TSM) streams dedup together — newest stream wins at equal ts. */
+GsGroupSet *gs_raw_set(GsCtx *ctx, const int64_t *counts, int64_t nseries);
+
 /* ---- string column decode (str_snappy_decode_to_array,
  * codec/string.rs:226-276 via data_buf_to_arrow_array,
  * tsm/reader.rs:658-731) ----
