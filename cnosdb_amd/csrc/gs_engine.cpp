@@ -407,25 +407,29 @@ __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
             budget -= int64_t(k);
         };
         auto fd = fdesc[wv];
+        /* 4 source lanes per store instruction (see k_gor_lds_filtered):
+           the 16-deep ring leaves 48/64 lanes idle in a one-source
+           flush; this shape keeps every lane useful — 16 stores per
+           flush instead of 64 */
+        const int f_idx = lane & 15;
+        const int f_sq = lane >> 4;
         auto flush = [&]() {
             fd[lane][0] = (uint64_t)(uintptr_t)(o + (r - rfill));
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
-            /* register-batched: issue 8 independent LDS reads, then their
-               8 stores — the serial read->store chain was the kernel's
-               dominant wait */
-            for (int base = 0; base < 64; base += 8) {
-                double vbuf[8];
-                uint64_t ob[8];
-                int cnt[8];
-                for (int j = 0; j < 8; j++) {
-                    vbuf[j] = rslot[lane][base + j];
-                    ob[j] = fd[base + j][0];
-                    cnt[j] = int(fd[base + j][1]);
+            for (int src0 = 0; src0 < 64; src0 += 16) {
+                double vbuf[4];
+                uint64_t ob[4];
+                int cnt[4];
+                for (int t = 0; t < 4; t++) {
+                    int src = src0 + f_sq + t * 4;
+                    vbuf[t] = rslot[f_idx][src];
+                    ob[t] = fd[src][0];
+                    cnt[t] = int(fd[src][1]);
                 }
-                for (int j = 0; j < 8; j++)
-                    if (lane < cnt[j])
-                        ((double *)(uintptr_t)ob[j])[lane] = vbuf[j];
+                for (int t = 0; t < 4; t++)
+                    if (f_idx < cnt[t])
+                        ((double *)(uintptr_t)ob[t])[f_idx] = vbuf[t];
             }
             rfill = 0;
         };
@@ -1767,25 +1771,31 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
             budget -= int64_t(k);
         };
         auto fd = fdesc[wv];
+        /* 4 source lanes per store instruction: lane = (sq:2, idx:4), sq
+           picks the source sub-lane, idx the ring slot — the ring is 16
+           deep, so a one-source-per-instruction flush leaves 48/64 lanes
+           idle; this shape keeps every lane useful and issues 16 stores
+           per flush instead of 64 (reads batched 4 deep before their
+           stores, the earlier register-batching lesson) */
+        const int f_idx = lane & 15;
+        const int f_sq = lane >> 4;
         auto flush = [&]() {
             fd[lane][0] = (uint64_t)(uintptr_t)(o + run0);
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
-            /* register-batched: issue 8 independent LDS reads, then their
-               8 stores — the serial read->store chain was the kernel's
-               dominant wait */
-            for (int base = 0; base < 64; base += 8) {
-                double vbuf[8];
-                uint64_t ob[8];
-                int cnt[8];
-                for (int j = 0; j < 8; j++) {
-                    vbuf[j] = rslot[lane][base + j];
-                    ob[j] = fd[base + j][0];
-                    cnt[j] = int(fd[base + j][1]);
+            for (int src0 = 0; src0 < 64; src0 += 16) {
+                double vbuf[4];
+                uint64_t ob[4];
+                int cnt[4];
+                for (int t = 0; t < 4; t++) {
+                    int src = src0 + f_sq + t * 4;
+                    vbuf[t] = rslot[f_idx][src];
+                    ob[t] = fd[src][0];
+                    cnt[t] = int(fd[src][1]);
                 }
-                for (int j = 0; j < 8; j++)
-                    if (lane < cnt[j])
-                        ((double *)(uintptr_t)ob[j])[lane] = vbuf[j];
+                for (int t = 0; t < 4; t++)
+                    if (f_idx < cnt[t])
+                        ((double *)(uintptr_t)ob[t])[f_idx] = vbuf[t];
             }
             rfill = 0;
         };
