@@ -411,18 +411,19 @@ __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
            the 16-deep ring leaves 48/64 lanes idle in a one-source
            flush; this shape keeps every lane useful — 16 stores per
            flush instead of 64 */
-        const int f_idx = lane & 15;
-        const int f_sq = lane >> 4;
+        const int f_idx = lane & (GS_RING - 1);
+        const int f_sq = lane / GS_RING;
         auto flush = [&]() {
             fd[lane][0] = (uint64_t)(uintptr_t)(o + (r - rfill));
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
-            for (int src0 = 0; src0 < 64; src0 += 16) {
+            constexpr int SRCP = 64 / GS_RING; /* sources per store */
+            for (int src0 = 0; src0 < 64; src0 += SRCP * 4) {
                 double vbuf[4];
                 uint64_t ob[4];
                 int cnt[4];
                 for (int t = 0; t < 4; t++) {
-                    int src = src0 + f_sq + t * 4;
+                    int src = src0 + f_sq + t * SRCP;
                     vbuf[t] = rslot[f_idx][src];
                     ob[t] = fd[src][0];
                     cnt[t] = int(fd[src][1]);
@@ -1777,18 +1778,19 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
            idle; this shape keeps every lane useful and issues 16 stores
            per flush instead of 64 (reads batched 4 deep before their
            stores, the earlier register-batching lesson) */
-        const int f_idx = lane & 15;
-        const int f_sq = lane >> 4;
+        const int f_idx = lane & (GS_RING - 1);
+        const int f_sq = lane / GS_RING;
         auto flush = [&]() {
             fd[lane][0] = (uint64_t)(uintptr_t)(o + run0);
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
-            for (int src0 = 0; src0 < 64; src0 += 16) {
+            constexpr int SRCP = 64 / GS_RING; /* sources per store */
+            for (int src0 = 0; src0 < 64; src0 += SRCP * 4) {
                 double vbuf[4];
                 uint64_t ob[4];
                 int cnt[4];
                 for (int t = 0; t < 4; t++) {
-                    int src = src0 + f_sq + t * 4;
+                    int src = src0 + f_sq + t * SRCP;
                     vbuf[t] = rslot[f_idx][src];
                     ob[t] = fd[src][0];
                     cnt[t] = int(fd[src][1]);
