@@ -447,6 +447,7 @@ int32_t gs_encode_f64_pages_omp(const double *vals, int64_t rows_per_page,
                                 int64_t npages, uint8_t *dst,
                                 int64_t cap_per_page, int64_t *out_lens,
                                 int32_t nthreads) {
+    (void)nthreads; /* used only via the OMP clause below */
     int32_t err = 0;
 #pragma omp parallel for schedule(dynamic, 1) num_threads(nthreads)
     for (int64_t p = 0; p < npages; p++) {

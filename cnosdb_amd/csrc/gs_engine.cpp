@@ -1446,7 +1446,7 @@ __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
         sred[threadIdx.x] = cnt;
         __syncthreads();
         for (int w = blockDim.x >> 1; w > 0; w >>= 1) {
-            if (threadIdx.x < w) sred[threadIdx.x] += sred[threadIdx.x + w];
+            if (threadIdx.x < unsigned(w)) sred[threadIdx.x] += sred[threadIdx.x + w];
             __syncthreads();
         }
         if (threadIdx.x == 0)
@@ -1571,7 +1571,7 @@ __global__ void k_scan_partials(const int64_t *__restrict__ in, int n,
     __syncthreads();
     /* block scan (exclusive) over per-thread sums */
     for (int off = 1; off < SCAN_BLOCK; off <<= 1) {
-        int64_t v = (threadIdx.x >= off) ? sh[threadIdx.x - off] : 0;
+        int64_t v = (threadIdx.x >= unsigned(off)) ? sh[threadIdx.x - off] : 0;
         __syncthreads();
         sh[threadIdx.x] += v;
         __syncthreads();
