@@ -50,3 +50,30 @@ def test_i64_roundtrip_with_nulls(vals, data):
     dec = orc.decode_i64(enc, a.size, valid)
     exp = np.where(valid, a, 0)
     assert (dec == exp).all()
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.binary(min_size=0, max_size=120), min_size=1,
+                max_size=300))
+def test_str_roundtrip(strs):
+    """product snappy encoder == oracle byte-for-byte, and the oracle
+    decoder restores the strings (codec/string.rs snappy block)."""
+    enc = gs.encode_str(strs)
+    assert enc == orc.encode_str(strs)
+    assert orc.decode_str(enc, len(strs)) == strs
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.sampled_from([b"north", b"south", b"east", b"west",
+                                 b"", b"x" * 90]),
+                min_size=1, max_size=500), st.data())
+def test_str_roundtrip_with_nulls(strs, data):
+    """repetitive tag-like corpora (snappy copies) with a validity mask:
+    null rows consume nothing from the payload (string.rs:226-276)."""
+    valid = data.draw(st.lists(st.booleans(), min_size=len(strs),
+                               max_size=len(strs)))
+    present = [s for s, v in zip(strs, valid) if v]
+    enc = gs.encode_str(present)
+    got = orc.decode_str(enc, len(strs), np.array(valid, dtype=bool))
+    exp = [s if v else None for s, v in zip(strs, valid)]
+    assert got == exp
