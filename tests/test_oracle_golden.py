@@ -212,3 +212,18 @@ def test_str_uncompressed_and_empty():
         len(s).to_bytes(8, "big") + s for s in strs)
     assert orc.decode_str(blk, 3) == strs
     assert orc.decode_str(b"", 4) == [None] * 4
+
+
+def test_ts_sub_encoding_selection():
+    """timestamp.rs:414-512: the encoder's RLE-vs-simple8b choice is part
+    of the byte contract (dst[1] >> 4); both the oracle and the product
+    encoder must pick the reference's sub-encoding and round-trip."""
+    import cnosdb_amd as gs
+    g = GOLD["ts_selection"]
+    for sub, cases in ((2, g["rle"]), (1, g["simple8b"])):
+        for case in cases:
+            a = np.array(case, dtype=np.int64)
+            for enc in (orc.encode_ts(a), gs.encode_ts(a)):
+                assert enc[1] >> 4 == sub, case[:4]
+            assert orc.encode_ts(a) == gs.encode_ts(a)
+            assert (orc.decode_i64(orc.encode_ts(a), a.size) == a).all()
