@@ -669,3 +669,62 @@ def test_scan_fused_agg_multigroup(engine):
     assert (gmx[ect > 0] == emx[ect > 0]).all()
     assert np.allclose(d_sum.cpu().numpy(), esm, rtol=1e-12)
     gset.free()
+
+
+def test_upload_packed_matches_upload(engine):
+    """The vectorized packed upload (one contiguous buffer + offset
+    arrays, used by bench.py) must produce the same decode and scan
+    results as the per-page upload path."""
+    r = np.random.default_rng(71)
+    t0 = 1_700_000_000_000_000_000
+    nser, npts = 12, 4096
+    pages = []
+    for s in range(nser):
+        ts = t0 + np.arange(npts, dtype=np.int64) * 10**9
+        vals = np.round(np.clip(np.cumsum(r.normal(0, 0.5, npts)) + 50, 0, 100), 1)
+        pages.append((gs.page_of(ts, gs.CT_TIME), gs.page_of(vals, gs.CT_F64)))
+    groups = [(s, [(tp, gs.CT_TIME), (vp, gs.CT_F64)])
+              for s, (tp, vp) in enumerate(pages)]
+    g1 = engine.upload(groups)
+    buf = b"".join(tp + vp for tp, vp in pages)
+    off, ln = [], []
+    cur = 0
+    for tp, vp in pages:
+        off += [cur, cur + len(tp)]
+        ln += [len(tp), len(vp)]
+        cur += len(tp) + len(vp)
+    g2 = engine.upload_packed(
+        np.frombuffer(buf, dtype=np.uint8),
+        np.array(off, dtype=np.int64), np.array(ln, dtype=np.int64),
+        np.full(nser * 2, npts, dtype=np.int64),
+        np.tile(np.array([gs.CT_TIME, gs.CT_F64], dtype=np.uint8), nser),
+        np.arange(nser, dtype=np.int64), 2)
+    assert g1.rows == g2.rows
+    outs = []
+    for g in (g1, g2):
+        rows = g.rows
+        d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+        d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+        d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+        d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+        nb = 16
+        d_max = torch.full((nb,), -np.inf, dtype=torch.float64, device="cuda")
+        d_sum = torch.zeros(nb, dtype=torch.float64, device="cuda")
+        d_cnt = torch.zeros(nb, dtype=torch.int64, device="cuda")
+        res = engine.scan(g, d_ts, d_val,
+                          time_range=(t0 + 100 * 10**9, t0 + 4000 * 10**9),
+                          d_out_ts=d_ots, d_out_val=d_oval,
+                          agg=dict(bucket_ns=300 * 10**9, t0=t0, n_buckets=nb,
+                                   d_max=d_max, d_sum=d_sum, d_count=d_cnt))
+        outs.append((res.out_rows, d_ots[:res.out_rows].cpu().numpy(),
+                     d_oval[:res.out_rows].cpu().numpy(),
+                     d_max.cpu().numpy(), d_sum.cpu().numpy(),
+                     d_cnt.cpu().numpy()))
+        g.free()
+    (n1, ts1, v1, m1, s1, c1), (n2, ts2, v2, m2, s2, c2) = outs
+    assert n1 == n2
+    assert (ts1 == ts2).all()
+    assert v1.view(np.uint64).tolist() == v2.view(np.uint64).tolist()
+    assert (c1 == c2).all()
+    assert m1.tolist() == m2.tolist()
+    assert s1.tolist() == s2.tolist()
