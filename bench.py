@@ -373,7 +373,7 @@ def bench_strings(args):
     rng = np.random.default_rng(231)
     page_rows = args.page_rows
     npages = args.series if args.series != 10000 else 24000
-    per_set = min(npages, 30000)  # 134M-row scan limit at 4000 rows/page
+    per_set = min(npages, 60000)  # 268M-row scan limit at 4000 rows/page
     uniq = min(args.unique, 256)
     tagpool = [b"hostname=host_%04d,region=region_%02d,rack=%02d"
                % (i, i % 16, i % 64) for i in range(100)]

@@ -2719,8 +2719,8 @@ GsStatus gs_decode_str(GsCtx *ctx, GsGroupSet *set, uint32_t col,
     SlotPages &sp = set->slots[col];
     if (sp.ctype != GS_CT_STR || sp.n[PC_STR] != int(sp.host[PC_STR].size()))
         return fail(GS_ERR, "column is not a string column");
-    if (set->total_rows > int64_t(SCAN_BLOCK) * SCAN_ITEMS * 65535)
-        return fail(GS_ERR, "string decode row limit exceeded (134M)");
+    if (set->total_rows > int64_t(SCAN_BLOCK) * SCAN_ITEMS * 131072)
+        return fail(GS_ERR, "string decode row limit exceeded (268M)");
     HIP_TRY(hipSetDevice(ctx->device));
     int n = sp.n[PC_STR];
     if (!set->d_str_scratch || set->str_scratch_cap < size_t(sp.str_total)) {
