@@ -372,7 +372,7 @@ def bench_strings(args):
     import cnosdb_amd as gs
     rng = np.random.default_rng(231)
     page_rows = args.page_rows
-    npages = args.series if args.series != 10000 else 24000
+    npages = args.series if args.series != 10000 else 60000
     per_set = min(npages, 60000)  # 268M-row scan limit at 4000 rows/page
     uniq = min(args.unique, 256)
     tagpool = [b"hostname=host_%04d,region=region_%02d,rack=%02d"
