@@ -346,75 +346,219 @@ __device__ __forceinline__ void gorilla_decode_page(
     }
 }
 
-/* Gorilla decode for all-valid pages with LDS-staged output: each lane
- * decodes its own page (bitstream is sequential), staging GS_RING values
- * in LDS; the wave flushes cooperatively with page-contiguous coalesced
- * stores.  The direct per-lane 8-B stores of the naive version touch 64
- * distinct cache lines per wave-store and were measured address-bound
- * (profiles/: storeonly 11.8 ms vs lds-staged ~3 ms for the same bytes).
- * Lanes run in lockstep (one value per iteration), so rings fill together;
- * lanes whose page ended keep cooperating in flushes until all are done. */
 #define GS_RING 16
 #define GS_GOR_BLOCK 256 /* 4 waves; RING 8/16/32 x block 128/256 swept: 16/256 optimal */
 
-__global__ void k_gor_lds(const uint8_t *__restrict__ blob,
-                          const DevPage *__restrict__ pages, int npages,
-                          double *__restrict__ out,
-                          uint8_t *__restrict__ valid,
-                          unsigned *__restrict__ err) {
-    __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1]; /* [wave][slot][lane] */
+/* --- Gorilla chunked decode (see DevGorChunk in gs_internal.h) ---
+ * k_gor_sync: one-time upload pre-pass.  One thread walks each multi-chunk
+ * all-valid Gorilla page with the same window parser as decode (no value
+ * stores) and records the parser state every GOR_CHUNK values, so decode
+ * can start mid-stream.  Errors are NOT raised here: a truncated or
+ * early-sentinel page poisons its unrecorded chunks (bitpos past the
+ * stream end), and the decode of that chunk raises DERR_SHORT exactly
+ * where the page-sequential kernel would have — same error surface,
+ * surfaced at decode/scan time like the reference's decode errors. */
+__global__ void k_gor_sync(const uint8_t *__restrict__ blob,
+                           const DevPage *__restrict__ pages, int npages,
+                           const int32_t *__restrict__ chunk_base,
+                           DevGorChunk *__restrict__ chunks,
+                           uint32_t chunk_rows) {
+    for (int pi = blockIdx.x * blockDim.x + threadIdx.x; pi < npages;
+         pi += gridDim.x * blockDim.x) {
+        DevPage pg = pages[pi];
+        int cb = chunk_base[pi];
+        int nch = chunk_base[pi + 1] - cb;
+        if (nch <= 1) continue; /* single chunk: decode reads the header */
+        const uint8_t *data = blob + pg.data_off;
+        int next_k = 1;
+        bool ok = false;
+        if (pg.data_len >= 10) {
+            int64_t budget0 = int64_t(pg.data_len - 10) * 8;
+            uint64_t val = dev_be64(data + 2);
+            const uint8_t *p = data + 10;
+            int64_t budget = budget0;
+            uint64_t hi = 0, lo = 0;
+            int nb = 0;
+            uint32_t trailing = 0, meaningful = 64;
+            uint32_t r = 1; /* header value = row 0 */
+            uint64_t nextw = dev_be64(p);
+            p += 8;
+            auto topup = [&]() { /* only with nb < 64 */
+                uint64_t x = nextw;
+                nextw = dev_be64(p);
+                p += 8;
+                if (nb == 0) { hi = x; lo = 0; }
+                else { hi |= x >> nb; lo = x << (64 - nb); }
+                nb += 64;
+            };
+            auto consume = [&](unsigned k) {
+                hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
+                lo = (k == 64) ? 0 : (lo << k);
+                nb -= int(k);
+                budget -= int64_t(k);
+            };
+            ok = true;
+            for (;;) {
+                if (r == uint32_t(next_k) * chunk_rows) {
+                    DevGorChunk &c = chunks[cb + next_k];
+                    c.bitpos = uint64_t(budget0 - budget);
+                    c.val = val;
+                    c.trailing = uint8_t(trailing);
+                    c.meaningful = uint8_t(meaningful);
+                    /* the previous chunk's whole range is now known
+                       decodable: the filtered kernel may stop it early */
+                    chunks[cb + next_k - 1].safe_stop = 1;
+                    next_k++;
+                    if (next_k == nch) break; /* tail chunk parses itself */
+                }
+                if (nb < 64) topup();
+                if (budget <= 0) { ok = false; break; }
+                uint32_t top13 = uint32_t(hi >> 51);
+                if (!(top13 & 0x1000)) { /* ctrl 0: repeat */
+                    consume(1);
+                } else {
+                    if (top13 & 0x0800) { /* ctrl 11: new window */
+                        uint32_t lead = (top13 >> 6) & 0x1f;
+                        meaningful = top13 & 0x3f;
+                        if (meaningful > 0) trailing = 64 - lead - meaningful;
+                        else { trailing = 0; meaningful = 64; }
+                        consume(13);
+                    } else { /* ctrl 10: reuse window */
+                        consume(2);
+                    }
+                    while (nb < int(meaningful)) topup();
+                    uint64_t sb =
+                        (meaningful == 64) ? hi : (hi >> (64 - meaningful));
+                    consume(meaningful);
+                    if (budget < 0) { ok = false; break; }
+                    val ^= sb << trailing;
+                    /* sentinel before the last sync point: rows missing */
+                    if (val == GORILLA_SENTINEL) { ok = false; break; }
+                }
+                r++;
+            }
+        }
+        if (!ok) { /* poison unrecorded chunks -> DERR_SHORT at decode */
+            for (int k = next_k; k < nch; k++) {
+                DevGorChunk &c = chunks[cb + k];
+                c.bitpos = uint64_t(pg.data_len) * 8;
+                c.val = 0;
+                c.trailing = 0;
+                c.meaningful = 64;
+            }
+        }
+    }
+}
+
+/* Parser window state of one chunk, positioned at DevGorChunk.bitpos. */
+struct GorChunkState {
+    const uint8_t *p; /* next 8-byte word to prefetch */
+    uint64_t hi, lo, nextw, nextw2;
+    uint64_t val;
+    int64_t budget;
+    int nb;
+    uint32_t trailing, meaningful;
+    bool bad;
+};
+
+__device__ __forceinline__ GorChunkState
+gor_chunk_init(const uint8_t *__restrict__ blob, const DevGorChunk &c) {
+    GorChunkState st;
+    st.bad = false;
+    const uint8_t *data = blob + c.data_off;
+    /* page data: [enc=6][0x10][first f64 BE][bitstream] (float.rs:418-444) */
+    if (c.data_len < 10) { st.bad = true; return st; }
+    int64_t total_bits = int64_t(c.data_len - 10) * 8;
+    if (c.row0 == 0) {
+        st.val = dev_be64(data + 2);
+        st.budget = total_bits;
+        st.trailing = 0;
+        st.meaningful = 64;
+        const uint8_t *p = data + 10;
+        st.hi = 0;
+        st.lo = 0;
+        st.nb = 0;
+        st.nextw = dev_be64(p);
+        st.nextw2 = dev_be64(p + 8);
+        st.p = p + 16;
+    } else {
+        st.val = c.val;
+        st.budget = total_bits - int64_t(c.bitpos);
+        st.trailing = c.trailing;
+        st.meaningful = c.meaningful;
+        if (st.budget <= 0) { st.bad = true; return st; } /* poisoned */
+        const uint8_t *p = data + 10 + (c.bitpos >> 6) * 8;
+        unsigned rem = unsigned(c.bitpos & 63);
+        uint64_t w0 = dev_be64(p);
+        p += 8;
+        st.hi = (rem == 0) ? w0 : (w0 << rem);
+        st.lo = 0;
+        st.nb = int(64 - rem);
+        st.nextw = dev_be64(p);
+        st.nextw2 = dev_be64(p + 8);
+        st.p = p + 16;
+    }
+    return st;
+}
+
+/* Chunk-parallel Gorilla decode with LDS-staged output: each lane decodes
+ * one chunk, staging GS_RING values in LDS; the wave flushes cooperatively
+ * with chunk-contiguous coalesced stores.  The direct per-lane 8-B stores
+ * of the naive version touch 64 distinct cache lines per wave-store and
+ * were measured address-bound (profiles/: storeonly 11.8 ms vs lds-staged
+ * ~3 ms for the same bytes).  Lanes run in lockstep (one value per
+ * iteration), so rings fill together; lanes whose chunk ended keep
+ * cooperating in flushes until all are done.  Chunks are page-major in
+ * the table, so a wave's lanes hold NEIGHBORING regions of one page:
+ * their refill streams share L2 lines and their branch patterns
+ * correlate (same series' data), unlike the round-1 page-per-lane shape. */
+__global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
+                             const DevGorChunk *__restrict__ chunks,
+                             int nchunks, double *__restrict__ out,
+                             uint8_t *__restrict__ valid,
+                             unsigned *__restrict__ err) {
+    __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1];
     __shared__ uint64_t fdesc[GS_GOR_BLOCK / 64][64][2];
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     auto rslot = ring[wv];
     int stride = gridDim.x * blockDim.x;
     int base_id = blockIdx.x * blockDim.x + threadIdx.x;
-    int rounds = (npages + stride - 1) / stride;
+    int rounds = (nchunks + stride - 1) / stride;
     for (int rd = 0; rd < rounds; rd++) {
-        int p0 = base_id + rd * stride;
-        bool have = p0 < npages;
-        DevPage pg = pages[have ? p0 : 0];
-        const uint8_t *s = blob + pg.data_off + 1;
-        double *o = out + pg.row_off;
-        uint32_t nrows = pg.nrows;
-        uint32_t slen = pg.data_len - 1;
-        uint64_t val = dev_be64(s + 1);
-        const uint8_t *p = s + 9;
-        int64_t budget = int64_t(slen - 9) * 8;
-        uint64_t hi = 0, lo = 0;
-        int nb = 0;
-        uint32_t trailing = 0, meaningful = 64;
-        int r = 0;
+        int ci = base_id + rd * stride;
+        bool have = ci < nchunks;
+        /* branch-free prologue: clamp to chunk 0 so every load issues
+           unconditionally (see k_gor_chunks_filtered) */
+        DevGorChunk ch = chunks[have ? ci : 0];
+        double *o = out + ch.row_off;
+        uint32_t r = ch.row0;
+        uint32_t end = ch.row0 + ch.cnt;
+        GorChunkState st = gor_chunk_init(blob, ch);
         int rfill = 0;
         bool done = !have;
-        if (have && slen < 9) { atomicOr(err, DERR_SHORT); done = true; }
-        uint64_t nextw = dev_be64(p);
-        uint64_t nextw2 = dev_be64(p + 8);
-        p += 16;
+        if (!have) { r = 0; end = 0; }
+        if (have && st.bad) { atomicOr(err, DERR_SHORT); done = true; }
         auto topup = [&]() { /* only with nb < 64 */
-            uint64_t x = nextw;
-            nextw = nextw2;
-            nextw2 = dev_be64(p);
-            p += 8;
-            if (nb == 0) { hi = x; lo = 0; }
-            else { hi |= x >> nb; lo = x << (64 - nb); }
-            nb += 64;
+            uint64_t x = st.nextw;
+            st.nextw = st.nextw2;
+            st.nextw2 = dev_be64(st.p);
+            st.p += 8;
+            if (st.nb == 0) { st.hi = x; st.lo = 0; }
+            else { st.hi |= x >> st.nb; st.lo = x << (64 - st.nb); }
+            st.nb += 64;
         };
         auto consume = [&](unsigned k) {
-            hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
-            lo = (k == 64) ? 0 : (lo << k);
-            nb -= int(k);
-            budget -= int64_t(k);
+            st.hi = (k == 64) ? st.lo : ((st.hi << k) | (st.lo >> (64 - k)));
+            st.lo = (k == 64) ? 0 : (st.lo << k);
+            st.nb -= int(k);
+            st.budget -= int64_t(k);
         };
         auto fd = fdesc[wv];
-        /* 4 source lanes per store instruction (see k_gor_lds_filtered):
-           the 16-deep ring leaves 48/64 lanes idle in a one-source
-           flush; this shape keeps every lane useful — 16 stores per
-           flush instead of 64 */
         const int f_idx = lane & (GS_RING - 1);
         const int f_sq = lane / GS_RING;
         auto flush = [&]() {
-            fd[lane][0] = (uint64_t)(uintptr_t)(o + (r - rfill));
+            fd[lane][0] = (uint64_t)(uintptr_t)(o + (int64_t(r) - rfill));
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
             constexpr int SRCP = 64 / GS_RING; /* sources per store */
@@ -434,58 +578,64 @@ __global__ void k_gor_lds(const uint8_t *__restrict__ blob,
             }
             rfill = 0;
         };
-        if (!done && r < int(nrows)) {
-            rslot[rfill][lane] = __longlong_as_double((long long)val);
+        if (!done && ch.row0 == 0 && r < end) { /* header value = row 0 */
+            rslot[rfill][lane] = __longlong_as_double((long long)st.val);
             rfill++;
             r++;
         }
         while (!__all(done)) {
             if (!done) {
-                if (nb < 64) topup();
-                if (budget <= 0) { atomicOr(err, DERR_SHORT); done = true; }
+                if (st.nb < 64) topup();
+                if (st.budget <= 0) { atomicOr(err, DERR_SHORT); done = true; }
             }
             if (!done) {
-                uint32_t top13 = uint32_t(hi >> 51);
-                bool stage = true;
+                uint32_t top13 = uint32_t(st.hi >> 51);
+                bool stg = true;
                 if (!(top13 & 0x1000)) {
                     consume(1);
                 } else {
                     if (top13 & 0x0800) {
                         uint32_t lead = (top13 >> 6) & 0x1f;
-                        meaningful = top13 & 0x3f;
-                        if (meaningful > 0) trailing = 64 - lead - meaningful;
-                        else { trailing = 0; meaningful = 64; }
+                        st.meaningful = top13 & 0x3f;
+                        if (st.meaningful > 0)
+                            st.trailing = 64 - lead - st.meaningful;
+                        else { st.trailing = 0; st.meaningful = 64; }
                         consume(13);
                     } else {
                         consume(2);
                     }
-                    while (nb < int(meaningful)) topup();
-                    uint64_t sb =
-                        (meaningful == 64) ? hi : (hi >> (64 - meaningful));
-                    consume(meaningful);
-                    if (budget < 0) {
+                    while (st.nb < int(st.meaningful)) topup();
+                    uint64_t sb = (st.meaningful == 64)
+                                      ? st.hi
+                                      : (st.hi >> (64 - st.meaningful));
+                    consume(st.meaningful);
+                    if (st.budget < 0) {
                         atomicOr(err, DERR_SHORT);
                         done = true;
-                        stage = false;
+                        stg = false;
                     } else {
-                        val ^= sb << trailing;
-                        if (val == GORILLA_SENTINEL) { done = true; stage = false; }
+                        st.val ^= sb << st.trailing;
+                        if (st.val == GORILLA_SENTINEL) { done = true; stg = false; }
                     }
                 }
-                if (stage && r < int(nrows)) {
-                    rslot[rfill][lane] = __longlong_as_double((long long)val);
+                if (stg && r < end) {
+                    rslot[rfill][lane] =
+                        __longlong_as_double((long long)st.val);
                     rfill++;
                     r++;
+                    /* non-last chunk: all rows produced, stop; the last
+                       chunk keeps parsing to the sentinel like the page-
+                       sequential kernel (a missing sentinel is an error) */
+                    if (r == end && !ch.last) done = true;
                 }
             }
             if (__any(rfill == GS_RING)) flush();
         }
         flush();
-        /* all-valid: every row must have been produced */
-        if (have && r < int(nrows)) atomicOr(err, DERR_SHORT);
+        if (have && r < end) atomicOr(err, DERR_SHORT);
         if (have && valid) {
-            uint8_t *vd = valid + pg.row_off;
-            for (uint32_t k = 0; k < nrows; k++) vd[k] = 1;
+            uint8_t *vd = valid + ch.row_off;
+            for (uint32_t k = ch.row0; k < end; k++) vd[k] = 1;
         }
     }
 }
@@ -1705,17 +1855,54 @@ __global__ void k_rle_ts_filtered(const uint8_t *__restrict__ blob,
     }
 }
 
-/* Gorilla decode writing only the selected span, compacted (LDS-staged
- * cooperative flush as k_gor_lds; each lane's staged run is contiguous in
- * the output by construction) */
-__global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
-                                   const DevPage *__restrict__ pages,
-                                   int npages,
-                                   const int64_t *__restrict__ sp_start,
-                                   const int64_t *__restrict__ sp_cnt,
-                                   const int64_t *__restrict__ out_off,
-                                   double *__restrict__ out,
-                                   unsigned *__restrict__ err) {
+/* Sub-page pruning for the fused scan: compact the indices of chunks whose
+ * row range overlaps their group's selected span (the chunk-granularity
+ * analog of the reference's page min/max pruning, reader/chunk.rs:12-49 —
+ * a chunk wholly outside the time range is never parsed at all).  Output
+ * order is wave-compacted (near table order); chunks write disjoint
+ * output rows so order never affects results. */
+__global__ void k_gor_active(const DevGorChunk *__restrict__ chunks,
+                             int nchunks,
+                             const int64_t *__restrict__ sp_start,
+                             const int64_t *__restrict__ sp_cnt,
+                             int *__restrict__ active,
+                             int *__restrict__ n_active) {
+    const int lane = threadIdx.x & 63;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < nchunks;
+         i += gridDim.x * blockDim.x) {
+        uint32_t grp = chunks[i].grp;
+        int64_t row0 = int64_t(chunks[i].row0);
+        int64_t rend = row0 + int64_t(chunks[i].cnt);
+        int64_t lo = sp_start[grp];
+        int64_t hi = lo + sp_cnt[grp];
+        bool act = row0 < hi && rend > lo;
+        uint64_t m = __ballot(act);
+        if (m) {
+            int leader = __ffsll((long long)m) - 1;
+            int base = 0;
+            if (lane == leader) base = atomicAdd(n_active, __popcll(m));
+            base = __shfl(base, leader);
+            if (act)
+                active[base + __popcll(m & ((1ull << lane) - 1))] = i;
+        }
+    }
+}
+
+/* Chunk-parallel Gorilla decode writing only the selected span, compacted
+ * (same LDS flush as k_gor_chunks; each lane's staged run is contiguous
+ * in the output by construction).  Runs over the k_gor_active index list;
+ * rows before the span are parsed but not staged (the bitstream is
+ * sequential), and a chunk whose decodability was proven by the upload
+ * pre-pass (safe_stop) stops as soon as the span's end is passed. */
+__global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
+                                      const DevGorChunk *__restrict__ chunks,
+                                      const int *__restrict__ active,
+                                      const int *__restrict__ n_active,
+                                      const int64_t *__restrict__ sp_start,
+                                      const int64_t *__restrict__ sp_cnt,
+                                      const int64_t *__restrict__ out_off,
+                                      double *__restrict__ out,
+                                      unsigned *__restrict__ err) {
     __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1];
     /* per-flush lane descriptors: {dst pointer, staged count} packed so the
        flush loop reads ONE broadcast ds_read per source lane instead of
@@ -1724,52 +1911,42 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     auto rslot = ring[wv];
+    const int nact = *n_active;
     int stride = gridDim.x * blockDim.x;
     int base_id = blockIdx.x * blockDim.x + threadIdx.x;
-    int rounds = (npages + stride - 1) / stride;
+    int rounds = (nact + stride - 1) / stride;
     for (int rd = 0; rd < rounds; rd++) {
-        int p0 = base_id + rd * stride;
-        bool have = p0 < npages;
-        /* branch-free prologue: clamp to page 0 so every load issues
+        int ai = base_id + rd * stride;
+        bool have = ai < nact;
+        /* branch-free prologue: clamp to entry 0 so every load issues
            unconditionally and the compiler batches them under one wait
-           (the predicated version serialized ~5 dependent vmcnt(0)s per
-           page); !have lanes are done immediately and never use these */
-        DevPage pg = pages[have ? p0 : 0];
-        const uint8_t *s = blob + pg.data_off + 1;
-        int64_t sel_lo = sp_start[pg.grp];
-        int64_t sel_hi = sel_lo + sp_cnt[pg.grp];
+           (the predicated version serialized ~5 dependent vmcnt(0)s) */
+        DevGorChunk ch = chunks[active[have ? ai : 0]];
+        int64_t sel_lo = sp_start[ch.grp];
+        int64_t sel_hi = sel_lo + sp_cnt[ch.grp];
         if (!have) { sel_lo = 0; sel_hi = 0; }
-        double *o = out + out_off[pg.grp] - sel_lo; /* o[r] valid for r in span */
-        uint32_t nrows = pg.nrows;
-        uint32_t slen = pg.data_len - 1;
-        uint64_t val = dev_be64(s + 1);
-        const uint8_t *p = s + 9;
-        int64_t budget = int64_t(slen - 9) * 8;
-        uint64_t hi = 0, lo = 0;
-        int nb = 0;
-        uint32_t trailing = 0, meaningful = 64;
-        int64_t r = 0;   /* rows produced */
-        int rfill = 0;   /* staged */
+        double *o = out + out_off[ch.grp] - sel_lo; /* o[r] valid in span */
+        int64_t r = int64_t(ch.row0);
+        int64_t end = r + int64_t(ch.cnt);
+        GorChunkState st = gor_chunk_init(blob, ch);
+        int rfill = 0;
         int64_t run0 = 0; /* output row of first staged entry */
         bool done = !have;
-        if (have && slen < 9) { atomicOr(err, DERR_SHORT); done = true; }
-        uint64_t nextw = dev_be64(p);
-        uint64_t nextw2 = dev_be64(p + 8);
-        p += 16;
-        auto topup = [&]() {
-            uint64_t x = nextw;
-            nextw = nextw2;
-            nextw2 = dev_be64(p);
-            p += 8;
-            if (nb == 0) { hi = x; lo = 0; }
-            else { hi |= x >> nb; lo = x << (64 - nb); }
-            nb += 64;
+        if (have && st.bad) { atomicOr(err, DERR_SHORT); done = true; }
+        auto topup = [&]() { /* only with nb < 64 */
+            uint64_t x = st.nextw;
+            st.nextw = st.nextw2;
+            st.nextw2 = dev_be64(st.p);
+            st.p += 8;
+            if (st.nb == 0) { st.hi = x; st.lo = 0; }
+            else { st.hi |= x >> st.nb; st.lo = x << (64 - st.nb); }
+            st.nb += 64;
         };
         auto consume = [&](unsigned k) {
-            hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
-            lo = (k == 64) ? 0 : (lo << k);
-            nb -= int(k);
-            budget -= int64_t(k);
+            st.hi = (k == 64) ? st.lo : ((st.hi << k) | (st.lo >> (64 - k)));
+            st.lo = (k == 64) ? 0 : (st.lo << k);
+            st.nb -= int(k);
+            st.budget -= int64_t(k);
         };
         auto fd = fdesc[wv];
         /* 4 source lanes per store instruction: lane = (sq:2, idx:4), sq
@@ -1808,47 +1985,54 @@ __global__ void k_gor_lds_filtered(const uint8_t *__restrict__ blob,
                 rfill++;
             }
             r++;
+            /* all rows produced (non-last chunk), or span passed on a
+               chunk the pre-pass proved decodable: stop parsing */
+            if ((r == end && !ch.last) || (r >= sel_hi && ch.safe_stop))
+                done = true;
         };
-        if (!done && r < int64_t(nrows)) stage_row(val);
+        if (!done && ch.row0 == 0 && r < end) stage_row(st.val);
         while (!__all(done)) {
             if (!done) {
-                if (nb < 64) topup();
-                if (budget <= 0) { atomicOr(err, DERR_SHORT); done = true; }
+                if (st.nb < 64) topup();
+                if (st.budget <= 0) { atomicOr(err, DERR_SHORT); done = true; }
             }
             if (!done) {
-                uint32_t top13 = uint32_t(hi >> 51);
+                uint32_t top13 = uint32_t(st.hi >> 51);
                 bool stg = true;
                 if (!(top13 & 0x1000)) {
                     consume(1);
                 } else {
                     if (top13 & 0x0800) {
                         uint32_t lead = (top13 >> 6) & 0x1f;
-                        meaningful = top13 & 0x3f;
-                        if (meaningful > 0) trailing = 64 - lead - meaningful;
-                        else { trailing = 0; meaningful = 64; }
+                        st.meaningful = top13 & 0x3f;
+                        if (st.meaningful > 0)
+                            st.trailing = 64 - lead - st.meaningful;
+                        else { st.trailing = 0; st.meaningful = 64; }
                         consume(13);
                     } else {
                         consume(2);
                     }
-                    while (nb < int(meaningful)) topup();
-                    uint64_t sb =
-                        (meaningful == 64) ? hi : (hi >> (64 - meaningful));
-                    consume(meaningful);
-                    if (budget < 0) {
+                    while (st.nb < int(st.meaningful)) topup();
+                    uint64_t sb = (st.meaningful == 64)
+                                      ? st.hi
+                                      : (st.hi >> (64 - st.meaningful));
+                    consume(st.meaningful);
+                    if (st.budget < 0) {
                         atomicOr(err, DERR_SHORT);
                         done = true;
                         stg = false;
                     } else {
-                        val ^= sb << trailing;
-                        if (val == GORILLA_SENTINEL) { done = true; stg = false; }
+                        st.val ^= sb << st.trailing;
+                        if (st.val == GORILLA_SENTINEL) { done = true; stg = false; }
                     }
                 }
-                if (stg && r < int64_t(nrows)) stage_row(val);
+                if (stg && r < end) stage_row(st.val);
             }
             if (__any(rfill == GS_RING)) flush();
         }
         flush();
-        if (have && r < int64_t(nrows)) atomicOr(err, DERR_SHORT);
+        if (have && r < end && !(r >= sel_hi && ch.safe_stop))
+            atomicOr(err, DERR_SHORT);
     }
 }
 
@@ -2156,6 +2340,13 @@ struct GsCtx {
     unsigned *d_err;
 };
 
+static int grid_for(int work, int per_block) {
+    int blocks = (work + per_block - 1) / per_block;
+    if (blocks > 2048) blocks = 2048; /* grid-stride beyond (Guideline 11) */
+    if (blocks < 1) blocks = 1;
+    return blocks;
+}
+
 struct SlotPages {
     uint8_t ctype;
     std::vector<DevPage> host[PC_NCLASS];
@@ -2167,6 +2358,11 @@ struct SlotPages {
     std::vector<int64_t> str_scr;
     int64_t str_total = 0;
     int64_t *d_str_scr = nullptr;
+    /* Gorilla chunk table (page-major over host[PC_GOR]; states filled by
+       the k_gor_sync upload pre-pass) */
+    DevGorChunk *d_gor_chunks = nullptr;
+    int n_gor_chunks = 0;
+    int32_t *d_gor_chunk_base = nullptr; /* [n[PC_GOR]+1] */
 };
 
 struct GsGroupSet {
@@ -2197,6 +2393,11 @@ struct GsGroupSet {
                                      row / RLE delta (k_spans_rle out) */
     int64_t *d_g_delta = nullptr;
     int max_span = 0; /* max page-groups per series-group */
+    /* fused-scan active-chunk list (lazy; sized to the largest slot's
+       chunk table) */
+    int *d_gor_active = nullptr;
+    int *d_gor_nactive = nullptr;
+    size_t gor_active_cap = 0;
     /* string decode scratch (lazy, cached across gs_decode_str calls) */
     uint8_t *d_str_scratch = nullptr;
     size_t str_scratch_cap = 0;
@@ -2503,6 +2704,64 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                            sp.n[PC_STR] * sizeof(int64_t),
                            hipMemcpyHostToDevice, ctx->stream);
         }
+        if (sp.n[PC_GOR]) {
+            /* chunk skeleton (page-major), states filled by k_gor_sync.
+               Chunk size is tunable for sweeps (GS_GOR_CHUNK env). */
+            static uint32_t chunk_rows = [] {
+                const char *e = getenv("GS_GOR_CHUNK");
+                long v = e ? atol(e) : 0;
+                return uint32_t(v >= 64 ? v : GOR_CHUNK);
+            }();
+            std::vector<int32_t> cbase(sp.n[PC_GOR] + 1);
+            std::vector<DevGorChunk> hch;
+            int32_t acc = 0;
+            for (int i = 0; i < sp.n[PC_GOR]; i++) {
+                const DevPage &pg = sp.host[PC_GOR][i];
+                cbase[i] = acc;
+                uint32_t nch =
+                    pg.nrows ? (pg.nrows + chunk_rows - 1) / chunk_rows : 1;
+                for (uint32_t k = 0; k < nch; k++) {
+                    DevGorChunk c{};
+                    c.data_off = pg.data_off;
+                    c.row_off = pg.row_off;
+                    c.grp = pg.grp;
+                    c.row0 = k * chunk_rows;
+                    c.cnt = pg.nrows > c.row0
+                                ? (pg.nrows - c.row0 < chunk_rows
+                                       ? pg.nrows - c.row0 : chunk_rows)
+                                : 0;
+                    c.data_len = pg.data_len;
+                    c.meaningful = 64;
+                    c.last = uint8_t(k == nch - 1);
+                    hch.push_back(c);
+                }
+                acc += int32_t(nch);
+            }
+            cbase[sp.n[PC_GOR]] = acc;
+            sp.n_gor_chunks = acc;
+            if (hipMalloc(&sp.d_gor_chunks, hch.size() * sizeof(DevGorChunk)) !=
+                    hipSuccess ||
+                hipMalloc(&sp.d_gor_chunk_base,
+                          cbase.size() * sizeof(int32_t)) != hipSuccess) {
+                fail(GS_ERR, "hipMalloc gor chunk table failed");
+                gs_groups_free(set); return nullptr;
+            }
+            hipMemcpyAsync(sp.d_gor_chunks, hch.data(),
+                           hch.size() * sizeof(DevGorChunk),
+                           hipMemcpyHostToDevice, ctx->stream);
+            hipMemcpyAsync(sp.d_gor_chunk_base, cbase.data(),
+                           cbase.size() * sizeof(int32_t),
+                           hipMemcpyHostToDevice, ctx->stream);
+            /* hch/cbase are stack-local: wait for the staged copies before
+               they go out of scope (upload path, not scan time) */
+            hipStreamSynchronize(ctx->stream);
+            /* one-time sync-point pre-pass (upload/setup, not scan time) */
+            hipLaunchKernelGGL(k_gor_sync, dim3(grid_for(sp.n[PC_GOR], 256)),
+                               dim3(256), 0, ctx->stream, set->d_blob,
+                               sp.dev[PC_GOR], sp.n[PC_GOR],
+                               sp.d_gor_chunk_base, sp.d_gor_chunks,
+                               chunk_rows);
+        }
     }
 
     /* group table + span scratch */
@@ -2574,7 +2833,11 @@ void gs_groups_free(GsGroupSet *set) {
         for (int k = 0; k < PC_NCLASS; k++)
             if (sp.dev[k]) hipFree(sp.dev[k]);
         if (sp.d_str_scr) hipFree(sp.d_str_scr);
+        if (sp.d_gor_chunks) hipFree(sp.d_gor_chunks);
+        if (sp.d_gor_chunk_base) hipFree(sp.d_gor_chunk_base);
     }
+    if (set->d_gor_active) hipFree(set->d_gor_active);
+    if (set->d_gor_nactive) hipFree(set->d_gor_nactive);
     if (set->d_str_scratch) hipFree(set->d_str_scratch);
     if (set->d_str_sz) hipFree(set->d_str_sz);
     if (set->d_str_pos) hipFree(set->d_str_pos);
@@ -2635,12 +2898,6 @@ static GsStatus check_dev_err(GsCtx *ctx) {
     return GS_OK;
 }
 
-static int grid_for(int work, int per_block) {
-    int blocks = (work + per_block - 1) / per_block;
-    if (blocks > 2048) blocks = 2048; /* grid-stride beyond (Guideline 11) */
-    if (blocks < 1) blocks = 1;
-    return blocks;
-}
 
 GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
                    uint8_t *d_valid) {
@@ -2669,10 +2926,10 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
                                d_valid, ctx->d_err);
     }
     if (sp.n[PC_GOR]) {
-        int n = sp.n[PC_GOR];
-        hipLaunchKernelGGL(k_gor_lds, dim3(grid_for(n, GS_GOR_BLOCK)),
+        int n = sp.n_gor_chunks;
+        hipLaunchKernelGGL(k_gor_chunks, dim3(grid_for(n, GS_GOR_BLOCK)),
                            dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
-                           sp.dev[PC_GOR], n, (double *)d_out, d_valid,
+                           sp.d_gor_chunks, n, (double *)d_out, d_valid,
                            ctx->d_err);
     }
     if (sp.n[PC_S8B]) {
@@ -2813,7 +3070,18 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
     hipEvent_t *ev = set->sev;
     int ng = int(set->ngroups);
     const DevPage *ts_pages = set->slots[0].dev[PC_RLE_TS];
-    const DevPage *f_pages = set->slots[1 + spec->field_col].dev[PC_GOR];
+    SlotPages &fsp = set->slots[1 + spec->field_col];
+    int nch = fsp.n_gor_chunks;
+    if (!set->d_gor_nactive &&
+        hipMalloc(&set->d_gor_nactive, sizeof(int)) != hipSuccess)
+        return fail(GS_ERR, "hipMalloc active counter failed");
+    if (set->gor_active_cap < size_t(nch)) {
+        if (set->d_gor_active) hipFree(set->d_gor_active);
+        if (hipMalloc(&set->d_gor_active, size_t(nch) * sizeof(int)) !=
+            hipSuccess)
+            return fail(GS_ERR, "hipMalloc active list failed");
+        set->gor_active_cap = size_t(nch);
+    }
     int nblocks = (ng + SCAN_BLOCK * SCAN_ITEMS - 1) /
                   (SCAN_BLOCK * SCAN_ITEMS);
     if (nblocks > 2048)
@@ -2838,10 +3106,16 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                        set->d_sp_start, set->d_sp_cnt, set->d_out_off,
                        spec->d_out_ts);
     HIP_TRY(hipEventRecord(ev[2], ctx->stream));
-    hipLaunchKernelGGL(k_gor_lds_filtered, dim3(grid_for(ng, GS_GOR_BLOCK)),
-                       dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
-                       f_pages, ng, set->d_sp_start, set->d_sp_cnt,
-                       set->d_out_off, spec->d_out_val, ctx->d_err);
+    HIP_TRY(hipMemsetAsync(set->d_gor_nactive, 0, sizeof(int), ctx->stream));
+    hipLaunchKernelGGL(k_gor_active, dim3(grid_for(nch, 256)), dim3(256), 0,
+                       ctx->stream, fsp.d_gor_chunks, nch, set->d_sp_start,
+                       set->d_sp_cnt, set->d_gor_active, set->d_gor_nactive);
+    hipLaunchKernelGGL(k_gor_chunks_filtered,
+                       dim3(grid_for(nch, GS_GOR_BLOCK)), dim3(GS_GOR_BLOCK),
+                       0, ctx->stream, set->d_blob, fsp.d_gor_chunks,
+                       set->d_gor_active, set->d_gor_nactive, set->d_sp_start,
+                       set->d_sp_cnt, set->d_out_off, spec->d_out_val,
+                       ctx->d_err);
     HIP_TRY(hipEventRecord(ev[3], ctx->stream));
     HIP_TRY(hipEventRecord(ev[4], ctx->stream));
     if (spec->n_buckets > 0) {

@@ -26,6 +26,38 @@ struct DevGroup {
     int32_t pad;
 };
 
+/* Gorilla chunked decode: reference-shaped pages hold up to
+ * max_datablock_size = 102,400 rows (comapcting_block_meta_group.rs:87,
+ * storage_config.rs:136-138) and the Gorilla bitstream is strictly
+ * sequential (float.rs:445-463), so thread-per-page parallelism collapses
+ * on big pages.  At upload, k_gor_sync walks each all-valid Gorilla page
+ * once and records the parser state (bit cursor, previous value, XOR
+ * window) every GOR_CHUNK values; decode then runs one chunk per lane,
+ * and the filtered scan skips chunks wholly outside the selected span
+ * (sub-page pruning — the chunk-granularity analog of the reference's
+ * page min/max pruning, tskv/src/reader/chunk.rs:12-49).  ~32 B of side
+ * table per 4096 values = ~0.1% of typical compressed size. */
+#define GOR_CHUNK 4096
+
+struct DevGorChunk {
+    uint64_t data_off;  /* page data offset in blob */
+    uint64_t bitpos;    /* bits consumed from stream start; 0 for row0==0
+                           (decode reads the page header itself there) */
+    uint64_t val;       /* bits of row (row0-1)'s value (pre-pass) */
+    int64_t row_off;    /* output row offset of the page's group */
+    uint32_t grp;       /* page-group index */
+    uint32_t row0;      /* first row index in page this chunk produces */
+    uint32_t cnt;       /* rows this chunk produces */
+    uint32_t data_len;  /* page data_len */
+    uint8_t trailing;   /* XOR window state at bitpos (pre-pass) */
+    uint8_t meaningful;
+    uint8_t last;       /* final chunk: page-end semantics (sentinel) */
+    uint8_t safe_stop;  /* pre-pass proved the whole chunk decodable (the
+                           next chunk's state was recorded): the filtered
+                           kernel may stop at the span end without losing
+                           the page's error surface */
+};
+
 /* launch-class partition of a column slot's pages */
 enum PageClass {
     PC_SEQ = 0,     /* universal sequential thread-per-page decoder */
