@@ -352,12 +352,25 @@ __device__ __forceinline__ void gorilla_decode_page(
 /* --- Gorilla chunked decode (see DevGorChunk in gs_internal.h) ---
  * k_gor_sync: one-time upload pre-pass.  One thread walks each multi-chunk
  * all-valid Gorilla page with the same window parser as decode (no value
- * stores) and records the parser state every GOR_CHUNK values, so decode
+ * stores) and records the parser state every chunk_rows values, so decode
  * can start mid-stream.  Errors are NOT raised here: a truncated or
  * early-sentinel page poisons its unrecorded chunks (bitpos past the
  * stream end), and the decode of that chunk raises DERR_SHORT exactly
  * where the page-sequential kernel would have — same error surface,
- * surfaced at decode/scan time like the reference's decode errors. */
+ * surfaced at decode/scan time like the reference's decode errors.
+ *
+ * Budget accounting (all Gorilla kernels here): instead of decrementing a
+ * bit budget per value (2 ops/value + 2 checks/value on the round-1
+ * parser), the cursor advances freely and correctness is enforced by
+ * (a) a load clamp: words at/after p_clamp read as zero, so no access
+ *     ever leaves the blob's 48-B tail pad;
+ * (b) an overrun flag: once the window would hold only pad (entering
+ *     word index >= W+2), the lane stops with DERR_SHORT — bounds every
+ *     loop on corrupt input within ~128 pad bits;
+ * (c) exact end checks, evaluated once per chunk (at the sentinel, or on
+ *     completing a non-final chunk): bits consumed > total_bits means the
+ *     parse ran past the real stream -> DERR_SHORT, reproducing the
+ *     reference's "unexpected end of block" (float.rs:462) surface. */
 __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
                            const DevPage *__restrict__ pages, int npages,
                            const int32_t *__restrict__ chunk_base,
@@ -373,35 +386,43 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
         int next_k = 1;
         bool ok = false;
         if (pg.data_len >= 10) {
-            int64_t budget0 = int64_t(pg.data_len - 10) * 8;
+            int64_t total_bits = int64_t(pg.data_len - 10) * 8;
+            const uint8_t *stream = data + 10;
+            uint64_t W = (uint64_t(total_bits) + 63) >> 6;
+            const uint8_t *p_clamp = stream + 8 * W + 24;
+            const uint8_t *p_over = stream + 8 * W + 40;
             uint64_t val = dev_be64(data + 2);
-            const uint8_t *p = data + 10;
-            int64_t budget = budget0;
+            const uint8_t *p = stream;
             uint64_t hi = 0, lo = 0;
             int nb = 0;
             uint32_t trailing = 0, meaningful = 64;
             uint32_t r = 1; /* header value = row 0 */
+            bool over = false;
             uint64_t nextw = dev_be64(p);
-            p += 8;
+            uint64_t nextw2 = dev_be64(p + 8);
+            p += 16;
             auto topup = [&]() { /* only with nb < 64 */
                 uint64_t x = nextw;
-                nextw = dev_be64(p);
+                nextw = nextw2;
+                over |= (p >= p_over);
+                nextw2 = (p < p_clamp) ? dev_be64(p) : 0;
                 p += 8;
                 if (nb == 0) { hi = x; lo = 0; }
                 else { hi |= x >> nb; lo = x << (64 - nb); }
                 nb += 64;
             };
-            auto consume = [&](unsigned k) {
+            auto consume = [&](unsigned k) { /* k in 1..=64 */
                 hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
                 lo = (k == 64) ? 0 : (lo << k);
                 nb -= int(k);
-                budget -= int64_t(k);
             };
             ok = true;
             for (;;) {
                 if (r == uint32_t(next_k) * chunk_rows) {
+                    int64_t used = int64_t(p - stream) * 8 - 128 - nb;
+                    if (used > total_bits) { ok = false; break; }
                     DevGorChunk &c = chunks[cb + next_k];
-                    c.bitpos = uint64_t(budget0 - budget);
+                    c.bitpos = uint64_t(used);
                     c.val = val;
                     c.trailing = uint8_t(trailing);
                     c.meaningful = uint8_t(meaningful);
@@ -412,25 +433,26 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
                     if (next_k == nch) break; /* tail chunk parses itself */
                 }
                 if (nb < 64) topup();
-                if (budget <= 0) { ok = false; break; }
+                if (over) { ok = false; break; }
                 uint32_t top13 = uint32_t(hi >> 51);
                 if (!(top13 & 0x1000)) { /* ctrl 0: repeat */
                     consume(1);
                 } else {
+                    unsigned shift = 2;
                     if (top13 & 0x0800) { /* ctrl 11: new window */
                         uint32_t lead = (top13 >> 6) & 0x1f;
                         meaningful = top13 & 0x3f;
                         if (meaningful > 0) trailing = 64 - lead - meaningful;
                         else { trailing = 0; meaningful = 64; }
-                        consume(13);
-                    } else { /* ctrl 10: reuse window */
-                        consume(2);
+                        shift = 13;
                     }
-                    while (nb < int(meaningful)) topup();
+                    unsigned need = shift + meaningful; /* <= 77 */
+                    while (nb < int(need)) topup();
+                    uint64_t w = (hi << shift) | (lo >> (64 - shift));
                     uint64_t sb =
-                        (meaningful == 64) ? hi : (hi >> (64 - meaningful));
-                    consume(meaningful);
-                    if (budget < 0) { ok = false; break; }
+                        (meaningful == 64) ? w : (w >> (64 - meaningful));
+                    if (need <= 64) consume(need);
+                    else { consume(shift); consume(meaningful); }
                     val ^= sb << trailing;
                     /* sentinel before the last sync point: rows missing */
                     if (val == GORILLA_SENTINEL) { ok = false; break; }
@@ -452,10 +474,13 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
 
 /* Parser window state of one chunk, positioned at DevGorChunk.bitpos. */
 struct GorChunkState {
-    const uint8_t *p; /* next 8-byte word to prefetch */
+    const uint8_t *p;       /* next 8-byte word to prefetch */
+    const uint8_t *stream;  /* bitstream start (page data + 10) */
+    const uint8_t *p_clamp; /* loads at/after here read as zero */
+    const uint8_t *p_over;  /* reaching here = stream overrun */
     uint64_t hi, lo, nextw, nextw2;
     uint64_t val;
-    int64_t budget;
+    int64_t total_bits;
     int nb;
     uint32_t trailing, meaningful;
     bool bad;
@@ -468,26 +493,31 @@ gor_chunk_init(const uint8_t *__restrict__ blob, const DevGorChunk &c) {
     const uint8_t *data = blob + c.data_off;
     /* page data: [enc=6][0x10][first f64 BE][bitstream] (float.rs:418-444) */
     if (c.data_len < 10) { st.bad = true; return st; }
-    int64_t total_bits = int64_t(c.data_len - 10) * 8;
+    const uint8_t *stream = data + 10;
+    st.stream = stream;
+    st.total_bits = int64_t(c.data_len - 10) * 8;
+    uint64_t W = (uint64_t(st.total_bits) + 63) >> 6;
+    st.p_clamp = stream + 8 * W + 24;
+    st.p_over = stream + 8 * W + 40;
     if (c.row0 == 0) {
         st.val = dev_be64(data + 2);
-        st.budget = total_bits;
         st.trailing = 0;
         st.meaningful = 64;
-        const uint8_t *p = data + 10;
         st.hi = 0;
         st.lo = 0;
         st.nb = 0;
-        st.nextw = dev_be64(p);
-        st.nextw2 = dev_be64(p + 8);
-        st.p = p + 16;
+        st.nextw = dev_be64(stream);
+        st.nextw2 = dev_be64(stream + 8);
+        st.p = stream + 16;
     } else {
+        if (int64_t(c.bitpos) >= st.total_bits) { /* poisoned / truncated */
+            st.bad = true;
+            return st;
+        }
         st.val = c.val;
-        st.budget = total_bits - int64_t(c.bitpos);
         st.trailing = c.trailing;
         st.meaningful = c.meaningful;
-        if (st.budget <= 0) { st.bad = true; return st; } /* poisoned */
-        const uint8_t *p = data + 10 + (c.bitpos >> 6) * 8;
+        const uint8_t *p = stream + (c.bitpos >> 6) * 8;
         unsigned rem = unsigned(c.bitpos & 63);
         uint64_t w0 = dev_be64(p);
         p += 8;
@@ -509,9 +539,7 @@ gor_chunk_init(const uint8_t *__restrict__ blob, const DevGorChunk &c) {
  * ~3 ms for the same bytes).  Lanes run in lockstep (one value per
  * iteration), so rings fill together; lanes whose chunk ended keep
  * cooperating in flushes until all are done.  Chunks are page-major in
- * the table, so a wave's lanes hold NEIGHBORING regions of one page:
- * their refill streams share L2 lines and their branch patterns
- * correlate (same series' data), unlike the round-1 page-per-lane shape. */
+ * the table, so a wave's lanes hold NEIGHBORING regions of one page. */
 __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                              const DevGorChunk *__restrict__ chunks,
                              int nchunks, double *__restrict__ out,
@@ -536,23 +564,26 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
         uint32_t end = ch.row0 + ch.cnt;
         GorChunkState st = gor_chunk_init(blob, ch);
         int rfill = 0;
-        bool done = !have;
+        bool done = !have, over = false;
         if (!have) { r = 0; end = 0; }
         if (have && st.bad) { atomicOr(err, DERR_SHORT); done = true; }
         auto topup = [&]() { /* only with nb < 64 */
             uint64_t x = st.nextw;
             st.nextw = st.nextw2;
-            st.nextw2 = dev_be64(st.p);
+            over |= (st.p >= st.p_over);
+            st.nextw2 = (st.p < st.p_clamp) ? dev_be64(st.p) : 0;
             st.p += 8;
             if (st.nb == 0) { st.hi = x; st.lo = 0; }
             else { st.hi |= x >> st.nb; st.lo = x << (64 - st.nb); }
             st.nb += 64;
         };
-        auto consume = [&](unsigned k) {
+        auto consume = [&](unsigned k) { /* k in 1..=64 */
             st.hi = (k == 64) ? st.lo : ((st.hi << k) | (st.lo >> (64 - k)));
             st.lo = (k == 64) ? 0 : (st.lo << k);
             st.nb -= int(k);
-            st.budget -= int64_t(k);
+        };
+        auto used_bits = [&]() {
+            return int64_t(st.p - st.stream) * 8 - 128 - st.nb;
         };
         auto fd = fdesc[wv];
         const int f_idx = lane & (GS_RING - 1);
@@ -586,7 +617,7 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
         while (!__all(done)) {
             if (!done) {
                 if (st.nb < 64) topup();
-                if (st.budget <= 0) { atomicOr(err, DERR_SHORT); done = true; }
+                if (over) { atomicOr(err, DERR_SHORT); done = true; }
             }
             if (!done) {
                 uint32_t top13 = uint32_t(st.hi >> 51);
@@ -594,28 +625,29 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                 if (!(top13 & 0x1000)) {
                     consume(1);
                 } else {
+                    unsigned shift = 2;
                     if (top13 & 0x0800) {
                         uint32_t lead = (top13 >> 6) & 0x1f;
                         st.meaningful = top13 & 0x3f;
                         if (st.meaningful > 0)
                             st.trailing = 64 - lead - st.meaningful;
                         else { st.trailing = 0; st.meaningful = 64; }
-                        consume(13);
-                    } else {
-                        consume(2);
+                        shift = 13;
                     }
-                    while (st.nb < int(st.meaningful)) topup();
+                    unsigned need = shift + st.meaningful; /* <= 77 */
+                    while (st.nb < int(need)) topup();
+                    uint64_t w = (st.hi << shift) | (st.lo >> (64 - shift));
                     uint64_t sb = (st.meaningful == 64)
-                                      ? st.hi
-                                      : (st.hi >> (64 - st.meaningful));
-                    consume(st.meaningful);
-                    if (st.budget < 0) {
-                        atomicOr(err, DERR_SHORT);
+                                      ? w
+                                      : (w >> (64 - st.meaningful));
+                    if (need <= 64) consume(need);
+                    else { consume(shift); consume(st.meaningful); }
+                    st.val ^= sb << st.trailing;
+                    if (st.val == GORILLA_SENTINEL) {
                         done = true;
                         stg = false;
-                    } else {
-                        st.val ^= sb << st.trailing;
-                        if (st.val == GORILLA_SENTINEL) { done = true; stg = false; }
+                        if (used_bits() > st.total_bits)
+                            atomicOr(err, DERR_SHORT);
                     }
                 }
                 if (stg && r < end) {
@@ -626,7 +658,11 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                     /* non-last chunk: all rows produced, stop; the last
                        chunk keeps parsing to the sentinel like the page-
                        sequential kernel (a missing sentinel is an error) */
-                    if (r == end && !ch.last) done = true;
+                    if (r == end && !ch.last) {
+                        done = true;
+                        if (used_bits() > st.total_bits)
+                            atomicOr(err, DERR_SHORT);
+                    }
                 }
             }
             if (__any(rfill == GS_RING)) flush();
@@ -1878,7 +1914,7 @@ __global__ void k_gor_active(const DevGorChunk *__restrict__ chunks,
         bool act = row0 < hi && rend > lo;
         uint64_t m = __ballot(act);
         if (m) {
-            int leader = __ffsll((long long)m) - 1;
+            int leader = __ffsll((unsigned long long)m) - 1;
             int base = 0;
             if (lane == leader) base = atomicAdd(n_active, __popcll(m));
             base = __shfl(base, leader);
@@ -1931,22 +1967,25 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
         GorChunkState st = gor_chunk_init(blob, ch);
         int rfill = 0;
         int64_t run0 = 0; /* output row of first staged entry */
-        bool done = !have;
+        bool done = !have, over = false, clean_stop = false;
         if (have && st.bad) { atomicOr(err, DERR_SHORT); done = true; }
         auto topup = [&]() { /* only with nb < 64 */
             uint64_t x = st.nextw;
             st.nextw = st.nextw2;
-            st.nextw2 = dev_be64(st.p);
+            over |= (st.p >= st.p_over);
+            st.nextw2 = (st.p < st.p_clamp) ? dev_be64(st.p) : 0;
             st.p += 8;
             if (st.nb == 0) { st.hi = x; st.lo = 0; }
             else { st.hi |= x >> st.nb; st.lo = x << (64 - st.nb); }
             st.nb += 64;
         };
-        auto consume = [&](unsigned k) {
+        auto consume = [&](unsigned k) { /* k in 1..=64 */
             st.hi = (k == 64) ? st.lo : ((st.hi << k) | (st.lo >> (64 - k)));
             st.lo = (k == 64) ? 0 : (st.lo << k);
             st.nb -= int(k);
-            st.budget -= int64_t(k);
+        };
+        auto used_bits = [&]() {
+            return int64_t(st.p - st.stream) * 8 - 128 - st.nb;
         };
         auto fd = fdesc[wv];
         /* 4 source lanes per store instruction: lane = (sq:2, idx:4), sq
@@ -1987,14 +2026,19 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
             r++;
             /* all rows produced (non-last chunk), or span passed on a
                chunk the pre-pass proved decodable: stop parsing */
-            if ((r == end && !ch.last) || (r >= sel_hi && ch.safe_stop))
+            if (r == end && !ch.last) {
                 done = true;
+                if (used_bits() > st.total_bits) atomicOr(err, DERR_SHORT);
+            } else if (r >= sel_hi && ch.safe_stop) {
+                done = true;
+                clean_stop = true;
+            }
         };
         if (!done && ch.row0 == 0 && r < end) stage_row(st.val);
         while (!__all(done)) {
             if (!done) {
                 if (st.nb < 64) topup();
-                if (st.budget <= 0) { atomicOr(err, DERR_SHORT); done = true; }
+                if (over) { atomicOr(err, DERR_SHORT); done = true; }
             }
             if (!done) {
                 uint32_t top13 = uint32_t(st.hi >> 51);
@@ -2002,28 +2046,29 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
                 if (!(top13 & 0x1000)) {
                     consume(1);
                 } else {
+                    unsigned shift = 2;
                     if (top13 & 0x0800) {
                         uint32_t lead = (top13 >> 6) & 0x1f;
                         st.meaningful = top13 & 0x3f;
                         if (st.meaningful > 0)
                             st.trailing = 64 - lead - st.meaningful;
                         else { st.trailing = 0; st.meaningful = 64; }
-                        consume(13);
-                    } else {
-                        consume(2);
+                        shift = 13;
                     }
-                    while (st.nb < int(st.meaningful)) topup();
+                    unsigned need = shift + st.meaningful; /* <= 77 */
+                    while (st.nb < int(need)) topup();
+                    uint64_t w = (st.hi << shift) | (st.lo >> (64 - shift));
                     uint64_t sb = (st.meaningful == 64)
-                                      ? st.hi
-                                      : (st.hi >> (64 - st.meaningful));
-                    consume(st.meaningful);
-                    if (st.budget < 0) {
-                        atomicOr(err, DERR_SHORT);
+                                      ? w
+                                      : (w >> (64 - st.meaningful));
+                    if (need <= 64) consume(need);
+                    else { consume(shift); consume(st.meaningful); }
+                    st.val ^= sb << st.trailing;
+                    if (st.val == GORILLA_SENTINEL) {
                         done = true;
                         stg = false;
-                    } else {
-                        st.val ^= sb << st.trailing;
-                        if (st.val == GORILLA_SENTINEL) { done = true; stg = false; }
+                        if (used_bits() > st.total_bits)
+                            atomicOr(err, DERR_SHORT);
                     }
                 }
                 if (stg && r < end) stage_row(st.val);
@@ -2031,8 +2076,7 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
             if (__any(rfill == GS_RING)) flush();
         }
         flush();
-        if (have && r < end && !(r >= sel_hi && ch.safe_stop))
-            atomicOr(err, DERR_SHORT);
+        if (have && r < end && !clean_stop) atomicOr(err, DERR_SHORT);
     }
 }
 
