@@ -447,12 +447,18 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
                         shift = 13;
                     }
                     unsigned need = shift + meaningful; /* <= 77 */
-                    while (nb < int(need)) topup();
-                    uint64_t w = (hi << shift) | (lo >> (64 - shift));
-                    uint64_t sb =
-                        (meaningful == 64) ? w : (w >> (64 - meaningful));
-                    if (need <= 64) consume(need);
-                    else { consume(shift); consume(meaningful); }
+                    uint64_t sb;
+                    if (int(need) <= nb) { /* window holds the whole value */
+                        uint64_t w = (hi << shift) | (lo >> (64 - shift));
+                        sb = (meaningful == 64) ? w : (w >> (64 - meaningful));
+                        if (need <= 64) consume(need);
+                        else { consume(shift); consume(meaningful); }
+                    } else { /* short window: two-step (topup needs nb<64) */
+                        consume(shift);
+                        while (nb < int(meaningful)) topup();
+                        sb = (meaningful == 64) ? hi : (hi >> (64 - meaningful));
+                        consume(meaningful);
+                    }
                     val ^= sb << trailing;
                     /* sentinel before the last sync point: rows missing */
                     if (val == GORILLA_SENTINEL) { ok = false; break; }
@@ -635,13 +641,23 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                         shift = 13;
                     }
                     unsigned need = shift + st.meaningful; /* <= 77 */
-                    while (st.nb < int(need)) topup();
-                    uint64_t w = (st.hi << shift) | (st.lo >> (64 - shift));
-                    uint64_t sb = (st.meaningful == 64)
-                                      ? w
-                                      : (w >> (64 - st.meaningful));
-                    if (need <= 64) consume(need);
-                    else { consume(shift); consume(st.meaningful); }
+                    uint64_t sb;
+                    if (int(need) <= st.nb) { /* window holds the value */
+                        uint64_t w =
+                            (st.hi << shift) | (st.lo >> (64 - shift));
+                        sb = (st.meaningful == 64)
+                                 ? w
+                                 : (w >> (64 - st.meaningful));
+                        if (need <= 64) consume(need);
+                        else { consume(shift); consume(st.meaningful); }
+                    } else { /* short window: two-step (topup needs nb<64) */
+                        consume(shift);
+                        while (st.nb < int(st.meaningful)) topup();
+                        sb = (st.meaningful == 64)
+                                 ? st.hi
+                                 : (st.hi >> (64 - st.meaningful));
+                        consume(st.meaningful);
+                    }
                     st.val ^= sb << st.trailing;
                     if (st.val == GORILLA_SENTINEL) {
                         done = true;
@@ -2056,13 +2072,23 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
                         shift = 13;
                     }
                     unsigned need = shift + st.meaningful; /* <= 77 */
-                    while (st.nb < int(need)) topup();
-                    uint64_t w = (st.hi << shift) | (st.lo >> (64 - shift));
-                    uint64_t sb = (st.meaningful == 64)
-                                      ? w
-                                      : (w >> (64 - st.meaningful));
-                    if (need <= 64) consume(need);
-                    else { consume(shift); consume(st.meaningful); }
+                    uint64_t sb;
+                    if (int(need) <= st.nb) { /* window holds the value */
+                        uint64_t w =
+                            (st.hi << shift) | (st.lo >> (64 - shift));
+                        sb = (st.meaningful == 64)
+                                 ? w
+                                 : (w >> (64 - st.meaningful));
+                        if (need <= 64) consume(need);
+                        else { consume(shift); consume(st.meaningful); }
+                    } else { /* short window: two-step (topup needs nb<64) */
+                        consume(shift);
+                        while (st.nb < int(st.meaningful)) topup();
+                        sb = (st.meaningful == 64)
+                                 ? st.hi
+                                 : (st.hi >> (64 - st.meaningful));
+                        consume(st.meaningful);
+                    }
                     st.val ^= sb << st.trailing;
                     if (st.val == GORILLA_SENTINEL) {
                         done = true;
