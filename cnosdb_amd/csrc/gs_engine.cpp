@@ -628,38 +628,41 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
             if (!done) {
                 uint32_t top13 = uint32_t(st.hi >> 51);
                 bool stg = true;
-                if (!(top13 & 0x1000)) {
-                    consume(1);
-                } else {
-                    unsigned shift = 2;
-                    if (top13 & 0x0800) {
-                        uint32_t lead = (top13 >> 6) & 0x1f;
-                        st.meaningful = top13 & 0x3f;
-                        if (st.meaningful > 0)
-                            st.trailing = 64 - lead - st.meaningful;
-                        else { st.trailing = 0; st.meaningful = 64; }
-                        shift = 13;
-                    }
-                    unsigned need = shift + st.meaningful; /* <= 77 */
+                {
+                    /* uniform path (no repeat/xor branch): a repeat is an
+                       XOR value with 0 meaningful bits (sb=0), so every
+                       lane runs the same straight-line code and the wave
+                       diverges only on the rare need>64 case — measured
+                       the round-1 two-branch parse at ~20% VALU
+                       utilization, divergence being the dominant loss */
+                    unsigned bit0 = (top13 >> 12) & 1; /* 1 = XOR value */
+                    unsigned bit1 = (top13 >> 11) & 1; /* 1 = new window */
+                    unsigned nw = bit0 & bit1;
+                    uint32_t lead = (top13 >> 6) & 0x1f;
+                    uint32_t mg_raw = top13 & 0x3f;
+                    uint32_t mg_new = mg_raw ? mg_raw : 64;
+                    uint32_t tr_new = mg_raw ? (64 - lead - mg_raw) : 0;
+                    st.meaningful = nw ? mg_new : st.meaningful;
+                    st.trailing = nw ? tr_new : st.trailing;
+                    unsigned shift = bit0 ? (bit1 ? 13u : 2u) : 1u;
+                    unsigned m_eff = bit0 ? st.meaningful : 0u;
+                    unsigned need = shift + m_eff;
                     uint64_t sb;
-                    if (int(need) <= st.nb) { /* window holds the value */
+                    if (__builtin_expect(need <= 64, 1)) {
+                        /* nb >= 64 after the loop-top topup */
                         uint64_t w =
                             (st.hi << shift) | (st.lo >> (64 - shift));
-                        sb = (st.meaningful == 64)
-                                 ? w
-                                 : (w >> (64 - st.meaningful));
-                        if (need <= 64) consume(need);
-                        else { consume(shift); consume(st.meaningful); }
-                    } else { /* short window: two-step (topup needs nb<64) */
+                        sb = m_eff ? (w >> ((64 - m_eff) & 63)) : 0;
+                        consume(need);
+                    } else { /* m_eff > 51: rare */
                         consume(shift);
-                        while (st.nb < int(st.meaningful)) topup();
-                        sb = (st.meaningful == 64)
-                                 ? st.hi
-                                 : (st.hi >> (64 - st.meaningful));
-                        consume(st.meaningful);
+                        while (st.nb < int(m_eff)) topup();
+                        sb = (m_eff == 64) ? st.hi
+                                           : (st.hi >> (64 - m_eff));
+                        consume(m_eff);
                     }
                     st.val ^= sb << st.trailing;
-                    if (st.val == GORILLA_SENTINEL) {
+                    if (bit0 && st.val == GORILLA_SENTINEL) {
                         done = true;
                         stg = false;
                         if (used_bits() > st.total_bits)
@@ -2059,38 +2062,41 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
             if (!done) {
                 uint32_t top13 = uint32_t(st.hi >> 51);
                 bool stg = true;
-                if (!(top13 & 0x1000)) {
-                    consume(1);
-                } else {
-                    unsigned shift = 2;
-                    if (top13 & 0x0800) {
-                        uint32_t lead = (top13 >> 6) & 0x1f;
-                        st.meaningful = top13 & 0x3f;
-                        if (st.meaningful > 0)
-                            st.trailing = 64 - lead - st.meaningful;
-                        else { st.trailing = 0; st.meaningful = 64; }
-                        shift = 13;
-                    }
-                    unsigned need = shift + st.meaningful; /* <= 77 */
+                {
+                    /* uniform path (no repeat/xor branch): a repeat is an
+                       XOR value with 0 meaningful bits (sb=0), so every
+                       lane runs the same straight-line code and the wave
+                       diverges only on the rare need>64 case — measured
+                       the round-1 two-branch parse at ~20% VALU
+                       utilization, divergence being the dominant loss */
+                    unsigned bit0 = (top13 >> 12) & 1; /* 1 = XOR value */
+                    unsigned bit1 = (top13 >> 11) & 1; /* 1 = new window */
+                    unsigned nw = bit0 & bit1;
+                    uint32_t lead = (top13 >> 6) & 0x1f;
+                    uint32_t mg_raw = top13 & 0x3f;
+                    uint32_t mg_new = mg_raw ? mg_raw : 64;
+                    uint32_t tr_new = mg_raw ? (64 - lead - mg_raw) : 0;
+                    st.meaningful = nw ? mg_new : st.meaningful;
+                    st.trailing = nw ? tr_new : st.trailing;
+                    unsigned shift = bit0 ? (bit1 ? 13u : 2u) : 1u;
+                    unsigned m_eff = bit0 ? st.meaningful : 0u;
+                    unsigned need = shift + m_eff;
                     uint64_t sb;
-                    if (int(need) <= st.nb) { /* window holds the value */
+                    if (__builtin_expect(need <= 64, 1)) {
+                        /* nb >= 64 after the loop-top topup */
                         uint64_t w =
                             (st.hi << shift) | (st.lo >> (64 - shift));
-                        sb = (st.meaningful == 64)
-                                 ? w
-                                 : (w >> (64 - st.meaningful));
-                        if (need <= 64) consume(need);
-                        else { consume(shift); consume(st.meaningful); }
-                    } else { /* short window: two-step (topup needs nb<64) */
+                        sb = m_eff ? (w >> ((64 - m_eff) & 63)) : 0;
+                        consume(need);
+                    } else { /* m_eff > 51: rare */
                         consume(shift);
-                        while (st.nb < int(st.meaningful)) topup();
-                        sb = (st.meaningful == 64)
-                                 ? st.hi
-                                 : (st.hi >> (64 - st.meaningful));
-                        consume(st.meaningful);
+                        while (st.nb < int(m_eff)) topup();
+                        sb = (m_eff == 64) ? st.hi
+                                           : (st.hi >> (64 - m_eff));
+                        consume(m_eff);
                     }
                     st.val ^= sb << st.trailing;
-                    if (st.val == GORILLA_SENTINEL) {
+                    if (bit0 && st.val == GORILLA_SENTINEL) {
                         done = true;
                         stg = false;
                         if (used_bits() > st.total_bits)
