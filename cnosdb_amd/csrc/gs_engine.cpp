@@ -620,6 +620,7 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
             rfill++;
             r++;
         }
+        unsigned it = 0;
         while (!__all(done)) {
             if (!done) {
                 if (st.nb < 64) topup();
@@ -684,7 +685,10 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                     }
                 }
             }
-            if (__any(rfill == GS_RING)) flush();
+            /* uniform cadence: <=1 staged/iteration, so a flush every
+               GS_RING iterations can never overflow the ring — no ballot,
+               no divergent branch */
+            if ((++it & (GS_RING - 1)) == 0) flush();
         }
         flush();
         if (have && r < end) atomicOr(err, DERR_SHORT);
@@ -2054,6 +2058,7 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
             }
         };
         if (!done && ch.row0 == 0 && r < end) stage_row(st.val);
+        unsigned it = 0;
         while (!__all(done)) {
             if (!done) {
                 if (st.nb < 64) topup();
@@ -2105,7 +2110,7 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
                 }
                 if (stg && r < end) stage_row(st.val);
             }
-            if (__any(rfill == GS_RING)) flush();
+            if ((++it & (GS_RING - 1)) == 0) flush();
         }
         flush();
         if (have && r < end && !clean_stop) atomicOr(err, DERR_SHORT);
