@@ -620,7 +620,7 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
             rfill++;
             r++;
         }
-        unsigned it = 0;
+        unsigned it = 1; /* header stage counts toward the first cadence */
         while (!__all(done)) {
             if (!done) {
                 if (st.nb < 64) topup();
@@ -2058,7 +2058,7 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
             }
         };
         if (!done && ch.row0 == 0 && r < end) stage_row(st.val);
-        unsigned it = 0;
+        unsigned it = 1; /* header stage counts toward the first cadence */
         while (!__all(done)) {
             if (!done) {
                 if (st.nb < 64) topup();
