@@ -37,7 +37,7 @@ struct DevGroup {
  * (sub-page pruning — the chunk-granularity analog of the reference's
  * page min/max pruning, tskv/src/reader/chunk.rs:12-49).  ~32 B of side
  * table per 4096 values = ~0.1% of typical compressed size. */
-#define GOR_CHUNK 1024
+#define GOR_CHUNK 2048
 
 struct DevGorChunk {
     uint64_t data_off;  /* page data offset in blob */

@@ -20,7 +20,7 @@ pytestmark = pytest.mark.gpu
 
 rng = np.random.default_rng(20250915)
 
-GOR_CHUNK = 1024  # keep in sync with gs_internal.h
+GOR_CHUNK = 2048  # keep in sync with gs_internal.h
 
 
 @pytest.fixture(scope="module")
