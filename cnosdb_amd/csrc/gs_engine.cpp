@@ -3466,21 +3466,23 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
     }
     HIP_TRY(hipEventRecord(ev[3], ctx->stream));
 
-    int64_t out_rows = 0;
     bool want_rows = spec->d_out_ts && spec->d_out_val;
-    if (want_rows || d_mask) {
-        std::vector<int64_t> cnt(set->ngroups), off(set->ngroups);
-        HIP_TRY(hipStreamSynchronize(ctx->stream));
-        HIP_TRY(hipMemcpy(cnt.data(), d_counts,
-                          set->ngroups * sizeof(int64_t),
-                          hipMemcpyDeviceToHost));
-        int64_t acc = 0;
-        for (size_t g = 0; g < set->ngroups; g++) { off[g] = acc; acc += cnt[g]; }
-        out_rows = acc;
+    {
+        /* device exclusive scan of selected counts -> offsets + total:
+           no host round-trip mid-pipeline (the round-1 version
+           synchronized here and prefix-summed on the host) */
+        int nblocks = (ng + SCAN_BLOCK * SCAN_ITEMS - 1) /
+                      (SCAN_BLOCK * SCAN_ITEMS);
+        if (nblocks > 2048)
+            return fail(GS_ERR, "too many groups for the span scan");
+        hipLaunchKernelGGL(k_scan_partials, dim3(nblocks), dim3(SCAN_BLOCK),
+                           0, ctx->stream, d_counts, ng, set->d_out_off,
+                           set->d_blocksums);
+        hipLaunchKernelGGL(k_scan_fixup, dim3(1), dim3(64), 0, ctx->stream,
+                           ng, set->d_out_off, set->d_blocksums, nblocks);
+        hipLaunchKernelGGL(k_scan_add, dim3(grid_for(ng, 256)), dim3(256), 0,
+                           ctx->stream, ng, set->d_out_off, set->d_blocksums);
         if (want_rows) {
-            HIP_TRY(hipMemcpyAsync(set->d_out_off, off.data(),
-                                   set->ngroups * sizeof(int64_t),
-                                   hipMemcpyHostToDevice, ctx->stream));
             if (d_mask)
                 hipLaunchKernelGGL(k_compact_masked,
                                    dim3(ng > 2048 ? 2048 : ng), dim3(256), 0,
@@ -3547,14 +3549,9 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
     HIP_TRY(hipEventElapsedTime(&ms, ev[4], ev[5]));
     result->ms_agg = ms;
 
-    if (!(spec->d_out_ts && spec->d_out_val)) {
-        /* still need out_rows for the result: sum span counts */
-        std::vector<int64_t> cnt(set->ngroups);
-        HIP_TRY(hipMemcpy(cnt.data(), set->d_sp_cnt,
-                          set->ngroups * sizeof(int64_t),
-                          hipMemcpyDeviceToHost));
-        for (size_t g = 0; g < set->ngroups; g++) out_rows += cnt[g];
-    }
+    int64_t out_rows = 0;
+    HIP_TRY(hipMemcpy(&out_rows, set->d_out_off + set->ngroups,
+                      sizeof(int64_t), hipMemcpyDeviceToHost));
     result->out_rows = out_rows;
     result->decoded_rows = set->total_rows;
     return GS_OK;
