@@ -488,9 +488,192 @@ def bench_strings(args):
     eng.close()
 
 
+
+def cpu_baseline_general(nseries_sample, npts, page_rows, lo, hi, tstones,
+                         pred_a, seed=231):
+    """Oracle C decode (real null bitsets) + numpy filter/tombstone/pred/
+    compact legs over a bounded general-shape sample (kind="port")."""
+    import ctypes
+    from oracle import pyoracle as orc
+    import cnosdb_amd as gs
+
+    rng = np.random.default_rng(seed)
+    npages = npts // page_rows
+    nb = (page_rows + 7) // 8
+    full = np.full(nb, 0xFF, dtype=np.uint8)
+    ts_datas = []
+    for p in range(npages):
+        ts = T0 + (np.arange(page_rows, dtype=np.int64) + p * page_rows) * NS
+        ts_datas.append(gs.encode_ts(ts))
+    val_datas, bitsets = [], []
+    for s in range(nseries_sample):
+        walk = np.round(np.clip(np.cumsum(rng.normal(0, 0.5, npts)) + 50,
+                                0, 100), 1)
+        valid = rng.random(npts) > 0.10
+        for p in range(npages):
+            sl = slice(p * page_rows, (p + 1) * page_rows)
+            v, m = walk[sl], valid[sl]
+            val_datas.append(gs.encode_f64(v[m]) if m.any() else b"")
+            bitsets.append(np.packbits(m, bitorder="little"))
+
+    o = orc.Oracle().lib
+
+    class PD(ctypes.Structure):
+        _fields_ = [("data", ctypes.c_void_p), ("data_len", ctypes.c_uint64),
+                    ("bitset", ctypes.c_void_p), ("nrows", ctypes.c_int64),
+                    ("out_off", ctypes.c_uint64), ("ctype", ctypes.c_uint8)]
+
+    total_pages = nseries_sample * npages
+    descs_v = (PD * total_pages)()
+    keep = []
+    for i, d in enumerate(val_datas):
+        a = np.frombuffer(d, dtype=np.uint8)
+        keep.append(a)
+        descs_v[i] = PD(a.ctypes.data if a.size else None, a.size,
+                        bitsets[i].ctypes.data, page_rows, i * page_rows, 1)
+    descs_t = (PD * total_pages)()
+    for s in range(nseries_sample):
+        for p in range(npages):
+            i = s * npages + p
+            a = np.frombuffer(ts_datas[p], dtype=np.uint8)
+            keep.append(a)
+            descs_t[i] = PD(a.ctypes.data, a.size, full.ctypes.data,
+                            page_rows, i * page_rows, 0)
+    rows = total_pages * page_rows
+    out_ts = np.zeros(rows, dtype=np.int64)
+    out_v = np.zeros(rows, dtype=np.float64)
+    cores = _host_threads()
+    t = time.perf_counter()
+    st = o.orc_decode_pages_omp(descs_t, total_pages,
+                                out_ts.ctypes.data_as(ctypes.c_void_p), cores)
+    assert st == 0
+    st = o.orc_decode_pages_omp(descs_v, total_pages,
+                                out_v.ctypes.data_as(ctypes.c_void_p), cores)
+    assert st == 0
+    # numpy legs: per-series time span + tombstone clear + value pred +
+    # compact + 5-min bucket agg (reader.rs:634-656 / filter.rs:91-142)
+    ots = out_ts.reshape(nseries_sample, npts)
+    ov = out_v.reshape(nseries_sample, npts)
+    valid_all = np.concatenate(
+        [np.unpackbits(b, bitorder="little")[:page_rows] for b in bitsets]
+    ).reshape(nseries_sample, npts).astype(bool)
+    nbuckets = int(npts * NS // BUCKET_NS) + 1
+    for s in range(nseries_sample):
+        a = np.searchsorted(ots[s], lo, side="left")
+        b = np.searchsorted(ots[s], hi, side="right")
+        tsl, vsl, msl = ots[s][a:b], ov[s][a:b], valid_all[s][a:b].copy()
+        for t0_, t1_ in tstones:
+            x = np.searchsorted(tsl, t0_, side="left")
+            y = np.searchsorted(tsl, t1_, side="right")
+            msl[x:y] = False
+        sel = msl & (vsl > pred_a)
+        ts_f, v_f = tsl[sel], vsl[sel]
+        bi = ((ts_f - T0) // BUCKET_NS).astype(np.int64)
+        np.bincount(bi, weights=v_f, minlength=nbuckets)      # sum leg
+        cnts = np.bincount(bi, minlength=nbuckets)            # count leg
+        if v_f.size:  # max leg: bi is nondecreasing (ts sorted)
+            offs_ = np.searchsorted(bi, np.flatnonzero(cnts))
+            np.maximum.reduceat(v_f, offs_)
+    dt = time.perf_counter() - t
+    values = nseries_sample * npts
+    return {"value": values / dt, "unit": "values/s", "cores": cores,
+            "kind": "port",
+            "sample": f"{nseries_sample} series x {npts} pts general shape "
+                      f"(10% nulls, 2 tombstones, pred; oracle C+OpenMP + numpy legs)"}
+
+
+def bench_general(args):
+    """General-shape scan: 10% nulls + 2 tombstone
+    ranges + value predicate v>50 — the non-TSBS shape that takes the
+    general gs_scan path (sequential null-scatter decode, decoded-ts span
+    search, value mask, masked compact, k_agg_partial).  Secondary
+    benchmark line."""
+    import torch
+    import cnosdb_amd as gs
+    rng = np.random.default_rng(231)
+    nseries, npts, page_rows = args.series, args.npts, args.page_rows
+    npages = npts // page_rows
+    lo = T0 + int(0.25 * npts) * NS
+    hi = T0 + int(0.75 * npts) * NS - 1
+    nbuckets = int(npts * NS // BUCKET_NS) + 1
+    tstones = [(T0 + int(0.30 * npts) * NS, T0 + int(0.33 * npts) * NS - 1),
+               (T0 + int(0.60 * npts) * NS, T0 + int(0.63 * npts) * NS - 1)]
+    pred = ("gt", 50.0)
+    eng = gs.Engine(0)
+    t_setup = time.perf_counter()
+    uniq = min(args.unique, nseries)
+    tpages = [gs.page_of(
+        T0 + (np.arange(page_rows, dtype=np.int64) + p * page_rows) * NS,
+        gs.CT_TIME) for p in range(npages)]
+    vpool = []
+    for _ in range(uniq):
+        walk = np.round(np.clip(
+            np.cumsum(rng.normal(0, 0.5, page_rows)) + 50, 0, 100), 1)
+        valid = rng.random(page_rows) > 0.10
+        vpool.append(gs.page_of(walk, gs.CT_F64, valid))
+    groups = []
+    for s in range(nseries):
+        for p in range(npages):
+            groups.append((s, [(tpages[p], gs.CT_TIME),
+                               (vpool[(s * npages + p) % uniq], gs.CT_F64)]))
+    gset = eng.upload(groups)
+    rows = gset.rows
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    agg = dict(bucket_ns=BUCKET_NS, t0=T0, n_buckets=nbuckets,
+               d_max=torch.full((nbuckets,), -np.inf, dtype=torch.float64,
+                                device="cuda"),
+               d_sum=torch.zeros(nbuckets, dtype=torch.float64, device="cuda"),
+               d_count=torch.zeros(nbuckets, dtype=torch.int64, device="cuda"))
+    setup_s = time.perf_counter() - t_setup
+
+    def step():
+        return eng.scan(gset, d_ts, d_val, time_range=(lo, hi),
+                        tombstones=tstones, d_out_ts=d_ots, d_out_val=d_oval,
+                        agg=agg, value_pred=pred)
+
+    for _ in range(args.warmup):
+        r = step()
+    torch.cuda.synchronize()
+    t0_ = time.perf_counter()
+    for _ in range(args.steps):
+        r = step()
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0_
+    value = rows * args.steps / dt
+    cpu_baseline = None
+    if not args.skip_cpu_baseline:
+        cpu_baseline = cpu_baseline_general(16, npts, page_rows, lo, hi,
+                                            tstones, pred[1])
+    line = {
+        "metric": "decoded+filtered values/sec & HBM GB/s, TSBS devops scan, 1/2/4/8 GPU",
+        "value": value, "unit": "values/s", "n_gpus": 1,
+        "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1000,
+        "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+        "dtype": "f64", "data": "synthetic",
+        "config": {
+            "workload": "general-shape scan (10% nulls + 2 tombstone ranges "
+                        "+ value pred v>50; general path, 1 stream)",
+            "series_per_gpu": nseries, "points_per_series": npts,
+            "page_rows": page_rows, "out_rows_per_step": int(r.out_rows),
+            "setup_s": round(setup_s, 1),
+            "phase_ms": {"decode_ts": r.ms_decode_ts,
+                         "decode_f64": r.ms_decode_val,
+                         "filter": r.ms_filter, "compact": r.ms_compact,
+                         "agg": r.ms_agg},
+        },
+        "roofline": None,
+        "cpu_baseline": cpu_baseline,
+    }
+    print(json.dumps(line))
+
+
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--mode", choices=["scan", "compact", "strings"], default="scan")
+    ap.add_argument("--mode", choices=["scan", "compact", "strings", "general"], default="scan")
     ap.add_argument("--gpus", type=int, default=1,
                     help="driver contract flag; the actual world size comes "
                          "from the torchrun environment (WORLD_SIZE)")
@@ -515,6 +698,11 @@ def main():
         return
     if args.mode == "strings":
         bench_strings(args)
+        return
+    if args.mode == "general":
+        if args.series == 10000:
+            args.series = 2500  # one resident set (ts+val+outs+masks ~34 B/row)
+        bench_general(args)
         return
 
     import torch
