@@ -428,7 +428,7 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
                     c.meaningful = uint8_t(meaningful);
                     /* the previous chunk's whole range is now known
                        decodable: the filtered kernel may stop it early */
-                    chunks[cb + next_k - 1].safe_stop = 1;
+                    chunks[cb + next_k - 1].flags = GORF_SAFE_STOP;
                     next_k++;
                     if (next_k == nch) break; /* tail chunk parses itself */
                 }
@@ -473,6 +473,7 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
                 c.val = 0;
                 c.trailing = 0;
                 c.meaningful = 64;
+                c.flags = 0;
             }
         }
     }
@@ -549,7 +550,6 @@ gor_chunk_init(const uint8_t *__restrict__ blob, const DevGorChunk &c) {
 __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                              const DevGorChunk *__restrict__ chunks,
                              int nchunks, double *__restrict__ out,
-                             uint8_t *__restrict__ valid,
                              unsigned *__restrict__ err) {
     __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1];
     __shared__ uint64_t fdesc[GS_GOR_BLOCK / 64][64][2];
@@ -692,10 +692,317 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
         }
         flush();
         if (have && r < end) atomicOr(err, DERR_SHORT);
-        if (have && valid) {
-            uint8_t *vd = valid + ch.row_off;
-            for (uint32_t k = ch.row0; k < end; k++) vd[k] = 1;
+    }
+}
+
+/* --- Gorilla chunked decode for NULL-carrying pages (PC_GORN) ---
+ * The encoded stream holds only non-null values, scattered to set bits of
+ * the validity bitset (tsm/reader.rs:763-825 / float.rs decode via
+ * bitset-driven builder).  The parser state is pipelined ONE VALUE AHEAD:
+ * `val` always holds the value pending for the next set-bit row, so a
+ * chunk boundary at an arbitrary row needs only (bitpos, pending val,
+ * window).  The page header's first f64 is the initial pending value.
+ * Chunks are row-ranges; null rows stage 0.0 through the same LDS ring
+ * (arrow append_null semantics); validity BYTES are produced separately
+ * by k_valid_expand (coalesced), not by this kernel. */
+__global__ void k_gor_sync_null(const uint8_t *__restrict__ blob,
+                                const DevPage *__restrict__ pages, int npages,
+                                const int32_t *__restrict__ chunk_base,
+                                DevGorChunk *__restrict__ chunks,
+                                uint32_t chunk_rows) {
+    for (int pi = blockIdx.x * blockDim.x + threadIdx.x; pi < npages;
+         pi += gridDim.x * blockDim.x) {
+        DevPage pg = pages[pi];
+        int cb = chunk_base[pi];
+        int nch = chunk_base[pi + 1] - cb;
+        if (nch <= 1) continue;
+        const uint8_t *data = blob + pg.data_off;
+        const uint8_t *bs = blob + pg.bitset_off;
+        int next_k = 1;
+        bool ok = false;
+        bool sent_seen = false;
+        if (pg.data_len >= 10) {
+            int64_t total_bits = int64_t(pg.data_len - 10) * 8;
+            const uint8_t *stream = data + 10;
+            uint64_t W = (uint64_t(total_bits) + 63) >> 6;
+            const uint8_t *p_clamp = stream + 8 * W + 24;
+            const uint8_t *p_over = stream + 8 * W + 40;
+            uint64_t val = dev_be64(data + 2); /* pending for 1st set row */
+            const uint8_t *p = stream;
+            uint64_t hi = 0, lo = 0;
+            int nb = 0;
+            uint32_t trailing = 0, meaningful = 64;
+            uint32_t r = 0; /* ROW cursor */
+            bool over = false;
+            uint64_t nextw = dev_be64(p);
+            uint64_t nextw2 = dev_be64(p + 8);
+            p += 16;
+            auto topup = [&]() {
+                uint64_t x = nextw;
+                nextw = nextw2;
+                over |= (p >= p_over);
+                nextw2 = (p < p_clamp) ? dev_be64(p) : 0;
+                p += 8;
+                if (nb == 0) { hi = x; lo = 0; }
+                else { hi |= x >> nb; lo = x << (64 - nb); }
+                nb += 64;
+            };
+            auto consume = [&](unsigned k) {
+                hi = (k == 64) ? lo : ((hi << k) | (lo >> (64 - k)));
+                lo = (k == 64) ? 0 : (lo << k);
+                nb -= int(k);
+            };
+            ok = true;
+            for (;;) {
+                if (r == uint32_t(next_k) * chunk_rows) {
+                    int64_t used = int64_t(p - stream) * 8 - 128 - nb;
+                    if (!sent_seen && used > total_bits) { ok = false; break; }
+                    DevGorChunk &c = chunks[cb + next_k];
+                    c.bitpos = uint64_t(used);
+                    c.val = val;
+                    c.trailing = uint8_t(trailing);
+                    c.meaningful = uint8_t(meaningful);
+                    c.flags = sent_seen ? GORF_SENT_SEEN : 0;
+                    next_k++;
+                    if (next_k == nch) break;
+                    if (sent_seen) continue; /* tail: just mark chunks */
+                }
+                if (sent_seen) { r++; continue; } /* walk rows to boundary */
+                if (nb < 64) topup();
+                if (over) { ok = false; break; }
+                if (dev_bit(bs, r)) {
+                    /* this set row consumes the pending value: parse the
+                       next one (uniform path, see k_gor_chunks) */
+                    uint32_t top13 = uint32_t(hi >> 51);
+                    unsigned bit0 = (top13 >> 12) & 1;
+                    unsigned bit1 = (top13 >> 11) & 1;
+                    unsigned nw = bit0 & bit1;
+                    uint32_t lead = (top13 >> 6) & 0x1f;
+                    uint32_t mg_raw = top13 & 0x3f;
+                    uint32_t mg_new = mg_raw ? mg_raw : 64;
+                    uint32_t tr_new = mg_raw ? (64 - lead - mg_raw) : 0;
+                    meaningful = nw ? mg_new : meaningful;
+                    trailing = nw ? tr_new : trailing;
+                    unsigned shift = bit0 ? (bit1 ? 13u : 2u) : 1u;
+                    unsigned m_eff = bit0 ? meaningful : 0u;
+                    unsigned need = shift + m_eff;
+                    uint64_t sb;
+                    if (need <= 64) {
+                        uint64_t w = (hi << shift) | (lo >> (64 - shift));
+                        sb = m_eff ? (w >> ((64 - m_eff) & 63)) : 0;
+                        consume(need);
+                    } else {
+                        consume(shift);
+                        while (nb < int(m_eff)) topup();
+                        sb = (m_eff == 64) ? hi : (hi >> (64 - m_eff));
+                        consume(m_eff);
+                    }
+                    val ^= sb << trailing;
+                    if (bit0 && val == GORILLA_SENTINEL) {
+                        int64_t used = int64_t(p - stream) * 8 - 128 - nb;
+                        if (used > total_bits) { ok = false; break; }
+                        sent_seen = true;
+                    }
+                }
+                r++;
+            }
         }
+        if (!ok) { /* poison unrecorded chunks -> DERR_SHORT at decode */
+            for (int k = next_k; k < nch; k++) {
+                DevGorChunk &c = chunks[cb + k];
+                c.bitpos = uint64_t(pg.data_len) * 8;
+                c.val = 0;
+                c.trailing = 0;
+                c.meaningful = 64;
+                c.flags = 0;
+            }
+        }
+    }
+}
+
+/* Chunk-parallel decode of null-carrying Gorilla pages: per iteration one
+ * ROW is staged (parsed pending value for set bits, 0.0 for nulls) and,
+ * when the row consumed a value, the next value is parsed ahead.  Same
+ * LDS ring/flush as k_gor_chunks. */
+__global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
+                                  const DevGorChunk *__restrict__ chunks,
+                                  int nchunks, double *__restrict__ out,
+                                  unsigned *__restrict__ err) {
+    __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1];
+    __shared__ uint64_t fdesc[GS_GOR_BLOCK / 64][64][2];
+    const int lane = threadIdx.x & 63;
+    const int wv = threadIdx.x >> 6;
+    auto rslot = ring[wv];
+    int stride = gridDim.x * blockDim.x;
+    int base_id = blockIdx.x * blockDim.x + threadIdx.x;
+    int rounds = (nchunks + stride - 1) / stride;
+    for (int rd = 0; rd < rounds; rd++) {
+        int ci = base_id + rd * stride;
+        bool have = ci < nchunks;
+        DevGorChunk ch = chunks[have ? ci : 0];
+        const uint8_t *bs = blob + ch.bitset_off;
+        double *o = out + ch.row_off;
+        uint32_t r = ch.row0;
+        uint32_t end = ch.row0 + ch.cnt;
+        bool sent_seen = (ch.flags & GORF_SENT_SEEN) != 0;
+        GorChunkState st;
+        st.bad = false;
+        if (sent_seen) { /* tail chunk after the sentinel: nulls only */
+            st.p = st.stream = st.p_clamp = st.p_over = nullptr;
+            st.hi = st.lo = st.nextw = st.nextw2 = st.val = 0;
+            st.total_bits = 0;
+            st.nb = 0;
+            st.trailing = 0;
+            st.meaningful = 64;
+        } else {
+            st = gor_chunk_init(blob, ch);
+        }
+        int rfill = 0;
+        bool done = !have, over = false;
+        if (!have) { r = 0; end = 0; }
+        if (have && st.bad) { atomicOr(err, DERR_SHORT); done = true; }
+        auto topup = [&]() {
+            uint64_t x = st.nextw;
+            st.nextw = st.nextw2;
+            over |= (st.p >= st.p_over);
+            st.nextw2 = (st.p < st.p_clamp) ? dev_be64(st.p) : 0;
+            st.p += 8;
+            if (st.nb == 0) { st.hi = x; st.lo = 0; }
+            else { st.hi |= x >> st.nb; st.lo = x << (64 - st.nb); }
+            st.nb += 64;
+        };
+        auto consume = [&](unsigned k) {
+            st.hi = (k == 64) ? st.lo : ((st.hi << k) | (st.lo >> (64 - k)));
+            st.lo = (k == 64) ? 0 : (st.lo << k);
+            st.nb -= int(k);
+        };
+        auto used_bits = [&]() {
+            return int64_t(st.p - st.stream) * 8 - 128 - st.nb;
+        };
+        auto parse_one = [&]() { /* uniform path; updates st / sent_seen */
+            uint32_t top13 = uint32_t(st.hi >> 51);
+            unsigned bit0 = (top13 >> 12) & 1;
+            unsigned bit1 = (top13 >> 11) & 1;
+            unsigned nw = bit0 & bit1;
+            uint32_t lead = (top13 >> 6) & 0x1f;
+            uint32_t mg_raw = top13 & 0x3f;
+            uint32_t mg_new = mg_raw ? mg_raw : 64;
+            uint32_t tr_new = mg_raw ? (64 - lead - mg_raw) : 0;
+            st.meaningful = nw ? mg_new : st.meaningful;
+            st.trailing = nw ? tr_new : st.trailing;
+            unsigned shift = bit0 ? (bit1 ? 13u : 2u) : 1u;
+            unsigned m_eff = bit0 ? st.meaningful : 0u;
+            unsigned need = shift + m_eff;
+            uint64_t sb;
+            if (__builtin_expect(need <= 64, 1)) {
+                uint64_t w = (st.hi << shift) | (st.lo >> (64 - shift));
+                sb = m_eff ? (w >> ((64 - m_eff) & 63)) : 0;
+                consume(need);
+            } else {
+                consume(shift);
+                while (st.nb < int(m_eff)) topup();
+                sb = (m_eff == 64) ? st.hi : (st.hi >> (64 - m_eff));
+                consume(m_eff);
+            }
+            st.val ^= sb << st.trailing;
+            if (bit0 && st.val == GORILLA_SENTINEL) {
+                if (used_bits() > st.total_bits) atomicOr(err, DERR_SHORT);
+                sent_seen = true;
+            }
+        };
+        auto fd = fdesc[wv];
+        const int f_idx = lane & (GS_RING - 1);
+        const int f_sq = lane / GS_RING;
+        auto flush = [&]() {
+            fd[lane][0] = (uint64_t)(uintptr_t)(o + (int64_t(r) - rfill));
+            fd[lane][1] = uint64_t(rfill);
+            __builtin_amdgcn_wave_barrier();
+            constexpr int SRCP = 64 / GS_RING;
+            for (int src0 = 0; src0 < 64; src0 += SRCP * 4) {
+                double vbuf[4];
+                uint64_t ob[4];
+                int cnt[4];
+                for (int t = 0; t < 4; t++) {
+                    int src = src0 + f_sq + t * SRCP;
+                    vbuf[t] = rslot[f_idx][src];
+                    ob[t] = fd[src][0];
+                    cnt[t] = int(fd[src][1]);
+                }
+                for (int t = 0; t < 4; t++)
+                    if (f_idx < cnt[t])
+                        ((double *)(uintptr_t)ob[t])[f_idx] = vbuf[t];
+            }
+            rfill = 0;
+        };
+        unsigned it = 0;
+        while (!__all(done)) {
+            if (!done && !sent_seen) {
+                if (st.nb < 64) topup();
+                if (over) { atomicOr(err, DERR_SHORT); done = true; }
+            }
+            if (!done) {
+                if (r < end) {
+                    int bit = dev_bit(bs, r);
+                    uint64_t sbits = 0;
+                    bool hole = false; /* set row after the sentinel */
+                    if (bit) {
+                        if (sent_seen) {
+                            atomicOr(err, DERR_SHORT);
+                            done = true;
+                            hole = true;
+                        } else {
+                            sbits = st.val;
+                        }
+                    }
+                    if (!hole) {
+                        rslot[rfill][lane] =
+                            __longlong_as_double((long long)sbits);
+                        rfill++;
+                        r++;
+                        if (bit && !sent_seen) parse_one();
+                        if (r == end) {
+                            if (!ch.last) {
+                                done = true;
+                                if (!sent_seen &&
+                                    used_bits() > st.total_bits)
+                                    atomicOr(err, DERR_SHORT);
+                            }
+                            /* last chunk: keep draining to the sentinel */
+                        }
+                    }
+                } else { /* last chunk drain: parse to the sentinel */
+                    if (sent_seen) done = true;
+                    else parse_one();
+                }
+            }
+            if ((++it & (GS_RING - 1)) == 0) flush();
+        }
+        flush();
+        if (have && r < end) atomicOr(err, DERR_SHORT);
+    }
+}
+
+/* validity bytes, decoupled from the decode kernels (coalesced):
+ * k_valid_fill: vd=1 over all-valid pages; k_valid_expand: vd=bitset. */
+__global__ void k_valid_fill(const DevPage *__restrict__ pages, int npages,
+                             uint8_t *__restrict__ valid) {
+    for (int p = blockIdx.x; p < npages; p += gridDim.x) {
+        DevPage pg = pages[p];
+        uint8_t *vd = valid + pg.row_off;
+        for (uint32_t r = threadIdx.x; r < pg.nrows; r += blockDim.x)
+            vd[r] = 1;
+    }
+}
+
+__global__ void k_valid_expand(const uint8_t *__restrict__ blob,
+                               const DevPage *__restrict__ pages, int npages,
+                               uint8_t *__restrict__ valid) {
+    for (int p = blockIdx.x; p < npages; p += gridDim.x) {
+        DevPage pg = pages[p];
+        const uint8_t *bs = blob + pg.bitset_off;
+        uint8_t *vd = valid + pg.row_off;
+        for (uint32_t r = threadIdx.x; r < pg.nrows; r += blockDim.x)
+            vd[r] = uint8_t(dev_bit(bs, r));
     }
 }
 
@@ -1420,7 +1727,16 @@ __global__ void k_agg_partial_rle(const DevGroup *__restrict__ sg, int nsg,
             }
             const int64_t n2 = (e - s - head) >> 1;
             const double2 *vp = (const double2 *)(v + s + head);
-            for (int64_t q = lane; q < n2; q += 64) {
+            int64_t q = lane;
+            for (; q + 64 < n2; q += 128) { /* 2 loads in flight per iter */
+                double2 x0 = vp[q], x1 = vp[q + 64];
+                if (x0.x > mx) mx = x0.x;
+                if (x0.y > mx) mx = x0.y;
+                if (x1.x > mx) mx = x1.x;
+                if (x1.y > mx) mx = x1.y;
+                sm += (x0.x + x0.y) + (x1.x + x1.y);
+            }
+            if (q < n2) {
                 double2 x = vp[q];
                 if (x.x > mx) mx = x.x;
                 if (x.y > mx) mx = x.y;
@@ -2060,7 +2376,7 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
             if (r == end && !ch.last) {
                 done = true;
                 if (used_bits() > st.total_bits) atomicOr(err, DERR_SHORT);
-            } else if (r >= sel_hi && ch.safe_stop) {
+            } else if (r >= sel_hi && (ch.flags & GORF_SAFE_STOP)) {
                 done = true;
                 clean_stop = true;
             }
@@ -2452,6 +2768,9 @@ struct SlotPages {
     DevGorChunk *d_gor_chunks = nullptr;
     int n_gor_chunks = 0;
     int32_t *d_gor_chunk_base = nullptr; /* [n[PC_GOR]+1] */
+    DevGorChunk *d_gorn_chunks = nullptr;
+    int n_gorn_chunks = 0;
+    int32_t *d_gorn_chunk_base = nullptr; /* [n[PC_GORN]+1] */
 };
 
 struct GsGroupSet {
@@ -2752,9 +3071,9 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                    (hp.dp.enc == GS_ENC_DELTATS || hp.dp.enc == GS_ENC_DELTA) &&
                    hp.dp.data_len > 2) {
             cls = PC_S8B;
-        } else if (hp.dp.all_valid && hp.dp.enc == GS_ENC_GORILLA &&
+        } else if (hp.dp.enc == GS_ENC_GORILLA &&
                    hp.dp.ctype == GS_CT_F64 && hp.dp.data_len >= 10) {
-            cls = PC_GOR;
+            cls = hp.dp.all_valid ? PC_GOR : PC_GORN;
         } else if (hp.dp.all_valid && hp.dp.enc == GS_ENC_NULL &&
                    hp.dp.ctype != GS_CT_BOOL && hp.dp.data_len >= 1) {
             cls = PC_RAW;
@@ -2793,25 +3112,29 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                            sp.n[PC_STR] * sizeof(int64_t),
                            hipMemcpyHostToDevice, ctx->stream);
         }
-        if (sp.n[PC_GOR]) {
-            /* chunk skeleton (page-major), states filled by k_gor_sync.
-               Chunk size is tunable for sweeps (GS_GOR_CHUNK env). */
+        for (int gcls = 0; gcls < 2; gcls++) {
+            /* chunk skeletons (page-major) for the two Gorilla classes,
+               states filled by k_gor_sync / k_gor_sync_null.  Chunk size
+               is tunable for sweeps (GS_GOR_CHUNK env). */
+            const int pc = gcls == 0 ? PC_GOR : PC_GORN;
+            if (!sp.n[pc]) continue;
             static uint32_t chunk_rows = [] {
                 const char *e = getenv("GS_GOR_CHUNK");
                 long v = e ? atol(e) : 0;
                 return uint32_t(v >= 64 ? v : GOR_CHUNK);
             }();
-            std::vector<int32_t> cbase(sp.n[PC_GOR] + 1);
+            std::vector<int32_t> cbase(sp.n[pc] + 1);
             std::vector<DevGorChunk> hch;
             int32_t acc = 0;
-            for (int i = 0; i < sp.n[PC_GOR]; i++) {
-                const DevPage &pg = sp.host[PC_GOR][i];
+            for (int i = 0; i < sp.n[pc]; i++) {
+                const DevPage &pg = sp.host[pc][i];
                 cbase[i] = acc;
                 uint32_t nch =
                     pg.nrows ? (pg.nrows + chunk_rows - 1) / chunk_rows : 1;
                 for (uint32_t k = 0; k < nch; k++) {
                     DevGorChunk c{};
                     c.data_off = pg.data_off;
+                    c.bitset_off = pg.bitset_off;
                     c.row_off = pg.row_off;
                     c.grp = pg.grp;
                     c.row0 = k * chunk_rows;
@@ -2826,30 +3149,39 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                 }
                 acc += int32_t(nch);
             }
-            cbase[sp.n[PC_GOR]] = acc;
-            sp.n_gor_chunks = acc;
-            if (hipMalloc(&sp.d_gor_chunks, hch.size() * sizeof(DevGorChunk)) !=
+            cbase[sp.n[pc]] = acc;
+            DevGorChunk **dchp = gcls == 0 ? &sp.d_gor_chunks
+                                           : &sp.d_gorn_chunks;
+            int32_t **dcbp = gcls == 0 ? &sp.d_gor_chunk_base
+                                       : &sp.d_gorn_chunk_base;
+            (gcls == 0 ? sp.n_gor_chunks : sp.n_gorn_chunks) = acc;
+            if (hipMalloc(dchp, hch.size() * sizeof(DevGorChunk)) !=
                     hipSuccess ||
-                hipMalloc(&sp.d_gor_chunk_base,
-                          cbase.size() * sizeof(int32_t)) != hipSuccess) {
+                hipMalloc(dcbp, cbase.size() * sizeof(int32_t)) !=
+                    hipSuccess) {
                 fail(GS_ERR, "hipMalloc gor chunk table failed");
                 gs_groups_free(set); return nullptr;
             }
-            hipMemcpyAsync(sp.d_gor_chunks, hch.data(),
+            hipMemcpyAsync(*dchp, hch.data(),
                            hch.size() * sizeof(DevGorChunk),
                            hipMemcpyHostToDevice, ctx->stream);
-            hipMemcpyAsync(sp.d_gor_chunk_base, cbase.data(),
+            hipMemcpyAsync(*dcbp, cbase.data(),
                            cbase.size() * sizeof(int32_t),
                            hipMemcpyHostToDevice, ctx->stream);
             /* hch/cbase are stack-local: wait for the staged copies before
                they go out of scope (upload path, not scan time) */
             hipStreamSynchronize(ctx->stream);
             /* one-time sync-point pre-pass (upload/setup, not scan time) */
-            hipLaunchKernelGGL(k_gor_sync, dim3(grid_for(sp.n[PC_GOR], 256)),
-                               dim3(256), 0, ctx->stream, set->d_blob,
-                               sp.dev[PC_GOR], sp.n[PC_GOR],
-                               sp.d_gor_chunk_base, sp.d_gor_chunks,
-                               chunk_rows);
+            if (gcls == 0)
+                hipLaunchKernelGGL(k_gor_sync,
+                                   dim3(grid_for(sp.n[pc], 256)), dim3(256),
+                                   0, ctx->stream, set->d_blob, sp.dev[pc],
+                                   sp.n[pc], *dcbp, *dchp, chunk_rows);
+            else
+                hipLaunchKernelGGL(k_gor_sync_null,
+                                   dim3(grid_for(sp.n[pc], 256)), dim3(256),
+                                   0, ctx->stream, set->d_blob, sp.dev[pc],
+                                   sp.n[pc], *dcbp, *dchp, chunk_rows);
         }
     }
 
@@ -2924,6 +3256,8 @@ void gs_groups_free(GsGroupSet *set) {
         if (sp.d_str_scr) hipFree(sp.d_str_scr);
         if (sp.d_gor_chunks) hipFree(sp.d_gor_chunks);
         if (sp.d_gor_chunk_base) hipFree(sp.d_gor_chunk_base);
+        if (sp.d_gorn_chunks) hipFree(sp.d_gorn_chunks);
+        if (sp.d_gorn_chunk_base) hipFree(sp.d_gorn_chunk_base);
     }
     if (set->d_gor_active) hipFree(set->d_gor_active);
     if (set->d_gor_nactive) hipFree(set->d_gor_nactive);
@@ -3018,8 +3352,24 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
         int n = sp.n_gor_chunks;
         hipLaunchKernelGGL(k_gor_chunks, dim3(grid_for(n, GS_GOR_BLOCK)),
                            dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
-                           sp.d_gor_chunks, n, (double *)d_out, d_valid,
-                           ctx->d_err);
+                           sp.d_gor_chunks, n, (double *)d_out, ctx->d_err);
+        if (d_valid)
+            hipLaunchKernelGGL(k_valid_fill,
+                               dim3(sp.n[PC_GOR] > 2048 ? 2048 : sp.n[PC_GOR]),
+                               dim3(256), 0, ctx->stream, sp.dev[PC_GOR],
+                               sp.n[PC_GOR], d_valid);
+    }
+    if (sp.n[PC_GORN]) {
+        int n = sp.n_gorn_chunks;
+        hipLaunchKernelGGL(k_gor_chunks_null, dim3(grid_for(n, GS_GOR_BLOCK)),
+                           dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
+                           sp.d_gorn_chunks, n, (double *)d_out, ctx->d_err);
+        if (d_valid)
+            hipLaunchKernelGGL(
+                k_valid_expand,
+                dim3(sp.n[PC_GORN] > 2048 ? 2048 : sp.n[PC_GORN]), dim3(256),
+                0, ctx->stream, set->d_blob, sp.dev[PC_GORN], sp.n[PC_GORN],
+                d_valid);
     }
     if (sp.n[PC_S8B]) {
         int n = sp.n[PC_S8B];

@@ -39,23 +39,32 @@ struct DevGroup {
  * table per 4096 values = ~0.1% of typical compressed size. */
 #define GOR_CHUNK 2048
 
+/* DevGorChunk.flags bits */
+#define GORF_SAFE_STOP 1 /* pre-pass proved the whole chunk decodable (the
+                            next chunk's state was recorded): the filtered
+                            kernel may stop at the span end without losing
+                            the page's error surface (PC_GOR only) */
+#define GORF_SENT_SEEN 2 /* the sentinel was already consumed before row0:
+                            this chunk holds only trailing null rows
+                            (PC_GORN only) */
+
 struct DevGorChunk {
-    uint64_t data_off;  /* page data offset in blob */
-    uint64_t bitpos;    /* bits consumed from stream start; 0 for row0==0
-                           (decode reads the page header itself there) */
-    uint64_t val;       /* bits of row (row0-1)'s value (pre-pass) */
-    int64_t row_off;    /* output row offset of the page's group */
-    uint32_t grp;       /* page-group index */
-    uint32_t row0;      /* first row index in page this chunk produces */
-    uint32_t cnt;       /* rows this chunk produces */
-    uint32_t data_len;  /* page data_len */
-    uint8_t trailing;   /* XOR window state at bitpos (pre-pass) */
+    uint64_t data_off;   /* page data offset in blob */
+    uint64_t bitset_off; /* page validity bitset offset (PC_GORN) */
+    uint64_t bitpos;     /* bits consumed from stream start; 0 for row0==0
+                            (decode reads the page header itself there) */
+    uint64_t val;        /* PC_GOR: bits of row (row0-1)'s value.
+                            PC_GORN: bits of the value PENDING for the next
+                            set-bit row >= row0 (pipelined one ahead). */
+    int64_t row_off;     /* output row offset of the page's group */
+    uint32_t grp;        /* page-group index */
+    uint32_t row0;       /* first row index in page this chunk produces */
+    uint32_t cnt;        /* rows this chunk produces */
+    uint32_t data_len;   /* page data_len */
+    uint8_t trailing;    /* XOR window state at bitpos (pre-pass) */
     uint8_t meaningful;
-    uint8_t last;       /* final chunk: page-end semantics (sentinel) */
-    uint8_t safe_stop;  /* pre-pass proved the whole chunk decodable (the
-                           next chunk's state was recorded): the filtered
-                           kernel may stop at the span end without losing
-                           the page's error surface */
+    uint8_t last;        /* final chunk: page-end semantics (sentinel) */
+    uint8_t flags;       /* GORF_* */
 };
 
 /* launch-class partition of a column slot's pages */
@@ -69,7 +78,9 @@ enum PageClass {
     PC_BOOL = 6,    /* BitPack + all-valid: parallel bit extract */
     PC_STR = 7,     /* string blocks (snappy / uncompressed): sequential
                        thread-per-page decompress, see gs_decode_str */
-    PC_NCLASS = 8,
+    PC_GORN = 8,    /* Gorilla + nulls: chunk-parallel pending-value
+                       pipeline over the validity bitset */
+    PC_NCLASS = 9,
 };
 
 extern "C" {

@@ -210,3 +210,78 @@ def test_chunked_truncation_inside_span_errors(engine):
                     time_range=(t0, t0 + npts * 10**9),
                     d_out_ts=d_ots, d_out_val=d_oval)
     gset.free()
+
+
+def _upload_f64_nulls(engine, cases):
+    """cases: list of (values, valid_mask)."""
+    groups = []
+    for i, (vals, valid) in enumerate(cases):
+        n = len(vals)
+        ts = np.arange(n, dtype=np.int64) * 1000
+        groups.append((i, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (gs.page_of(np.asarray(vals), gs.CT_F64, valid),
+                            gs.CT_F64)]))
+    return engine.upload(groups)
+
+
+def test_chunked_null_pages_bit_exact(engine):
+    """Null-carrying Gorilla pages (PC_GORN): the chunk-parallel
+    pending-value pipeline must scatter values to set bits exactly like
+    the reference's bitset-driven builder (tsm/reader.rs:763-825), with
+    k_valid_expand producing the validity bytes."""
+    cases = []
+    for n in (100, GOR_CHUNK + 3, 5_000, 100_000):
+        vals = _walk(n)
+        valid = rng.random(n) > 0.10
+        if not valid.any():
+            valid[0] = True
+        cases.append((vals, valid))
+    # trailing-null page: values only in the first 30% (the sentinel is
+    # consumed long before the last rows -> GORF_SENT_SEEN tail chunks)
+    n = 60_000
+    vals = _walk(n)
+    valid = np.zeros(n, dtype=bool)
+    valid[:n * 3 // 10] = rng.random(n * 3 // 10) > 0.05
+    valid[0] = True
+    cases.append((vals, valid))
+    # leading-null page
+    valid2 = np.zeros(n, dtype=bool)
+    valid2[n // 2:] = True
+    cases.append((vals, valid2))
+    gset = _upload_f64_nulls(engine, cases)
+    out = torch.zeros(gset.rows, dtype=torch.float64, device="cuda")
+    dv = torch.zeros(gset.rows, dtype=torch.uint8, device="cuda")
+    engine.decode(gset, 1, out, dv)
+    offs = gset.row_offsets()
+    host, hv = out.cpu().numpy(), dv.cpu().numpy()
+    for i, (vals, valid) in enumerate(cases):
+        n = len(vals)
+        present = vals[valid]
+        data = gs.encode_f64(present) if present.size else b""
+        exp = orc.decode_f64(data, n, valid)
+        got = host[offs[i]:offs[i] + n]
+        assert got.view(np.uint64).tolist() == exp.view(np.uint64).tolist(), \
+            f"null chunked case {i}"
+        assert (hv[offs[i]:offs[i] + n] == valid.astype(np.uint8)).all(), \
+            f"valid bytes case {i}"
+    gset.free()
+
+
+def test_chunked_null_truncated_errors(engine):
+    """Truncated null-carrying multi-chunk page must raise at decode."""
+    n = 80_000
+    vals = _walk(n)
+    valid = rng.random(n) > 0.10
+    valid[0] = True
+    data = gs.encode_f64(vals[valid])
+    cut = data[:len(data) // 3]
+    nb = (n + 7) // 8
+    bitset = np.packbits(valid, bitorder="little").tobytes()
+    page = gs.build_page(cut, n, bitset)
+    ts = np.arange(n, dtype=np.int64) * 1000
+    gset = engine.upload([(0, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                               (page, gs.CT_F64)])])
+    out = torch.zeros(gset.rows, dtype=torch.float64, device="cuda")
+    with pytest.raises(RuntimeError):
+        engine.decode(gset, 1, out)
+    gset.free()
