@@ -695,6 +695,15 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
     }
 }
 
+/* bitset word fetch: LSB-first bytes -> little-endian u64 gives bit index
+ * == row index within the word.  One (possibly unaligned) 8-B load per 64
+ * rows instead of a divergent byte load per row. */
+__device__ __forceinline__ uint64_t dev_bits64(const uint8_t *bs, uint32_t w) {
+    uint64_t v;
+    __builtin_memcpy(&v, bs + (size_t(w) << 3), 8);
+    return v;
+}
+
 /* --- Gorilla chunked decode for NULL-carrying pages (PC_GORN) ---
  * The encoded stream holds only non-null values, scattered to set bits of
  * the validity bitset (tsm/reader.rs:763-825 / float.rs decode via
@@ -734,6 +743,8 @@ __global__ void k_gor_sync_null(const uint8_t *__restrict__ blob,
             uint32_t trailing = 0, meaningful = 64;
             uint32_t r = 0; /* ROW cursor */
             bool over = false;
+            uint64_t bw = 0;
+            uint32_t bw_idx = 0xffffffffu;
             uint64_t nextw = dev_be64(p);
             uint64_t nextw2 = dev_be64(p + 8);
             p += 16;
@@ -770,7 +781,11 @@ __global__ void k_gor_sync_null(const uint8_t *__restrict__ blob,
                 if (sent_seen) { r++; continue; } /* walk rows to boundary */
                 if (nb < 64) topup();
                 if (over) { ok = false; break; }
-                if (dev_bit(bs, r)) {
+                {
+                    const uint32_t wi = r >> 6;
+                    if (wi != bw_idx) { bw = dev_bits64(bs, wi); bw_idx = wi; }
+                }
+                if ((bw >> (r & 63)) & 1) {
                     /* this set row consumes the pending value: parse the
                        next one (uniform path, see k_gor_chunks) */
                     uint32_t top13 = uint32_t(hi >> 51);
@@ -843,6 +858,8 @@ __global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
         const uint8_t *bs = blob + ch.bitset_off;
         double *o = out + ch.row_off;
         uint32_t r = ch.row0;
+        uint64_t bw = 0;
+        uint32_t bw_idx = 0xffffffffu; /* lazy per-64-row bit word */
         uint32_t end = ch.row0 + ch.cnt;
         bool sent_seen = (ch.flags & GORF_SENT_SEEN) != 0;
         GorChunkState st;
@@ -942,7 +959,9 @@ __global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
             }
             if (!done) {
                 if (r < end) {
-                    int bit = dev_bit(bs, r);
+                    const uint32_t wi = r >> 6;
+                    if (wi != bw_idx) { bw = dev_bits64(bs, wi); bw_idx = wi; }
+                    const int bit = int((bw >> (r & 63)) & 1);
                     uint64_t sbits = 0;
                     bool hole = false; /* set row after the sentinel */
                     if (bit) {
@@ -1001,7 +1020,16 @@ __global__ void k_valid_expand(const uint8_t *__restrict__ blob,
         DevPage pg = pages[p];
         const uint8_t *bs = blob + pg.bitset_off;
         uint8_t *vd = valid + pg.row_off;
-        for (uint32_t r = threadIdx.x; r < pg.nrows; r += blockDim.x)
+        const uint32_t nfull = pg.nrows >> 3; /* whole bytes -> u64 writes */
+        for (uint32_t b = threadIdx.x; b < nfull; b += blockDim.x) {
+            uint64_t ex = 0;
+            uint8_t byte = bs[b];
+            for (int k = 0; k < 8; k++)
+                ex |= uint64_t((byte >> k) & 1) << (8 * k);
+            __builtin_memcpy(vd + (size_t(b) << 3), &ex, 8);
+        }
+        for (uint32_t r = nfull * 8 + threadIdx.x; r < pg.nrows;
+             r += blockDim.x)
             vd[r] = uint8_t(dev_bit(bs, r));
     }
 }

@@ -275,8 +275,7 @@ def test_chunked_null_truncated_errors(engine):
     valid[0] = True
     data = gs.encode_f64(vals[valid])
     cut = data[:len(data) // 3]
-    nb = (n + 7) // 8
-    bitset = np.packbits(valid, bitorder="little").tobytes()
+    bitset = np.packbits(valid, bitorder="little")
     page = gs.build_page(cut, n, bitset)
     ts = np.arange(n, dtype=np.int64) * 1000
     gset = engine.upload([(0, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
