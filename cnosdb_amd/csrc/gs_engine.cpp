@@ -3316,6 +3316,21 @@ void gs_groups_free(GsGroupSet *set) {
     delete set;
 }
 
+/* debug: copy a slot's GORN chunk table to host (repro tooling only) */
+int64_t gs_debug_gorn_table(GsCtx *ctx, GsGroupSet *set, uint32_t col,
+                            DevGorChunk *out, int64_t cap) {
+    if (!ctx || !set || col >= set->ncols) return -1;
+    SlotPages &sp = set->slots[col];
+    int64_t n = sp.n_gorn_chunks;
+    if (!out || cap < n) return n;
+    hipSetDevice(ctx->device);
+    hipStreamSynchronize(ctx->stream);
+    if (hipMemcpy(out, sp.d_gorn_chunks, size_t(n) * sizeof(DevGorChunk),
+                  hipMemcpyDeviceToHost) != hipSuccess)
+        return -1;
+    return n;
+}
+
 int64_t gs_set_rows(const GsGroupSet *set) { return set ? set->total_rows : -1; }
 int64_t gs_set_series(const GsGroupSet *set) { return set ? set->nsgroups : -1; }
 

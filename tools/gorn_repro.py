@@ -98,5 +98,70 @@ def main():
     print("dumped to gpurun_out/fail_{walk,valid}.npy")
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and not (len(sys.argv) > 1 and sys.argv[1] == "dump"):
     main()
+
+
+def dump_states(nentries=98304, page_rows=4000):
+    """Upload a failing-size set and diff chunk states of identical pool
+    pages at low vs high page index."""
+    import ctypes
+
+    class DevGorChunk(ctypes.Structure):
+        _fields_ = [("data_off", ctypes.c_uint64),
+                    ("bitset_off", ctypes.c_uint64),
+                    ("bitpos", ctypes.c_uint64),
+                    ("val", ctypes.c_uint64),
+                    ("row_off", ctypes.c_int64),
+                    ("grp", ctypes.c_uint32),
+                    ("row0", ctypes.c_uint32),
+                    ("cnt", ctypes.c_uint32),
+                    ("data_len", ctypes.c_uint32),
+                    ("trailing", ctypes.c_uint8),
+                    ("meaningful", ctypes.c_uint8),
+                    ("last", ctypes.c_uint8),
+                    ("flags", ctypes.c_uint8)]
+
+    eng = gs.Engine(0)
+    rng = np.random.default_rng(231)
+    pool = make_pool(rng, page_rows, 256)
+    entries = [pool[i % 256] for i in range(nentries)]
+    groups = []
+    for i, (walk, valid) in enumerate(entries):
+        ts = T0 + (np.arange(page_rows, dtype=np.int64)) * NS
+        groups.append((i, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (gs.page_of(walk, gs.CT_F64, valid), gs.CT_F64)]))
+    gset = eng.upload(groups)
+    lib = eng.lib
+    lib.gs_debug_gorn_table.restype = ctypes.c_int64
+    lib.gs_debug_gorn_table.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                        ctypes.c_uint32, ctypes.c_void_p,
+                                        ctypes.c_int64]
+    n = lib.gs_debug_gorn_table(eng._ctx, gset._h, 1, None, 0)
+    print("gorn chunks:", n)
+    arr = (DevGorChunk * n)()
+    r = lib.gs_debug_gorn_table(eng._ctx, gset._h, 1, arr, n)
+    assert r == n
+    per_page = n // nentries
+    print("chunks/page:", per_page)
+    bad = 0
+    for p in range(nentries):
+        ref_p = p % 256  # pool twin at low index
+        for k in range(per_page):
+            a, b = arr[ref_p * per_page + k], arr[p * per_page + k]
+            if (a.bitpos, a.val, a.trailing, a.meaningful, a.flags) != \
+               (b.bitpos, b.val, b.trailing, b.meaningful, b.flags):
+                if bad < 8:
+                    print(f"page {p} chunk {k}: bitpos {b.bitpos} vs "
+                          f"{a.bitpos}, val {b.val:#x} vs {a.val:#x}, "
+                          f"fl {b.flags} vs {a.flags}, dlen {b.data_len}")
+                bad += 1
+    print("mismatched chunk states:", bad)
+    # also scan for poisoned states
+    pois = sum(1 for c in arr if c.row0 and c.bitpos >= c.data_len * 8)
+    print("poisoned chunks:", pois)
+    gset.free()
+
+
+if len(sys.argv) > 1 and sys.argv[1] == "dump":
+    dump_states(int(sys.argv[2]) if len(sys.argv) > 2 else 98304)
