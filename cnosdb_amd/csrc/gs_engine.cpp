@@ -3331,6 +3331,17 @@ int64_t gs_debug_gorn_table(GsCtx *ctx, GsGroupSet *set, uint32_t col,
     return n;
 }
 
+int64_t gs_debug_read_blob(GsCtx *ctx, GsGroupSet *set, uint64_t off,
+                           uint64_t len, uint8_t *out) {
+    if (!ctx || !set || off + len > set->blob_len) return -1;
+    hipSetDevice(ctx->device);
+    hipStreamSynchronize(ctx->stream);
+    if (hipMemcpy(out, set->d_blob + off, len, hipMemcpyDeviceToHost) !=
+        hipSuccess)
+        return -1;
+    return int64_t(len);
+}
+
 int64_t gs_set_rows(const GsGroupSet *set) { return set ? set->total_rows : -1; }
 int64_t gs_set_series(const GsGroupSet *set) { return set ? set->nsgroups : -1; }
 

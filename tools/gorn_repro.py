@@ -157,6 +157,34 @@ def dump_states(nentries=98304, page_rows=4000):
                           f"fl {b.flags} vs {a.flags}, dlen {b.data_len}")
                 bad += 1
     print("mismatched chunk states:", bad)
+    # offsets of the first mismatching page vs its twin + blob byte diff
+    lib.gs_debug_read_blob.restype = ctypes.c_int64
+    lib.gs_debug_read_blob.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.c_uint64, ctypes.c_uint64,
+                                       ctypes.c_void_p]
+
+    def blob_bytes(off, ln):
+        buf = np.zeros(ln, dtype=np.uint8)
+        r2 = lib.gs_debug_read_blob(eng._ctx, gset._h, off, ln,
+                                    buf.ctypes.data_as(ctypes.c_void_p))
+        assert r2 == ln, (off, ln, r2)
+        return buf
+
+    for p in (65535, 65536, 65537):
+        ref_p = p % 256
+        a, b = arr[ref_p * per_page], arr[p * per_page]
+        print(f"page {p}: data_off {b.data_off} bitset_off {b.bitset_off} "
+              f"dlen {b.data_len}; twin data_off {a.data_off} dlen {a.data_len}")
+        da = blob_bytes(a.data_off, a.data_len)
+        db = blob_bytes(b.data_off, b.data_len)
+        same_data = (da == db).all() if a.data_len == b.data_len else False
+        ba = blob_bytes(a.bitset_off, 500)
+        bb = blob_bytes(b.bitset_off, 500)
+        print(f"  data identical: {same_data}; bitset identical: "
+              f"{(ba == bb).all()}")
+        if not same_data and a.data_len == b.data_len:
+            d = np.flatnonzero(da != db)
+            print(f"  first byte diffs at {d[:6].tolist()} of {a.data_len}")
     # also scan for poisoned states
     pois = sum(1 for c in arr if c.row0 and c.bitpos >= c.data_len * 8)
     print("poisoned chunks:", pois)
