@@ -3342,6 +3342,23 @@ int64_t gs_debug_read_blob(GsCtx *ctx, GsGroupSet *set, uint64_t off,
     return int64_t(len);
 }
 
+int64_t gs_debug_rerun_gorn_sync(GsCtx *ctx, GsGroupSet *set,
+                                 uint32_t col) {
+    if (!ctx || !set || col >= set->ncols) return -1;
+    SlotPages &sp = set->slots[col];
+    if (!sp.n[PC_GORN]) return 0;
+    hipSetDevice(ctx->device);
+    const char *e = getenv("GS_GOR_CHUNK");
+    long v = e ? atol(e) : 0;
+    uint32_t chunk_rows = uint32_t(v >= 64 ? v : GOR_CHUNK);
+    hipLaunchKernelGGL(k_gor_sync_null, dim3(grid_for(sp.n[PC_GORN], 256)),
+                       dim3(256), 0, ctx->stream, set->d_blob,
+                       sp.dev[PC_GORN], sp.n[PC_GORN], sp.d_gorn_chunk_base,
+                       sp.d_gorn_chunks, chunk_rows);
+    hipStreamSynchronize(ctx->stream);
+    return 1;
+}
+
 int64_t gs_set_rows(const GsGroupSet *set) { return set ? set->total_rows : -1; }
 int64_t gs_set_series(const GsGroupSet *set) { return set ? set->nsgroups : -1; }
 

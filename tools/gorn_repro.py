@@ -188,6 +188,22 @@ def dump_states(nentries=98304, page_rows=4000):
     # also scan for poisoned states
     pois = sum(1 for c in arr if c.row0 and c.bitpos >= c.data_len * 8)
     print("poisoned chunks:", pois)
+    # rerun the sync kernel post-upload: if states become twin-consistent,
+    # something later in upload stomped the table
+    lib.gs_debug_rerun_gorn_sync.restype = ctypes.c_int64
+    lib.gs_debug_rerun_gorn_sync.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                             ctypes.c_uint32]
+    lib.gs_debug_rerun_gorn_sync(eng._ctx, gset._h, 1)
+    r = lib.gs_debug_gorn_table(eng._ctx, gset._h, 1, arr, n)
+    bad2 = 0
+    for p in range(nentries):
+        ref_p = p % 256
+        for k in range(per_page):
+            a, b = arr[ref_p * per_page + k], arr[p * per_page + k]
+            if (a.bitpos, a.val, a.trailing, a.meaningful, a.flags) != \
+               (b.bitpos, b.val, b.trailing, b.meaningful, b.flags):
+                bad2 += 1
+    print("mismatched after sync rerun:", bad2)
     gset.free()
 
 
