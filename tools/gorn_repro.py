@@ -170,7 +170,9 @@ def dump_states(nentries=98304, page_rows=4000):
         assert r2 == ln, (off, ln, r2)
         return buf
 
-    for p in (65535, 65536, 65537):
+    for p in (65535, 65536, 65537, 90000, 98000):
+        if p >= nentries:
+            continue
         ref_p = p % 256
         a, b = arr[ref_p * per_page], arr[p * per_page]
         print(f"page {p}: data_off {b.data_off} bitset_off {b.bitset_off} "
@@ -208,4 +210,5 @@ def dump_states(nentries=98304, page_rows=4000):
 
 
 if len(sys.argv) > 1 and sys.argv[1] == "dump":
-    dump_states(int(sys.argv[2]) if len(sys.argv) > 2 else 98304)
+    dump_states(int(sys.argv[2]) if len(sys.argv) > 2 else 98304,
+                int(sys.argv[3]) if len(sys.argv) > 3 else 4000)
