@@ -157,6 +157,21 @@ def dump_states(nentries=98304, page_rows=4000):
                           f"fl {b.flags} vs {a.flags}, dlen {b.data_len}")
                 bad += 1
     print("mismatched chunk states:", bad)
+    skel_bad = 0
+    for p in range(nentries):
+        for k in range(per_page):
+            b = arr[p * per_page + k]
+            exp_row0 = k * 2048
+            exp_cnt = min(2048, page_rows - exp_row0)
+            a = arr[(p % 256) * per_page + k]
+            if (b.row0 != exp_row0 or b.cnt != exp_cnt or
+                    b.grp != p or b.data_len != a.data_len or
+                    b.last != (1 if k == per_page - 1 else 0)):
+                if skel_bad < 5:
+                    print(f"SKEL page {p} chunk {k}: row0={b.row0} cnt={b.cnt} "
+                          f"grp={b.grp} dlen={b.data_len} last={b.last}")
+                skel_bad += 1
+    print("skeleton-corrupt entries:", skel_bad)
     # offsets of the first mismatching page vs its twin + blob byte diff
     lib.gs_debug_read_blob.restype = ctypes.c_int64
     lib.gs_debug_read_blob.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
