@@ -3359,6 +3359,31 @@ int64_t gs_debug_rerun_gorn_sync(GsCtx *ctx, GsGroupSet *set,
     return 1;
 }
 
+__global__ void k_debug_crc(const uint8_t *__restrict__ p, uint64_t len,
+                            uint64_t *__restrict__ out) {
+    /* simple order-independent mix so any byte difference shows */
+    uint64_t acc = 0;
+    for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < len;
+         i += uint64_t(gridDim.x) * blockDim.x)
+        acc += (uint64_t(p[i]) + 1) * (i + 0x9e3779b97f4a7c15ULL);
+    atomicAdd((unsigned long long *)out, (unsigned long long)acc);
+}
+
+int64_t gs_debug_kernel_crc(GsCtx *ctx, GsGroupSet *set, uint64_t off,
+                            uint64_t len, uint64_t *out_sum) {
+    if (!ctx || !set || off + len > set->blob_len) return -1;
+    hipSetDevice(ctx->device);
+    uint64_t *d;
+    if (hipMalloc(&d, 8) != hipSuccess) return -1;
+    hipMemset(d, 0, 8);
+    hipLaunchKernelGGL(k_debug_crc, dim3(64), dim3(256), 0, ctx->stream,
+                       set->d_blob + off, len, d);
+    hipStreamSynchronize(ctx->stream);
+    hipMemcpy(out_sum, d, 8, hipMemcpyDeviceToHost);
+    hipFree(d);
+    return 0;
+}
+
 int64_t gs_set_rows(const GsGroupSet *set) { return set ? set->total_rows : -1; }
 int64_t gs_set_series(const GsGroupSet *set) { return set ? set->nsgroups : -1; }
 

@@ -172,6 +172,32 @@ def dump_states(nentries=98304, page_rows=4000):
                           f"grp={b.grp} dlen={b.data_len} last={b.last}")
                 skel_bad += 1
     print("skeleton-corrupt entries:", skel_bad)
+    # kernel-view vs copy-engine-view of the same bytes
+    lib.gs_debug_kernel_crc.restype = ctypes.c_int64
+    lib.gs_debug_kernel_crc.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                        ctypes.c_uint64, ctypes.c_uint64,
+                                        ctypes.c_void_p]
+
+    def kcrc(off, ln):
+        o = ctypes.c_uint64(0)
+        assert lib.gs_debug_kernel_crc(eng._ctx, gset._h, off, ln,
+                                       ctypes.byref(o)) == 0
+        return o.value
+
+    def hcrc(off, ln):
+        b = blob_bytes(off, ln).astype(np.uint64)
+        idx = np.arange(ln, dtype=np.uint64)
+        return int(((b + 1) * (idx + np.uint64(0x9E3779B97F4A7C15))).sum(
+            dtype=np.uint64))
+
+    for p in (0, 60000, 65536, 70000, nentries - 1):
+        if p >= nentries:
+            continue
+        c = arr[p * per_page]
+        kv = kcrc(c.data_off, c.data_len)
+        hv = hcrc(c.data_off, c.data_len)
+        print(f"page {p} data_off={c.data_off}: kernel-crc "
+              f"{'==' if kv == hv else '!='} copy-crc")
     # offsets of the first mismatching page vs its twin + blob byte diff
     lib.gs_debug_read_blob.restype = ctypes.c_int64
     lib.gs_debug_read_blob.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
