@@ -185,7 +185,14 @@ def dump_states(nentries=98304, page_rows=4000):
         return o.value
 
     def hcrc(off, ln):
-        b = blob_bytes(off, ln).astype(np.uint64)
+        buf = np.zeros(ln, dtype=np.uint8)
+        lib.gs_debug_read_blob.restype = ctypes.c_int64
+        lib.gs_debug_read_blob.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                           ctypes.c_uint64, ctypes.c_uint64,
+                                           ctypes.c_void_p]
+        assert lib.gs_debug_read_blob(eng._ctx, gset._h, off, ln,
+                                      buf.ctypes.data_as(ctypes.c_void_p)) == ln
+        b = buf.astype(np.uint64)
         idx = np.arange(ln, dtype=np.uint64)
         return int(((b + 1) * (idx + np.uint64(0x9E3779B97F4A7C15))).sum(
             dtype=np.uint64))
