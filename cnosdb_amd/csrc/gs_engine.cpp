@@ -781,11 +781,7 @@ __global__ void k_gor_sync_null(const uint8_t *__restrict__ blob,
                 if (sent_seen) { r++; continue; } /* walk rows to boundary */
                 if (nb < 64) topup();
                 if (over) { ok = false; break; }
-                {
-                    const uint32_t wi = r >> 6;
-                    if (wi != bw_idx) { bw = dev_bits64(bs, wi); bw_idx = wi; }
-                }
-                if ((bw >> (r & 63)) & 1) {
+                if (dev_bit(bs, r)) {
                     /* this set row consumes the pending value: parse the
                        next one (uniform path, see k_gor_chunks) */
                     uint32_t top13 = uint32_t(hi >> 51);
