@@ -695,15 +695,6 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
     }
 }
 
-/* bitset word fetch: LSB-first bytes -> little-endian u64 gives bit index
- * == row index within the word.  One (possibly unaligned) 8-B load per 64
- * rows instead of a divergent byte load per row. */
-__device__ __forceinline__ uint64_t dev_bits64(const uint8_t *bs, uint32_t w) {
-    uint64_t v;
-    __builtin_memcpy(&v, bs + (size_t(w) << 3), 8);
-    return v;
-}
-
 /* --- Gorilla chunked decode for NULL-carrying pages (PC_GORN) ---
  * The encoded stream holds only non-null values, scattered to set bits of
  * the validity bitset (tsm/reader.rs:763-825 / float.rs decode via
@@ -743,8 +734,6 @@ __global__ void k_gor_sync_null(const uint8_t *__restrict__ blob,
             uint32_t trailing = 0, meaningful = 64;
             uint32_t r = 0; /* ROW cursor */
             bool over = false;
-            uint64_t bw = 0;
-            uint32_t bw_idx = 0xffffffffu;
             uint64_t nextw = dev_be64(p);
             uint64_t nextw2 = dev_be64(p + 8);
             p += 16;
@@ -854,8 +843,11 @@ __global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
         const uint8_t *bs = blob + ch.bitset_off;
         double *o = out + ch.row_off;
         uint32_t r = ch.row0;
-        uint64_t bw = 0;
-        uint32_t bw_idx = 0xffffffffu; /* lazy per-64-row bit word */
+        uint32_t bb = 0;
+        uint32_t bb_idx = 0xffffffffu; /* lazy per-8-row bitset byte (byte
+                                          loads are always aligned; the
+                                          8-B word variant misread flakily
+                                          at high blob offsets) */
         uint32_t end = ch.row0 + ch.cnt;
         bool sent_seen = (ch.flags & GORF_SENT_SEEN) != 0;
         GorChunkState st;
@@ -955,9 +947,9 @@ __global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
             }
             if (!done) {
                 if (r < end) {
-                    const uint32_t wi = r >> 6;
-                    if (wi != bw_idx) { bw = dev_bits64(bs, wi); bw_idx = wi; }
-                    const int bit = int((bw >> (r & 63)) & 1);
+                    const uint32_t bi = r >> 3;
+                    if (bi != bb_idx) { bb = bs[bi]; bb_idx = bi; }
+                    const int bit = int((bb >> (r & 7)) & 1);
                     uint64_t sbits = 0;
                     bool hole = false; /* set row after the sentinel */
                     if (bit) {
