@@ -96,6 +96,11 @@ __device__ __forceinline__ bool dev_varint(const uint8_t *p, uint32_t len,
 /* error codes accumulated into d_err */
 #define DERR_FORMAT 1u
 #define DERR_SHORT 2u
+#define DERR_NONMONO 4u /* a CRC-valid but non-monotonic time
+                           page would silently mis-select rows via the
+                           binary-search span/tombstone kernels; fail
+                           loudly instead (the reference's row-wise
+                           filtering cannot mis-select) */
 
 /* --------------------------------------------- pull-style value iterators */
 
@@ -164,9 +169,17 @@ __global__ void k_seq_i64(const uint8_t *__restrict__ blob,
         if (enc == GS_ENC_NULL) { /* raw BE per valid slot */
             const uint8_t *q = data + 1;
             uint32_t avail = (pg.data_len - 1) / 8, used = 0;
+            int64_t prev_ts = INT64_MIN;
             for (uint32_t r = 0; r < n; r++) {
                 int v = pg.all_valid ? 1 : dev_bit(bs, r);
-                if (v && used < avail) { o[r] = int64_t(dev_be64(q + 8ull * used)); used++; }
+                if (v && used < avail) {
+                    int64_t x = int64_t(dev_be64(q + 8ull * used)); used++;
+                    if (pg.ctype == GS_CT_TIME) {
+                        if (x < prev_ts) atomicOr(err, DERR_NONMONO);
+                        prev_ts = x;
+                    }
+                    o[r] = x;
+                }
                 else { o[r] = 0; v = 0; }
                 if (vd) vd[r] = uint8_t(v);
             }
@@ -195,8 +208,12 @@ __global__ void k_seq_i64(const uint8_t *__restrict__ blob,
                 }
                 uint64_t raw = dev_be64(q + 8ull * used);
                 used++;
-                prev = is_ts ? int64_t(uint64_t(prev) + raw)
-                             : int64_t(uint64_t(prev) + uint64_t(dev_zzdec(raw)));
+                int64_t nxt = is_ts ? int64_t(uint64_t(prev) + raw)
+                                    : int64_t(uint64_t(prev) +
+                                              uint64_t(dev_zzdec(raw)));
+                if (is_ts && nxt < prev && used > 1)
+                    atomicOr(err, DERR_NONMONO);
+                prev = nxt;
                 o[r] = prev;
                 if (vd) vd[r] = 1;
             }
@@ -217,6 +234,7 @@ __global__ void k_seq_i64(const uint8_t *__restrict__ blob,
             int64_t cur, delta;
             if (is_ts) { cur = int64_t(first_raw); delta = int64_t(dv * scaler); }
             else { cur = dev_zzdec(first_raw); delta = dev_zzdec(dv); }
+            if (is_ts && delta < 0 && n > 1) atomicOr(err, DERR_NONMONO);
             bool first = true;
             for (uint32_t r = 0; r < n; r++) {
                 int v = pg.all_valid ? 1 : dev_bit(bs, r);
@@ -247,8 +265,11 @@ __global__ void k_seq_i64(const uint8_t *__restrict__ blob,
                 if (first) { o[r] = cur; first = false; if (vd) vd[r] = 1; continue; }
                 uint64_t u;
                 if (!it.next(&u)) { o[r] = 0; if (vd) vd[r] = 1; continue; } /* iterator exhaustion: builder skips */
-                cur = is_ts ? int64_t(uint64_t(cur) + u * scaler)
-                            : int64_t(uint64_t(cur) + uint64_t(dev_zzdec(u)));
+                int64_t nxt2 = is_ts ? int64_t(uint64_t(cur) + u * scaler)
+                                     : int64_t(uint64_t(cur) +
+                                               uint64_t(dev_zzdec(u)));
+                if (is_ts && nxt2 < cur) atomicOr(err, DERR_NONMONO);
+                cur = nxt2;
                 o[r] = cur;
                 if (vd) vd[r] = 1;
             }
@@ -1139,6 +1160,8 @@ __global__ void k_rle_par(const uint8_t *__restrict__ blob,
         int64_t first, delta;
         if (is_ts) { first = int64_t(first_raw); delta = int64_t(dv * scaler); }
         else { first = dev_zzdec(first_raw); delta = dev_zzdec(dv); }
+        if (is_ts && delta < 0 && pg.nrows > 1 && threadIdx.x == 0)
+            atomicOr(err, DERR_NONMONO);
         int64_t *o = out + pg.row_off;
         uint8_t *vd = valid ? valid + pg.row_off : nullptr;
         for (uint32_t r = threadIdx.x; r < pg.nrows; r += blockDim.x) {
@@ -1249,9 +1272,15 @@ __global__ void k_s8b_par(const uint8_t *__restrict__ blob,
                     }
                 } else {
                     uint64_t v = w;
+                    int64_t pr = int64_t(acc);
                     for (uint32_t k = 0; k < cnt; k++) {
                         uint64_t d = v & mask;
                         acc += is_ts ? d * scaler : uint64_t(dev_zzdec(d));
+                        /* ts: i64 wrap of the delta prefix = corrupt page
+                           (times must not decrease) */
+                        if (is_ts && int64_t(acc) < pr)
+                            atomicOr(err, DERR_NONMONO);
+                        pr = int64_t(acc);
                         if (voff + k < n) o[voff + k] = int64_t(acc);
                         v >>= bits;
                     }
@@ -1282,7 +1311,8 @@ __global__ void k_s8b_par(const uint8_t *__restrict__ blob,
 __global__ void k_raw_par(const uint8_t *__restrict__ blob,
                           const DevPage *__restrict__ pages, int npages,
                           int64_t *__restrict__ out,
-                          uint8_t *__restrict__ valid) {
+                          uint8_t *__restrict__ valid,
+                          unsigned *__restrict__ err) {
     for (int p = blockIdx.x; p < npages; p += gridDim.x) {
         DevPage pg = pages[p];
         const uint8_t *q = blob + pg.data_off + 1;
@@ -1290,8 +1320,15 @@ __global__ void k_raw_par(const uint8_t *__restrict__ blob,
         int64_t *o = out + pg.row_off;
         uint8_t *vd = valid ? valid + pg.row_off : nullptr;
         uint32_t n = pg.nrows;
+        const bool ck_ts = pg.ctype == GS_CT_TIME;
         for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
-            o[r] = (r < avail) ? int64_t(dev_be64(q + 8ull * r)) : 0;
+            int64_t x = (r < avail) ? int64_t(dev_be64(q + 8ull * r)) : 0;
+            /* absolute (Null-encoded) time pages may be unsorted: the
+               previous input word is an L1 hit */
+            if (ck_ts && r > 0 && r < avail &&
+                x < int64_t(dev_be64(q + 8ull * (r - 1))))
+                atomicOr(err, DERR_NONMONO);
+            o[r] = x;
             if (vd) vd[r] = 1;
         }
     }
@@ -1892,8 +1929,15 @@ __global__ void k_str_decode(const uint8_t *__restrict__ blob,
         for (uint32_t r = 0; r < pg.nrows; r++) {
             int v = pg.all_valid ? 1 : dev_bit(bs, r);
             if (!v) continue;
-            if (i >= uint64_t(pn)) break; /* payload exhausted: remaining
-                                             valid rows decode null */
+            if (i >= uint64_t(pn)) { /* payload exhausted with valid rows
+                                        remaining: the reference emits a
+                                        SHORTER array here (rows dropped,
+                                        str_snappy_decode_to_array); fail
+                                        loudly instead of silently
+                                        nulling */
+                atomicOr(err, DERR_SHORT);
+                break;
+            }
             uint64_t slen = 0;
             if (be) {
                 if (i + 8 > uint64_t(pn)) { atomicOr(err, DERR_FORMAT); break; }
@@ -2203,7 +2247,7 @@ __global__ void k_spans_rle(const uint8_t *__restrict__ blob,
         int64_t delta = int64_t(dv * scaler);
         int64_t n = pg.nrows;
         int64_t st, en; /* selected rows = [st, en) */
-        if (delta < 0) { atomicOr(err, DERR_FORMAT); st = en = 0; }
+        if (delta < 0) { atomicOr(err, DERR_NONMONO); st = en = 0; }
         else if (delta == 0) {
             bool in = first >= lo && first <= hi;
             st = 0; en = in ? n : 0;
@@ -3399,6 +3443,9 @@ static GsStatus check_dev_err(GsCtx *ctx) {
     HIP_TRY(hipMemcpy(&e, ctx->d_err, sizeof(unsigned), hipMemcpyDeviceToHost));
     if (e) {
         hipMemset(ctx->d_err, 0, sizeof(unsigned));
+        if (e & DERR_NONMONO)
+            return fail(GS_ERR_FORMAT,
+                        "decode: non-monotonic time page (corrupt input)");
         if (e & DERR_SHORT) return fail(GS_ERR_FORMAT, "decode: stream shorter than validity demands");
         return fail(GS_ERR_FORMAT, "decode: malformed page data");
     }
@@ -3471,7 +3518,7 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
         int n = sp.n[PC_RAW];
         hipLaunchKernelGGL(k_raw_par, dim3(n > 65535 ? 65535 : n), dim3(256),
                            0, ctx->stream, set->d_blob, sp.dev[PC_RAW], n,
-                           (int64_t *)d_out, d_valid);
+                           (int64_t *)d_out, d_valid, ctx->d_err);
     }
     if (sp.n[PC_RLE_TS])
         hipLaunchKernelGGL(k_rle_par,

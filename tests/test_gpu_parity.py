@@ -728,3 +728,39 @@ def test_upload_packed_matches_upload(engine):
     assert (c1 == c2).all()
     assert m1.tolist() == m2.tolist()
     assert s1.tolist() == s2.tolist()
+
+
+def test_nonmonotonic_ts_page_errors(engine):
+    """a CRC-valid but unsorted time page must fail decode
+    loudly (the span/tombstone binary searches assume ascending ts; the
+    reference's row-wise filtering cannot mis-select)."""
+    ts = np.array([1000, 500, 2000, 1500], dtype=np.int64)  # unsorted
+    data = bytes([1]) + ts.astype(">i8").tobytes()  # Null encoding, raw BE
+    tpage = gs.build_page(data, 4)
+    vals = np.array([1.0, 2.0, 3.0, 4.0])
+    gset = engine.upload([(0, [(tpage, gs.CT_TIME),
+                               (gs.page_of(vals, gs.CT_F64), gs.CT_F64)])])
+    out = torch.zeros(gset.rows, dtype=torch.int64, device="cuda")
+    with pytest.raises(RuntimeError, match="non-monotonic"):
+        engine.decode(gset, 0, out)
+    gset.free()
+
+
+def test_truncated_string_payload_errors(engine):
+    """a string page whose payload ends while valid rows
+    remain must error (the reference emits a shorter array there —
+    silently nulling would be an undetectable divergence)."""
+    import struct
+    # Null-encoded string block ([1][u64 BE len][bytes]..., string.rs:169-183)
+    # holding only 3 strings while the bitset marks 4 valid rows
+    payload = b"".join(struct.pack(">Q", 6) + b"abcdef" for _ in range(3))
+    page = gs.build_page(bytes([1]) + payload, 4)
+    ts = np.arange(4, dtype=np.int64) * 1000
+    gset = engine.upload([(0, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                               (page, gs.CT_STR)])])
+    rows = gset.rows
+    d_off = torch.zeros(rows + 1, dtype=torch.int64, device="cuda")
+    d_bytes = torch.zeros(rows * 64, dtype=torch.uint8, device="cuda")
+    with pytest.raises(RuntimeError):
+        engine.decode_str(gset, 1, d_off, d_bytes)
+    gset.free()
