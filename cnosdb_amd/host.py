@@ -620,3 +620,52 @@ class Engine:
         if st != 0:
             raise RuntimeError(f"gs_scan failed ({st}): {self._pl.err()}")
         return res
+
+
+def prune_column_groups(stats, time_range=None, value_pred=None):
+    """Host-side mirror of `filter_column_groups` (tskv/src/reader/
+    chunk.rs:12-49): the reference evaluates a DataFusion PruningPredicate
+    over per-column-group min/max statistics (PageMeta ValueStatistics,
+    tsm/page.rs:599-639) and drops groups no row of which can match.  The
+    shim runs this BEFORE gs_groups_upload, so pruned pages are never
+    read, uploaded, or decoded (a pruned page's corruption is therefore
+    never observed — same as the reference never reading it).
+
+    stats: iterable of (min_ts, max_ts, min_val, max_val); min_val/max_val
+    may be None when the field column has no stats (never prunable then).
+    time_range: closed interval (lo, hi) (TimeRange semantics,
+    common/models/src/predicate/domain.rs:36-39).
+    value_pred: ("gt"|"ge"|"lt"|"le"|"eq"|"ne"|"between", a[, b]) with
+    PruningPredicate keep-if-maybe semantics (nulls fail predicates but
+    other rows may pass, so stats ranges decide only certain misses).
+
+    Returns a list of bools (True = keep), like the reference's indices.
+    """
+    keep = []
+    for st in stats:
+        min_ts, max_ts, min_v, max_v = st
+        k = True
+        if time_range is not None:
+            lo, hi = time_range
+            if max_ts < lo or min_ts > hi:
+                k = False
+        if k and value_pred is not None and min_v is not None \
+                and max_v is not None:
+            op, a = value_pred[0], value_pred[1]
+            if op == "gt":
+                k = max_v > a
+            elif op == "ge":
+                k = max_v >= a
+            elif op == "lt":
+                k = min_v < a
+            elif op == "le":
+                k = min_v <= a
+            elif op == "eq":
+                k = min_v <= a <= max_v
+            elif op == "ne":
+                k = not (min_v == max_v == a)
+            elif op == "between":
+                b = value_pred[2]
+                k = max_v >= a and min_v <= b
+        keep.append(bool(k))
+    return keep
