@@ -2019,45 +2019,17 @@ __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
         const int64_t *t = a.ts[f] + g.row_off;
         uint8_t *fl = a.flags[f] + g.row_off;
         long long cnt = 0;
-        /* contiguous per-thread tiles + per-newer-stream cursors: the
-           lower-bound rank advances monotonically within a tile, so the
-           per-element binary search of round 1 (log2(n) dependent global
-           loads x (k-1) streams) becomes an amortized O(1) gallop.  The
-           cursor array stays in registers via the unrolled fixed-bound
-           loop (f is wave-uniform, the guards are scalar branches). */
-        int64_t T = (g.nrows + blockDim.x - 1) / blockDim.x;
-        int64_t j0 = int64_t(threadIdx.x) * T;
-        int64_t j1 = j0 + T < g.nrows ? j0 + T : g.nrows;
-        int32_t curs[GS_MAX_STREAMS];
-        if (j0 < j1) {
-            int64_t x0 = t[j0];
-#pragma unroll
-            for (int f2 = 0; f2 < GS_MAX_STREAMS; f2++) {
-                if (f2 <= f || f2 >= nsets) continue;
-                DevGroup g2 = a.groups[f2][s];
-                const int64_t *t2 = a.ts[f2] + g2.row_off;
-                int64_t lo = 0, hi2 = g2.nrows;
-                while (lo < hi2) {
-                    int64_t m = (lo + hi2) >> 1;
-                    if (t2[m] < x0) lo = m + 1; else hi2 = m;
-                }
-                curs[f2] = int32_t(lo);
-            }
-        }
-        for (int64_t j = j0; j < j1; j++) {
+        for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
             int64_t x = t[j];
-            if (j > 0 && t[j - 1] >= x)
-                atomicOr(err, DERR_FORMAT); /* "data in stream is not sorted" */
+            if (j > 0 && t[j - 1] >= x) atomicOr(err, DERR_FORMAT); /* "data in stream is not sorted" */
             bool owner = true;
-#pragma unroll
-            for (int f2 = 0; f2 < GS_MAX_STREAMS; f2++) {
-                if (f2 <= f || f2 >= nsets) continue;
+            for (int f2 = f + 1; f2 < nsets; f2++) {
                 DevGroup g2 = a.groups[f2][s];
-                const int64_t *t2 = a.ts[f2] + g2.row_off;
-                int32_t pos = curs[f2];
-                while (pos < g2.nrows && t2[pos] < x) pos++;
-                curs[f2] = pos;
-                if (pos < g2.nrows && t2[pos] == x) owner = false;
+                int64_t pos;
+                if (dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &pos)) {
+                    owner = false;
+                    break;
+                }
             }
             fl[j] = owner;
             cnt += owner;
@@ -2072,7 +2044,6 @@ __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
         }
         if (threadIdx.x == 0)
             a.counts[size_t(f) * nseries + s] = sred[0];
-        __syncthreads();
     }
 }
 
@@ -2122,62 +2093,37 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
         const uint8_t *fl = a.flags[f] + g.row_off;
         const int32_t *pf = a.prefix[f] + g.row_off;
         int64_t base = a.out_off[s];
-        /* contiguous tiles + galloping cursors, as k_cm_flags: the merged
-           position (owners with smaller ts across all streams) and the
-           dedup hits come from the cursor's lower bound instead of a
-           fresh binary search per element (batch_builder.rs:106-155
-           semantics: this element is the owner, so no NEWER stream
-           contains x; the newest non-null among {f, older hits} wins,
-           all-null -> null) */
-        int64_t T = (g.nrows + blockDim.x - 1) / blockDim.x;
-        int64_t j0 = int64_t(threadIdx.x) * T;
-        int64_t j1 = j0 + T < g.nrows ? j0 + T : g.nrows;
-        int32_t curs[GS_MAX_STREAMS];
-        if (j0 < j1) {
-            int64_t x0 = t[j0];
-#pragma unroll
-            for (int f2 = 0; f2 < GS_MAX_STREAMS; f2++) {
-                if (f2 == f || f2 >= nsets) continue;
-                DevGroup g2 = a.groups[f2][s];
-                const int64_t *t2 = a.ts[f2] + g2.row_off;
-                int64_t lo = 0, hi2 = g2.nrows;
-                while (lo < hi2) {
-                    int64_t m = (lo + hi2) >> 1;
-                    if (t2[m] < x0) lo = m + 1; else hi2 = m;
-                }
-                curs[f2] = int32_t(lo);
-            }
-        }
-        for (int64_t j = j0; j < j1; j++) {
-            if (!fl[j]) continue; /* cursors lag, the gallop catches up */
+        for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
+            if (!fl[j]) continue;
             int64_t x = t[j];
+            /* merged position (owners with smaller ts across all streams)
+               and, in the same pass, where x occurs in OLDER streams —
+               the dedup value walk reuses those hits instead of searching
+               again (batch_builder.rs:139-151 semantics: this element is
+               the owner, so no NEWER stream contains x; the newest
+               non-null among {f, older hits} wins, all-null -> null) */
             int64_t pos = base + pf[j];
             int64_t hit[GS_MAX_STREAMS];
-            hit[f] = g.row_off + j;
-#pragma unroll
-            for (int f2 = 0; f2 < GS_MAX_STREAMS; f2++) {
-                if (f2 == f || f2 >= nsets) continue;
+            for (int f2 = 0; f2 < nsets; f2++) {
+                if (f2 == f) { hit[f2] = g.row_off + j; continue; }
                 DevGroup g2 = a.groups[f2][s];
-                const int64_t *t2 = a.ts[f2] + g2.row_off;
-                int32_t p2 = curs[f2];
-                while (p2 < g2.nrows && t2[p2] < x) p2++;
-                curs[f2] = p2;
-                bool found = p2 < int32_t(g2.nrows) && t2[p2] == x;
-                hit[f2] = (found && f2 < f) ? g2.row_off + p2 : -1;
-                pos += (p2 > 0) ? a.prefix[f2][g2.row_off + p2 - 1] +
-                                      a.flags[f2][g2.row_off + p2 - 1]
-                                : 0;
+                int64_t ins;
+                bool found =
+                    dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &ins);
+                hit[f2] = (found && f2 < f) ? g2.row_off + ins : -1;
+                pos += (ins > 0) ? a.prefix[f2][g2.row_off + ins - 1] +
+                                       a.flags[f2][g2.row_off + ins - 1]
+                                 : 0;
             }
             double v = 0.0;
             uint8_t ok = 0;
-#pragma unroll
-            for (int fi = GS_MAX_STREAMS - 1; fi >= 0; fi--) {
-                if (fi > f || ok) continue;
-                if (hit[fi] < 0) continue;
-                const uint8_t *vd2 = a.valid[fi];
-                if (!vd2 || vd2[hit[fi]]) {
-                    v = a.val[fi][hit[fi]];
+            for (int f2 = f; f2 >= 0; f2--) {
+                if (hit[f2] < 0) continue;
+                const uint8_t *vd2 = a.valid[f2];
+                if (!vd2 || vd2[hit[f2]]) {
+                    v = a.val[f2][hit[f2]];
                     ok = 1;
+                    break;
                 }
             }
             out_ts[pos] = x;
