@@ -2011,28 +2011,6 @@ __device__ __forceinline__ bool dev_ts_contains(const int64_t *t, int64_t n,
 }
 
 /* block per (set, series): owner flags + per-series owner count */
-/* lower_bound of x in sorted t2[0..n2), starting from a lagging cursor:
- * gallop (doubling probes) to bracket, then binary-search the bracket.
- * For a lane whose elements are blockDim rows apart this costs ~log2 of
- * the local density instead of log2(n2) (k_cm_* kernels). */
-__device__ __forceinline__ int32_t dev_lb_gallop(const int64_t *t2,
-                                                 int32_t n2, int32_t cur,
-                                                 int64_t x) {
-    if (cur >= n2 || t2[cur] >= x) return cur;
-    int32_t step = 1, lo = cur, hi = cur + 1;
-    while (hi < n2 && t2[hi] < x) {
-        lo = hi;
-        step <<= 1;
-        hi = (n2 - hi > step) ? hi + step : n2;
-    }
-    /* t2[lo] < x <= t2[hi] (or hi==n2) */
-    while (lo + 1 < hi) {
-        int32_t m = (lo + hi) >> 1;
-        if (t2[m] < x) lo = m; else hi = m;
-    }
-    return hi;
-}
-
 __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
                            unsigned *__restrict__ err) {
     int f = blockIdx.y;
@@ -2041,26 +2019,17 @@ __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
         const int64_t *t = a.ts[f] + g.row_off;
         uint8_t *fl = a.flags[f] + g.row_off;
         long long cnt = 0;
-        /* strided (coalesced) element loop; per-lane per-stream cursors
-           turn each rank lookup into a gallop over the ~blockDim-row gap
-           since the lane's previous element (registers via the unrolled
-           fixed-bound loop; f is wave-uniform so the guards are scalar) */
-        int32_t curs[GS_MAX_STREAMS];
-#pragma unroll
-        for (int f2 = 0; f2 < GS_MAX_STREAMS; f2++) curs[f2] = 0;
         for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
             int64_t x = t[j];
             if (j > 0 && t[j - 1] >= x) atomicOr(err, DERR_FORMAT); /* "data in stream is not sorted" */
             bool owner = true;
-#pragma unroll
-            for (int f2 = 0; f2 < GS_MAX_STREAMS; f2++) {
-                if (f2 <= f || f2 >= nsets) continue;
+            for (int f2 = f + 1; f2 < nsets; f2++) {
                 DevGroup g2 = a.groups[f2][s];
-                const int64_t *t2 = a.ts[f2] + g2.row_off;
-                int32_t pos = dev_lb_gallop(t2, int32_t(g2.nrows),
-                                            curs[f2], x);
-                curs[f2] = pos;
-                if (pos < int32_t(g2.nrows) && t2[pos] == x) owner = false;
+                int64_t pos;
+                if (dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &pos)) {
+                    owner = false;
+                    break;
+                }
             }
             fl[j] = owner;
             cnt += owner;
@@ -2124,45 +2093,37 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
         const uint8_t *fl = a.flags[f] + g.row_off;
         const int32_t *pf = a.prefix[f] + g.row_off;
         int64_t base = a.out_off[s];
-        /* strided + cursor-gallop ranks, as k_cm_flags; the merged
-           position (owners with smaller ts across all streams) and the
-           dedup hits come from the cursor's lower bound
-           (batch_builder.rs:106-155 semantics: this element is the
-           owner, so no NEWER stream contains x; the newest non-null
-           among {f, older hits} wins, all-null -> null) */
-        int32_t curs[GS_MAX_STREAMS];
-#pragma unroll
-        for (int f2 = 0; f2 < GS_MAX_STREAMS; f2++) curs[f2] = 0;
         for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
-            if (!fl[j]) continue; /* cursors lag; the gallop catches up */
+            if (!fl[j]) continue;
             int64_t x = t[j];
+            /* merged position (owners with smaller ts across all streams)
+               and, in the same pass, where x occurs in OLDER streams —
+               the dedup value walk reuses those hits instead of searching
+               again (batch_builder.rs:139-151 semantics: this element is
+               the owner, so no NEWER stream contains x; the newest
+               non-null among {f, older hits} wins, all-null -> null) */
             int64_t pos = base + pf[j];
             int64_t hit[GS_MAX_STREAMS];
-            hit[f] = g.row_off + j;
-#pragma unroll
-            for (int f2 = 0; f2 < GS_MAX_STREAMS; f2++) {
-                if (f2 == f || f2 >= nsets) continue;
+            for (int f2 = 0; f2 < nsets; f2++) {
+                if (f2 == f) { hit[f2] = g.row_off + j; continue; }
                 DevGroup g2 = a.groups[f2][s];
-                const int64_t *t2 = a.ts[f2] + g2.row_off;
-                int32_t p2 = dev_lb_gallop(t2, int32_t(g2.nrows),
-                                           curs[f2], x);
-                curs[f2] = p2;
-                bool found = p2 < int32_t(g2.nrows) && t2[p2] == x;
-                hit[f2] = (found && f2 < f) ? g2.row_off + p2 : -1;
-                pos += (p2 > 0) ? a.prefix[f2][g2.row_off + p2 - 1] +
-                                      a.flags[f2][g2.row_off + p2 - 1]
-                                : 0;
+                int64_t ins;
+                bool found =
+                    dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &ins);
+                hit[f2] = (found && f2 < f) ? g2.row_off + ins : -1;
+                pos += (ins > 0) ? a.prefix[f2][g2.row_off + ins - 1] +
+                                       a.flags[f2][g2.row_off + ins - 1]
+                                 : 0;
             }
             double v = 0.0;
             uint8_t ok = 0;
-#pragma unroll
-            for (int fi = GS_MAX_STREAMS - 1; fi >= 0; fi--) {
-                if (fi > f || ok) continue;
-                if (hit[fi] < 0) continue;
-                const uint8_t *vd2 = a.valid[fi];
-                if (!vd2 || vd2[hit[fi]]) {
-                    v = a.val[fi][hit[fi]];
+            for (int f2 = f; f2 >= 0; f2--) {
+                if (hit[f2] < 0) continue;
+                const uint8_t *vd2 = a.valid[f2];
+                if (!vd2 || vd2[hit[f2]]) {
+                    v = a.val[f2][hit[f2]];
                     ok = 1;
+                    break;
                 }
             }
             out_ts[pos] = x;
