@@ -2847,6 +2847,7 @@ struct GsGroupSet {
                                       groups merged) for aggregation */
     int nsgroups = 0;
     std::vector<int32_t> sgroup_span; /* page-groups per series-group */
+    std::vector<int64_t> sgroup_row_off; /* first row of each series-group */
     DevGroup *d_sgroups_out = nullptr; /* series-group layout of the fused
                                           compacted output (rebuilt per scan) */
     int32_t *d_sgroup_first = nullptr; /* [nsgroups+1] first page-group idx */
@@ -3263,6 +3264,7 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
         } else {
             hsg.push_back(hg[g]);
             set->sgroup_span.push_back(1);
+            set->sgroup_row_off.push_back(hg[g].row_off);
         }
     }
     set->nsgroups = int(hsg.size());
@@ -3438,6 +3440,109 @@ GsStatus gs_set_row_offsets(const GsGroupSet *set, int64_t *out) {
     return GS_OK;
 }
 
+/* --------------------------------------------- GROUP BY tag
+ * TSBS GROUP-BY-hostname over the decoded varbinary tag column.  In the
+ * reference a tag is part of the SeriesKey and therefore CONSTANT within
+ * a series (tag columns are materialized per series from the key,
+ * reader/series.rs:23-100); the GPU group-by exploits exactly that:
+ * hash the tag of each series' first row into a device dictionary
+ * (first-occurrence = min series index, deterministic), then merge the
+ * per-series bucket partials of the preceding aggregate scan per
+ * (tag, bucket) in series order — a deterministic fixed-order reduction,
+ * like the aggregate itself. */
+__device__ __forceinline__ uint64_t dev_fnv1a(const uint8_t *p, int64_t n) {
+    uint64_t h = 1469598103934665603ULL;
+    for (int64_t i = 0; i < n; i++) {
+        h ^= p[i];
+        h *= 1099511628211ULL;
+    }
+    return h;
+}
+
+/* dictionary slot, one atomic u64: [fp32 (nonzero) | min_series32].
+ * Claim by CAS from 0; equal-tag series then atomicMin the whole word —
+ * the high fp bits are equal so the min acts on the series index,
+ * keeping the deterministic first-occurrence owner.  fp collisions are
+ * resolved by byte comparison against the slot owner's tag (the owner
+ * field is valid from the moment the slot exists — single-word claim). */
+__global__ void k_tag_series(const DevGroup *__restrict__ sg, int nsg,
+                             const int64_t *__restrict__ str_pos,
+                             const int64_t *__restrict__ str_sz,
+                             const uint8_t *__restrict__ scratch,
+                             unsigned long long *__restrict__ table,
+                             int cap_mask, int32_t *__restrict__ slot_of,
+                             unsigned *__restrict__ err) {
+    for (int g = blockIdx.x * blockDim.x + threadIdx.x; g < nsg;
+         g += gridDim.x * blockDim.x) {
+        const int64_t r0 = sg[g].row_off; /* tag of the series' first row */
+        const uint8_t *tb = scratch + str_pos[r0];
+        const int64_t tn = str_sz[r0];
+        const uint64_t fphi = (dev_fnv1a(tb, tn) >> 32) | 1ULL;
+        const unsigned long long claim =
+            (fphi << 32) | (unsigned long long)(uint32_t)g;
+        int slot = int(fphi * 0x9e3779b1u) & cap_mask;
+        bool placed = false;
+        for (int probes = 0; probes <= cap_mask && !placed; probes++) {
+            unsigned long long cur = table[slot];
+            if (cur == 0) cur = atomicCAS(&table[slot], 0ULL, claim);
+            if (cur == 0 || (cur >> 32) == fphi) {
+                bool same = true;
+                if (cur != 0) { /* verify bytes vs the slot owner */
+                    const int own = int(cur & 0xffffffffULL);
+                    if (own != g) {
+                        const int64_t ro = sg[own].row_off;
+                        if (str_sz[ro] != tn) same = false;
+                        else {
+                            const uint8_t *ob = scratch + str_pos[ro];
+                            for (int64_t i = 0; i < tn && same; i++)
+                                same = ob[i] == tb[i];
+                        }
+                    }
+                }
+                if (same) {
+                    atomicMin(&table[slot], claim);
+                    slot_of[g] = slot;
+                    placed = true;
+                    break;
+                }
+            }
+            slot = (slot + 1) & cap_mask;
+        }
+        if (!placed) atomicOr(err, DERR_FORMAT); /* table full */
+    }
+}
+
+/* thread per (gid, bucket): ordered walk of the gid's series list */
+__global__ void k_agg_tag(const int32_t *__restrict__ csr_off,
+                          const int32_t *__restrict__ csr_sg, int ngids,
+                          int nbuckets, const double *__restrict__ pmax,
+                          const double *__restrict__ psum,
+                          const long long *__restrict__ pcnt,
+                          double *__restrict__ out_max,
+                          double *__restrict__ out_sum,
+                          long long *__restrict__ out_cnt) {
+    int64_t total = int64_t(ngids) * nbuckets;
+    for (int64_t i = blockIdx.x * int64_t(blockDim.x) + threadIdx.x;
+         i < total; i += int64_t(gridDim.x) * blockDim.x) {
+        const int gid = int(i / nbuckets);
+        const int b = int(i % nbuckets);
+        double mx = -__builtin_inf(), sm = 0.0;
+        long long c = 0;
+        for (int32_t k = csr_off[gid]; k < csr_off[gid + 1]; k++) {
+            const size_t cell = size_t(csr_sg[k]) * nbuckets + b;
+            const long long pc = pcnt[cell];
+            if (pc <= 0) continue;
+            const double m2 = pmax[cell];
+            if (m2 > mx) mx = m2;
+            sm += psum[cell];
+            c += pc;
+        }
+        out_max[i] = mx;
+        out_sum[i] = sm;
+        out_cnt[i] = c;
+    }
+}
+
 static GsStatus check_dev_err(GsCtx *ctx) {
     unsigned e = 0;
     HIP_TRY(hipMemcpy(&e, ctx->d_err, sizeof(unsigned), hipMemcpyDeviceToHost));
@@ -3533,6 +3638,106 @@ GsStatus gs_decode(GsCtx *ctx, GsGroupSet *set, uint32_t col, void *d_out,
                            sp.dev[PC_RLE_I64], sp.n[PC_RLE_I64], (int64_t *)d_out,
                            d_valid, 0, ctx->d_err);
     HIP_TRY(hipStreamSynchronize(ctx->stream));
+    return check_dev_err(ctx);
+}
+
+/* GROUP BY tag over a completed aggregate scan: the tag column (decoded
+ * by gs_decode_str: per-row pos/sz into its scratch) keys a dictionary of
+ * the per-series tag (constant within a series, SeriesKey semantics);
+ * outputs per-(tag, bucket) max/sum/count in deterministic
+ * first-occurrence tag order.  out arrays are caller device buffers of
+ * cap_gids * n_buckets; tag_rep_row[gid] (host) = a row whose string is
+ * the tag value.  Requires the previous gs_scan on this set to have run
+ * with the same n_buckets (the per-series partials are reused). */
+GsStatus gs_groupby_tag(GsCtx *ctx, GsGroupSet *set, int tag_col,
+                        int n_buckets, double *d_out_max, double *d_out_sum,
+                        long long *d_out_count, int64_t *tag_rep_row,
+                        int cap_gids, int *out_ngids) {
+    if (!ctx || !set || !d_out_max || !d_out_sum || !d_out_count ||
+        !out_ngids || n_buckets <= 0)
+        return fail(GS_ERR, "bad args to gs_groupby_tag");
+    if (!set->d_pmax ||
+        set->partials_cap < size_t(set->nsgroups) * size_t(n_buckets))
+        return fail(GS_ERR, "gs_groupby_tag needs a prior aggregate scan");
+    if (!set->d_str_sz || !set->d_str_pos || !set->d_str_scratch)
+        return fail(GS_ERR, "gs_groupby_tag needs a decoded string column");
+    (void)tag_col;
+    HIP_TRY(hipSetDevice(ctx->device));
+    const int nsg = set->nsgroups;
+    int cap = 64;
+    while (cap < 2 * nsg) cap <<= 1;
+    unsigned long long *d_table = nullptr;
+    int32_t *d_slot_of = nullptr;
+    if (hipMalloc(&d_table, size_t(cap) * 8) != hipSuccess ||
+        hipMalloc(&d_slot_of, size_t(nsg) * 4) != hipSuccess) {
+        if (d_table) hipFree(d_table);
+        return fail(GS_ERR, "hipMalloc tag table failed");
+    }
+    HIP_TRY(hipMemsetAsync(d_table, 0, size_t(cap) * 8, ctx->stream));
+    hipLaunchKernelGGL(k_tag_series, dim3(grid_for(nsg, 256)), dim3(256), 0,
+                       ctx->stream, set->d_sgroups, nsg, set->d_str_pos,
+                       set->d_str_sz, set->d_str_scratch, d_table, cap - 1,
+                       d_slot_of, ctx->d_err);
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    GsStatus st = check_dev_err(ctx);
+    if (st != GS_OK) { hipFree(d_table); hipFree(d_slot_of); return st; }
+    /* host finalize (tiny, deterministic): dense ids in first-occurrence
+       order + CSR of the series of each tag */
+    std::vector<unsigned long long> table(cap);
+    std::vector<int32_t> slot_of(nsg);
+    HIP_TRY(hipMemcpy(table.data(), d_table, size_t(cap) * 8,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(slot_of.data(), d_slot_of, size_t(nsg) * 4,
+                      hipMemcpyDeviceToHost));
+    hipFree(d_table);
+    hipFree(d_slot_of);
+    std::vector<std::pair<int, int>> owners; /* (min series, slot) */
+    for (int sl = 0; sl < cap; sl++)
+        if (table[sl])
+            owners.push_back({int(table[sl] & 0xffffffffULL), sl});
+    std::sort(owners.begin(), owners.end());
+    const int ngids = int(owners.size());
+    *out_ngids = ngids;
+    if (ngids > cap_gids)
+        return fail(GS_ERR, "gs_groupby_tag: cap_gids too small");
+    std::vector<int32_t> gid_of_slot(cap, -1);
+    for (int gidx = 0; gidx < ngids; gidx++)
+        gid_of_slot[owners[gidx].second] = gidx;
+    std::vector<int32_t> csr_off(ngids + 1, 0), csr_sg(nsg);
+    for (int g = 0; g < nsg; g++) csr_off[gid_of_slot[slot_of[g]] + 1]++;
+    for (int gidx = 0; gidx < ngids; gidx++) csr_off[gidx + 1] += csr_off[gidx];
+    {
+        std::vector<int32_t> cur(csr_off.begin(), csr_off.end() - 1);
+        for (int g = 0; g < nsg; g++) /* series order within a tag */
+            csr_sg[cur[gid_of_slot[slot_of[g]]]++] = g;
+    }
+    if (tag_rep_row)
+        for (int gidx = 0; gidx < ngids; gidx++) {
+            int sgi = owners[gidx].first;
+            tag_rep_row[gidx] = size_t(sgi) < set->sgroup_row_off.size()
+                                    ? set->sgroup_row_off[sgi] : -1;
+        }
+    int32_t *d_csr_off = nullptr, *d_csr_sg = nullptr;
+    if (hipMalloc(&d_csr_off, (ngids + 1) * 4) != hipSuccess ||
+        hipMalloc(&d_csr_sg, size_t(nsg) * 4) != hipSuccess) {
+        if (d_csr_off) hipFree(d_csr_off);
+        return fail(GS_ERR, "hipMalloc tag csr failed");
+    }
+    hipMemcpyAsync(d_csr_off, csr_off.data(), (ngids + 1) * 4,
+                   hipMemcpyHostToDevice, ctx->stream);
+    hipMemcpyAsync(d_csr_sg, csr_sg.data(), size_t(nsg) * 4,
+                   hipMemcpyHostToDevice, ctx->stream);
+    hipStreamSynchronize(ctx->stream);
+    const int64_t total = int64_t(ngids) * n_buckets;
+    const int blocks = int(total > int64_t(2048) * 256
+                               ? 2048 : (total + 255) / 256);
+    hipLaunchKernelGGL(k_agg_tag, dim3(blocks < 1 ? 1 : blocks), dim3(256),
+                       0, ctx->stream, d_csr_off, d_csr_sg, ngids, n_buckets,
+                       set->d_pmax, set->d_psum, set->d_pcnt, d_out_max,
+                       d_out_sum, d_out_count);
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    hipFree(d_csr_off);
+    hipFree(d_csr_sg);
     return check_dev_err(ctx);
 }
 

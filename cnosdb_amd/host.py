@@ -669,3 +669,32 @@ def prune_column_groups(stats, time_range=None, value_pred=None):
                 k = max_v >= a and min_v <= b
         keep.append(bool(k))
     return keep
+
+
+def _bind_groupby(lib):
+    import ctypes
+    if getattr(lib, "_gb_bound", False):
+        return
+    lib.gs_groupby_tag.restype = ctypes.c_int
+    lib.gs_groupby_tag.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int, ctypes.POINTER(ctypes.c_int)]
+    lib._gb_bound = True
+
+
+def groupby_tag(engine, gset, n_buckets, d_max, d_sum, d_count,
+                cap_gids, tag_col=0):
+    """GROUP BY tag over a completed aggregate scan + decoded tag column
+    (gs_groupby_tag).  Returns (ngids, tag_rep_rows ndarray)."""
+    import ctypes
+    _bind_groupby(engine.lib)
+    rep = np.zeros(cap_gids, dtype=np.int64)
+    ng = ctypes.c_int(0)
+    st = engine.lib.gs_groupby_tag(
+        engine._ctx, gset._h, tag_col, n_buckets,
+        d_max.data_ptr(), d_sum.data_ptr(), d_count.data_ptr(),
+        rep.ctypes.data_as(ctypes.c_void_p), cap_gids, ctypes.byref(ng))
+    if st != 0:
+        raise RuntimeError(f"gs_groupby_tag failed ({st}): {engine._pl.err()}")
+    return ng.value, rep[:ng.value]

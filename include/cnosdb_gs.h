@@ -268,6 +268,20 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
  * synchronizing; gs_scan_wait synchronizes the ctx stream and fills the
  * result.  Lets a caller overlap sub-batches on two ctx streams (the
  * Gorilla decode is ALU-bound, the aggregate HBM-bound). */
+/* GROUP BY tag over a completed aggregate scan: keys the
+ * per-series bucket partials by the decoded varbinary tag column (tags
+ * are SeriesKey members, constant within a series —
+ * tskv/src/reader/series.rs:23-100; the group-by above TskvExec is stock
+ * DataFusion hash-agg in the reference).  Outputs per-(tag, bucket)
+ * max/sum/count in deterministic first-occurrence tag order;
+ * tag_rep_row[gid] = a row index whose string value IS the tag.
+ * Precondition: gs_scan with the same n_buckets, and gs_decode_str of
+ * the tag column, both on this set. */
+GsStatus gs_groupby_tag(GsCtx *ctx, GsGroupSet *set, int tag_col,
+                        int n_buckets, double *d_out_max, double *d_out_sum,
+                        long long *d_out_count, int64_t *tag_rep_row,
+                        int cap_gids, int *out_ngids);
+
 GsStatus gs_scan_async(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec);
 GsStatus gs_scan_wait(GsCtx *ctx, GsGroupSet *set, GsScanResult *result);
 
