@@ -671,9 +671,108 @@ def bench_general(args):
     print(json.dumps(line))
 
 
+
+def bench_groupby(args):
+    """GROUP BY tag leg (SURVEY 8f continuation): fused aggregate scan +
+    gs_groupby_tag keyed on the decoded varbinary tag column (tags are
+    SeriesKey members, constant per series).  One step = scan (decode +
+    filter + per-series 5-min buckets) + per-(tag, bucket) merge.  The
+    tag column itself is decoded once at setup (its decode rate is the
+    strings benchmark line)."""
+    import torch
+    import cnosdb_amd as gs
+    rng = np.random.default_rng(231)
+    nseries, npts, page_rows = args.series, args.npts, args.page_rows
+    npages = npts // page_rows
+    uniq_tags = 256
+    tagpool = [b"host_%06d" % i for i in range(uniq_tags)]
+    t_setup = time.perf_counter()
+    tpages = [gs.page_of(
+        T0 + (np.arange(page_rows, dtype=np.int64) + p * page_rows) * NS,
+        gs.CT_TIME) for p in range(npages)]
+    uniq_f = min(64, nseries)
+    fpool = [gs.page_of(np.round(np.clip(
+        np.cumsum(rng.normal(0, 0.5, page_rows)) + 50, 0, 100), 1),
+        gs.CT_F64) for _ in range(uniq_f)]
+    spool = {}
+    groups = []
+    for s_ in range(nseries):
+        tag = tagpool[s_ % uniq_tags]
+        if tag not in spool:
+            spool[tag] = gs.str_page_of([tag] * page_rows)
+        for p in range(npages):
+            groups.append((s_, [(tpages[p], gs.CT_TIME),
+                                (fpool[(s_ + p) % uniq_f], gs.CT_F64),
+                                (spool[tag], gs.CT_STR)]))
+    eng = gs.Engine(0)
+    gset = eng.upload(groups)
+    rows = gset.rows
+    lo = T0 + int(0.25 * npts) * NS
+    hi = T0 + int(0.75 * npts) * NS - 1
+    nb = int(npts * NS // BUCKET_NS) + 1
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    agg = dict(bucket_ns=BUCKET_NS, t0=T0, n_buckets=nb,
+               d_max=torch.full((nb,), -np.inf, dtype=torch.float64,
+                                device="cuda"),
+               d_sum=torch.zeros(nb, dtype=torch.float64, device="cuda"),
+               d_count=torch.zeros(nb, dtype=torch.int64, device="cuda"))
+    d_soff = torch.zeros(rows + 1, dtype=torch.int64, device="cuda")
+    d_sbytes = torch.zeros(rows * 12, dtype=torch.uint8, device="cuda")
+    eng.decode_str(gset, 2, d_soff, d_sbytes)  # setup: tag column resident
+    cap = 512
+    g_max = torch.zeros(cap * nb, dtype=torch.float64, device="cuda")
+    g_sum = torch.zeros(cap * nb, dtype=torch.float64, device="cuda")
+    g_cnt = torch.zeros(cap * nb, dtype=torch.int64, device="cuda")
+    setup_s = time.perf_counter() - t_setup
+
+    def step():
+        eng.scan(gset, d_ts, d_val, time_range=(lo, hi),
+                 d_out_ts=d_ots, d_out_val=d_oval, agg=agg)
+        return gs.groupby_tag(eng, gset, nb, g_max, g_sum, g_cnt, cap)
+
+    for _ in range(args.warmup):
+        ngids, _ = step()
+    torch.cuda.synchronize()
+    t0_ = time.perf_counter()
+    for _ in range(args.steps):
+        ngids, _ = step()
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0_
+    # isolate the group-by increment
+    torch.cuda.synchronize()
+    t1_ = time.perf_counter()
+    for _ in range(args.steps):
+        gs.groupby_tag(eng, gset, nb, g_max, g_sum, g_cnt, cap)
+    torch.cuda.synchronize()
+    gb_ms = (time.perf_counter() - t1_) / args.steps * 1000
+    value = rows * args.steps / dt
+    line = {
+        "metric": "decoded+filtered values/sec & HBM GB/s, TSBS devops scan, 1/2/4/8 GPU",
+        "value": value, "unit": "values/s", "n_gpus": 1,
+        "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1000,
+        "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+        "dtype": "f64", "data": "synthetic",
+        "config": {
+            "workload": "TSBS GROUP-BY-hostname: fused agg scan + "
+                        "per-(tag,bucket) merge over the decoded tag column "
+                        "(256 hostnames)",
+            "series_per_gpu": nseries, "points_per_series": npts,
+            "page_rows": page_rows, "n_tags": int(ngids),
+            "n_buckets": nb, "groupby_ms_per_step": gb_ms,
+            "setup_s": round(setup_s, 1),
+        },
+        "roofline": None, "cpu_baseline": None,
+    }
+    print(json.dumps(line))
+
+
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--mode", choices=["scan", "compact", "strings", "general"], default="scan")
+    ap.add_argument("--mode", choices=["scan", "compact", "strings", "general", "groupby"], default="scan")
     ap.add_argument("--gpus", type=int, default=1,
                     help="driver contract flag; the actual world size comes "
                          "from the torchrun environment (WORLD_SIZE)")
@@ -681,7 +780,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--series", type=int, default=10000)
     ap.add_argument("--npts", type=int, default=1_000_000)
-    ap.add_argument("--page-rows", type=int, default=4000)
+    ap.add_argument("--page-rows", type=int, default=125000,
+                help="rows per page; default = reference-shaped (1M pts / 8 pages,\n                max_datablock_size ~100KiB rows, comapcting_block_meta_group.rs:87)")
     ap.add_argument("--sub-batches", type=int, default=2)
     ap.add_argument("--unique", type=int, default=256)
     ap.add_argument("--fields", type=int, default=1,
@@ -702,7 +802,16 @@ def main():
     if args.mode == "general":
         if args.series == 10000:
             args.series = 2500  # one resident set (ts+val+outs+masks ~34 B/row)
+        if args.page_rows == 125000:
+            args.page_rows = 4000  # null pages: chunked GORN path, smaller pages
         bench_general(args)
+        return
+    if args.mode == "groupby":
+        if args.series == 10000:
+            args.series = 2000
+        if args.npts == 1_000_000:
+            args.npts = 250_000
+        bench_groupby(args)
         return
 
     import torch
@@ -850,7 +959,15 @@ def main():
     if os.path.exists(pmc_path):
         try:
             pmc = json.load(open(pmc_path))
-            if pmc.get("kernel") == "k_gor_lds_filtered":
+            shape = pmc.get("launch_shape", {})
+            # attach only when the PMC collection ran the same launch
+            # shape (a stale half-size traffic number
+            # was attached to a full-size run)
+            if (pmc.get("kernel") == "k_gor_chunks_filtered"
+                    and shape.get("rows_per_launch") == sb_rows
+                    and shape.get("page_rows") == page_rows
+                    and shape.get("sub_batches") == args.sub_batches
+                    and shape.get("series") == nseries):
                 traffic = pmc.get("bytes_per_launch")
         except Exception:
             pass
@@ -876,7 +993,7 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": "tsbs-devops-scan (BASELINE configs[1]: 10k series x 1M pts, delta-ts + Gorilla-f64 decode, ts-range filter 50%, fused 5-min max/sum/count)",
+                "workload": "tsbs-devops-scan (BASELINE configs[1]: 10k series x 1M pts, reference-shaped 125k-row pages, delta-ts + Gorilla-f64 decode, ts-range filter 50%, fused 5-min max/sum/count)",
                 "series_per_gpu": nseries,
                 "fields": args.fields,
                 "points_per_series": npts,
