@@ -797,6 +797,8 @@ def main():
         bench_compact(args)
         return
     if args.mode == "strings":
+        if args.page_rows == 125000:
+            args.page_rows = 4000  # round-1 strings shape: 60k x 4000-row pages
         bench_strings(args)
         return
     if args.mode == "general":
@@ -808,7 +810,7 @@ def main():
         return
     if args.mode == "groupby":
         if args.series == 10000:
-            args.series = 2000
+            args.series = 1000  # decode_str per-set row cap is 268M
         if args.npts == 1_000_000:
             args.npts = 250_000
         bench_groupby(args)
