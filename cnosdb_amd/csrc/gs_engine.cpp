@@ -1855,7 +1855,17 @@ __device__ int64_t dev_snappy_decompress(const uint8_t *__restrict__ src,
             }
             l += 1;
             if (ip + l > end || op + l > op_end) return -1;
-            for (uint32_t k = 0; k < l; k++) op[k] = ip[k];
+            { /* word-wide literal copy (disjoint buffers; bounds above
+                 guarantee the 8-B reads/writes stay inside [ip,ip+l) /
+                 [op,op+l)) */
+                uint32_t k = 0;
+                for (; k + 8 <= l; k += 8) {
+                    uint64_t w;
+                    __builtin_memcpy(&w, ip + k, 8);
+                    __builtin_memcpy(op + k, &w, 8);
+                }
+                for (; k < l; k++) op[k] = ip[k];
+            }
             ip += l;
             op += l;
             continue;
@@ -1881,7 +1891,17 @@ __device__ int64_t dev_snappy_decompress(const uint8_t *__restrict__ src,
         }
         if (off == 0 || uint64_t(op - dst) < off || op + l > op_end) return -1;
         const uint8_t *cp = op - off; /* may overlap: strictly in order */
-        for (uint32_t k = 0; k < l; k++) op[k] = cp[k];
+        if (off >= 8) { /* no overlap within a word: copy 8 B at a time */
+            uint32_t k = 0;
+            for (; k + 8 <= l; k += 8) {
+                uint64_t w;
+                __builtin_memcpy(&w, cp + k, 8);
+                __builtin_memcpy(op + k, &w, 8);
+            }
+            for (; k < l; k++) op[k] = cp[k];
+        } else { /* short offset: byte-serial repeat semantics */
+            for (uint32_t k = 0; k < l; k++) op[k] = cp[k];
+        }
         op += l;
     }
     return (op == op_end && ip == end) ? int64_t(ulen) : -1;
