@@ -137,3 +137,43 @@ def test_str_decode_corrupt_block_errors(engine):
     with pytest.raises(RuntimeError):
         engine.decode_str(gset, 1, d_off, d_bytes)
     gset.free()
+
+
+def test_strings_scale_parity(engine):
+    """Large-set parity for the word-wide snappy copies: 4096 pages deep
+    into a multi-GB scratch slab must decode byte-identically to the
+    oracle (guards the unaligned-access hazard found on the GORN path)."""
+    rng2 = np.random.default_rng(77)
+    page_rows = 4000
+    npages = 4096
+    pool = []
+    tagpool = [b"host_%06d" % i for i in range(64)]
+    for u in range(32):
+        idx = rng2.integers(0, len(tagpool), page_rows)
+        blk = [tagpool[i] for i in idx]
+        pool.append((gs.str_page_of(blk), blk))
+    groups = []
+    expect = []
+    for p in range(npages):
+        spage, blk = pool[p % 32]
+        ts = np.arange(page_rows, dtype=np.int64) * 1000
+        groups.append((p, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (spage, gs.CT_STR)]))
+        expect.append(blk)
+    gset = engine.upload(groups)
+    rows = gset.rows
+    d_off = torch.zeros(rows + 1, dtype=torch.int64, device="cuda")
+    d_bytes = torch.zeros(rows * 16, dtype=torch.uint8, device="cuda")
+    engine.decode_str(gset, 1, d_off, d_bytes)
+    offs = d_off.cpu().numpy()
+    data = d_bytes.cpu().numpy()
+    # verify pages spread across the whole set, incl. the deepest ones
+    for p in (0, 1, npages // 2, npages - 2, npages - 1):
+        base = p * page_rows
+        for r in (0, 1, page_rows // 2, page_rows - 1):
+            got = data[offs[base + r]:offs[base + r + 1]].tobytes()
+            assert got == expect[p][r], (p, r)
+    # whole-set checksum vs expectation
+    total = sum(len(b) for blk in expect for b in blk)
+    assert int(offs[rows]) == total
+    gset.free()
