@@ -1992,7 +1992,13 @@ __global__ void k_str_gather(const uint8_t *__restrict__ scratch,
         const int64_t n2 = sz[r];
         const uint8_t *s = scratch + pos[r];
         uint8_t *d = out + off[r];
-        for (int64_t k = 0; k < n2; k++) d[k] = s[k];
+        int64_t k = 0;
+        for (; k + 8 <= n2; k += 8) { /* word-wide gather */
+            uint64_t w;
+            __builtin_memcpy(&w, s + k, 8);
+            __builtin_memcpy(d + k, &w, 8);
+        }
+        for (; k < n2; k++) d[k] = s[k];
     }
 }
 
