@@ -636,8 +636,11 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
             }
             rfill = 0;
         };
+        double *rcur = &rslot[0][lane]; /* strength-reduced ring cursor
+                                            (row stride 65*8 B) */
         if (!done && ch.row0 == 0 && r < end) { /* header value = row 0 */
-            rslot[rfill][lane] = __longlong_as_double((long long)st.val);
+            *rcur = __longlong_as_double((long long)st.val);
+            rcur += 65;
             rfill++;
             r++;
         }
@@ -670,13 +673,16 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                     unsigned m_eff = bit0 ? st.meaningful : 0u;
                     unsigned need = shift + m_eff;
                     uint64_t sb;
-                    if (__builtin_expect(need <= 64, 1)) {
-                        /* nb >= 64 after the loop-top topup */
+                    if (__builtin_expect(need < 64, 1)) {
+                        /* nb >= 64 after the loop-top topup; need < 64
+                           keeps the consume branch-free (no k==64 case) */
                         uint64_t w =
                             (st.hi << shift) | (st.lo >> (64 - shift));
                         sb = m_eff ? (w >> ((64 - m_eff) & 63)) : 0;
-                        consume(need);
-                    } else { /* m_eff > 51: rare */
+                        st.hi = (st.hi << need) | (st.lo >> (64 - need));
+                        st.lo <<= need;
+                        st.nb -= int(need);
+                    } else { /* m_eff >= 51: rare */
                         consume(shift);
                         while (st.nb < int(m_eff)) topup();
                         sb = (m_eff == 64) ? st.hi
@@ -692,8 +698,8 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                     }
                 }
                 if (stg && r < end) {
-                    rslot[rfill][lane] =
-                        __longlong_as_double((long long)st.val);
+                    *rcur = __longlong_as_double((long long)st.val);
+                    rcur += 65;
                     rfill++;
                     r++;
                     /* non-last chunk: all rows produced, stop; the last
@@ -920,10 +926,12 @@ __global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
             unsigned m_eff = bit0 ? st.meaningful : 0u;
             unsigned need = shift + m_eff;
             uint64_t sb;
-            if (__builtin_expect(need <= 64, 1)) {
+            if (__builtin_expect(need < 64, 1)) {
                 uint64_t w = (st.hi << shift) | (st.lo >> (64 - shift));
                 sb = m_eff ? (w >> ((64 - m_eff) & 63)) : 0;
-                consume(need);
+                st.hi = (st.hi << need) | (st.lo >> (64 - need));
+                st.lo <<= need;
+                st.nb -= int(need);
             } else {
                 consume(shift);
                 while (st.nb < int(m_eff)) topup();
@@ -2407,15 +2415,17 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
            unconditionally and the compiler batches them under one wait
            (the predicated version serialized ~5 dependent vmcnt(0)s) */
         DevGorChunk ch = chunks[active[have ? ai : 0]];
-        int64_t sel_lo = sp_start[ch.grp];
-        int64_t sel_hi = sel_lo + sp_cnt[ch.grp];
+        /* spans/rows are page-local (< 2^31): 32-bit compares in the
+           per-value hot loop */
+        int32_t sel_lo = int32_t(sp_start[ch.grp]);
+        int32_t sel_hi = int32_t(sel_lo + sp_cnt[ch.grp]);
         if (!have) { sel_lo = 0; sel_hi = 0; }
         double *o = out + out_off[ch.grp] - sel_lo; /* o[r] valid in span */
-        int64_t r = int64_t(ch.row0);
-        int64_t end = r + int64_t(ch.cnt);
+        int32_t r = int32_t(ch.row0);
+        int32_t end = r + int32_t(ch.cnt);
         GorChunkState st = gor_chunk_init(blob, ch);
         int rfill = 0;
-        int64_t run0 = 0; /* output row of first staged entry */
+        int32_t run0 = 0; /* output row of first staged entry */
         bool done = !have, over = false, clean_stop = false;
         if (have && st.bad) { atomicOr(err, DERR_SHORT); done = true; }
         auto topup = [&]() { /* only with nb < 64 */
@@ -2446,7 +2456,7 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
         const int f_idx = lane & (GS_RING - 1);
         const int f_sq = lane / GS_RING;
         auto flush = [&]() {
-            fd[lane][0] = (uint64_t)(uintptr_t)(o + run0);
+            fd[lane][0] = (uint64_t)(uintptr_t)(o + int64_t(run0));
             fd[lane][1] = uint64_t(rfill);
             __builtin_amdgcn_wave_barrier();
             constexpr int SRCP = 64 / GS_RING; /* sources per store */
@@ -2513,13 +2523,16 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
                     unsigned m_eff = bit0 ? st.meaningful : 0u;
                     unsigned need = shift + m_eff;
                     uint64_t sb;
-                    if (__builtin_expect(need <= 64, 1)) {
-                        /* nb >= 64 after the loop-top topup */
+                    if (__builtin_expect(need < 64, 1)) {
+                        /* nb >= 64 after the loop-top topup; need < 64
+                           keeps the consume branch-free (no k==64 case) */
                         uint64_t w =
                             (st.hi << shift) | (st.lo >> (64 - shift));
                         sb = m_eff ? (w >> ((64 - m_eff) & 63)) : 0;
-                        consume(need);
-                    } else { /* m_eff > 51: rare */
+                        st.hi = (st.hi << need) | (st.lo >> (64 - need));
+                        st.lo <<= need;
+                        st.nb -= int(need);
+                    } else { /* m_eff >= 51: rare */
                         consume(shift);
                         while (st.nb < int(m_eff)) topup();
                         sb = (m_eff == 64) ? st.hi
