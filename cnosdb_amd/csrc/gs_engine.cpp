@@ -615,6 +615,8 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
         auto fd = fdesc[wv];
         const int f_idx = lane & (GS_RING - 1);
         const int f_sq = lane / GS_RING;
+        double *rcur = &rslot[0][lane]; /* strength-reduced ring
+                                           cursor (row stride 65*8 B) */
         auto flush = [&]() {
             fd[lane][0] = (uint64_t)(uintptr_t)(o + (int64_t(r) - rfill));
             fd[lane][1] = uint64_t(rfill);
@@ -635,9 +637,8 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
                         ((double *)(uintptr_t)ob[t])[f_idx] = vbuf[t];
             }
             rfill = 0;
+            rcur = &rslot[0][lane];
         };
-        double *rcur = &rslot[0][lane]; /* strength-reduced ring cursor
-                                            (row stride 65*8 B) */
         if (!done && ch.row0 == 0 && r < end) { /* header value = row 0 */
             *rcur = __longlong_as_double((long long)st.val);
             rcur += 65;
@@ -2455,6 +2456,7 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
            stores, the earlier register-batching lesson) */
         const int f_idx = lane & (GS_RING - 1);
         const int f_sq = lane / GS_RING;
+        double *rcur = &rslot[0][lane]; /* strength-reduced ring cursor */
         auto flush = [&]() {
             fd[lane][0] = (uint64_t)(uintptr_t)(o + int64_t(run0));
             fd[lane][1] = uint64_t(rfill);
@@ -2475,11 +2477,13 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
                         ((double *)(uintptr_t)ob[t])[f_idx] = vbuf[t];
             }
             rfill = 0;
+            rcur = &rslot[0][lane];
         };
         auto stage_row = [&](uint64_t bits_) {
             if (r >= sel_lo && r < sel_hi) {
                 if (rfill == 0) run0 = r;
-                rslot[rfill][lane] = __longlong_as_double((long long)bits_);
+                *rcur = __longlong_as_double((long long)bits_);
+                rcur += 65;
                 rfill++;
             }
             r++;
