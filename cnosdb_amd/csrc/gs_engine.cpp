@@ -396,13 +396,18 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
                            const DevPage *__restrict__ pages, int npages,
                            const int32_t *__restrict__ chunk_base,
                            DevGorChunk *__restrict__ chunks,
-                           uint32_t chunk_rows) {
+                           uint32_t chunk_rows, int il_stride) {
     for (int pi = blockIdx.x * blockDim.x + threadIdx.x; pi < npages;
          pi += gridDim.x * blockDim.x) {
         DevPage pg = pages[pi];
         int cb = chunk_base[pi];
         int nch = chunk_base[pi + 1] - cb;
         if (nch <= 1) continue; /* single chunk: decode reads the header */
+        /* chunk k of page pi lives at cb+k (page-major) or
+           k*il_stride+pi (interleaved) */
+        auto cidx = [&](int k) {
+            return il_stride ? k * il_stride + pi : cb + k;
+        };
         const uint8_t *data = blob + pg.data_off;
         int next_k = 1;
         bool ok = false;
@@ -442,14 +447,14 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
                 if (r == uint32_t(next_k) * chunk_rows) {
                     int64_t used = int64_t(p - stream) * 8 - 128 - nb;
                     if (used > total_bits) { ok = false; break; }
-                    DevGorChunk &c = chunks[cb + next_k];
+                    DevGorChunk &c = chunks[cidx(next_k)];
                     c.bitpos = uint64_t(used);
                     c.val = val;
                     c.trailing = uint8_t(trailing);
                     c.meaningful = uint8_t(meaningful);
                     /* the previous chunk's whole range is now known
                        decodable: the filtered kernel may stop it early */
-                    chunks[cb + next_k - 1].flags = GORF_SAFE_STOP;
+                    chunks[cidx(next_k - 1)].flags = GORF_SAFE_STOP;
                     next_k++;
                     if (next_k == nch) break; /* tail chunk parses itself */
                 }
@@ -489,7 +494,7 @@ __global__ void k_gor_sync(const uint8_t *__restrict__ blob,
         }
         if (!ok) { /* poison unrecorded chunks -> DERR_SHORT at decode */
             for (int k = next_k; k < nch; k++) {
-                DevGorChunk &c = chunks[cb + k];
+                DevGorChunk &c = chunks[cidx(k)];
                 c.bitpos = uint64_t(pg.data_len) * 8;
                 c.val = 0;
                 c.trailing = 0;
@@ -589,6 +594,7 @@ __global__ void k_gor_chunks(const uint8_t *__restrict__ blob,
         double *o = out + ch.row_off;
         uint32_t r = ch.row0;
         uint32_t end = ch.row0 + ch.cnt;
+        if (ch.cnt == 0) have = false; /* interleave-padding entry */
         GorChunkState st = gor_chunk_init(blob, ch);
         int rfill = 0;
         bool done = !have, over = false;
@@ -3270,6 +3276,34 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                 acc += int32_t(nch);
             }
             cbase[sp.n[pc]] = acc;
+            /* layout experiment (GS_GOR_ORDER=interleave, PC_GOR only):
+               chunk-position-major table — a wave's 64 lanes then write
+               64 DIFFERENT series' output regions instead of one series'
+               contiguous region (DRAM channel spread).  Ragged pages are
+               padded with cnt=0 entries the decode kernels skip. */
+            static int il = [] {
+                const char *e = getenv("GS_GOR_ORDER");
+                return e && strcmp(e, "interleave") == 0;
+            }();
+            int il_stride = 0;
+            if (il && gcls == 0 && sp.n[pc] > 0) {
+                uint32_t maxc = 0;
+                for (int i = 0; i < sp.n[pc]; i++)
+                    maxc = uint32_t(cbase[i + 1] - cbase[i]) > maxc
+                               ? uint32_t(cbase[i + 1] - cbase[i]) : maxc;
+                if (maxc > 1) {
+                    il_stride = sp.n[pc];
+                    std::vector<DevGorChunk> h2(size_t(maxc) * sp.n[pc]);
+                    for (auto &c : h2) { c = DevGorChunk{}; c.cnt = 0; }
+                    for (int i = 0; i < sp.n[pc]; i++) {
+                        int nchi = cbase[i + 1] - cbase[i];
+                        for (int k = 0; k < nchi; k++)
+                            h2[size_t(k) * sp.n[pc] + i] = hch[cbase[i] + k];
+                    }
+                    hch.swap(h2);
+                    acc = int32_t(hch.size());
+                }
+            }
             DevGorChunk **dchp = gcls == 0 ? &sp.d_gor_chunks
                                            : &sp.d_gorn_chunks;
             int32_t **dcbp = gcls == 0 ? &sp.d_gor_chunk_base
@@ -3296,7 +3330,8 @@ GsGroupSet *gs_groups_upload(GsCtx *ctx, const GsColumnGroupDesc *groups,
                 hipLaunchKernelGGL(k_gor_sync,
                                    dim3(grid_for(sp.n[pc], 256)), dim3(256),
                                    0, ctx->stream, set->d_blob, sp.dev[pc],
-                                   sp.n[pc], *dcbp, *dchp, chunk_rows);
+                                   sp.n[pc], *dcbp, *dchp, chunk_rows,
+                                   il_stride);
             else
                 hipLaunchKernelGGL(k_gor_sync_null,
                                    dim3(grid_for(sp.n[pc], 256)), dim3(256),
