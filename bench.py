@@ -868,36 +868,46 @@ def main():
             raw_f64_bytes_sb0 = int(lens[m].sum()) // args.fields
     del sub
     sb_rows = sets[0].rows
+    nf = args.fields
     douts = [(torch.zeros(sb_rows, dtype=torch.int64, device=device),
-              torch.zeros(sb_rows, dtype=torch.float64, device=device))
+              torch.zeros(nf * sb_rows, dtype=torch.float64, device=device))
              for _ in range(2)]
     aggs = []
     for _ in range(2):
         aggs.append(dict(
             bucket_ns=BUCKET_NS, t0=T0, n_buckets=nbuckets,
-            d_max=torch.full((nbuckets,), -np.inf, dtype=torch.float64,
+            d_max=torch.full((nf * nbuckets,), -np.inf, dtype=torch.float64,
                              device=device),
-            d_sum=torch.zeros(nbuckets, dtype=torch.float64, device=device),
-            d_count=torch.zeros(nbuckets, dtype=torch.int64, device=device)))
-    d_sum = torch.zeros(nbuckets, dtype=torch.float64, device=device)
-    d_cnt = torch.zeros(nbuckets, dtype=torch.int64, device=device)
-    d_max = torch.zeros(nbuckets, dtype=torch.float64, device=device)
+            d_sum=torch.zeros(nf * nbuckets, dtype=torch.float64,
+                              device=device),
+            d_count=torch.zeros(nf * nbuckets, dtype=torch.int64,
+                                device=device)))
+    d_sum = torch.zeros(nf * nbuckets, dtype=torch.float64, device=device)
+    d_cnt = torch.zeros(nf * nbuckets, dtype=torch.int64, device=device)
+    d_max = torch.zeros(nf * nbuckets, dtype=torch.float64, device=device)
     setup_s = time.perf_counter() - t_setup
+
+    fields_list = list(range(nf))
+    dummy = [(torch.zeros(1, dtype=torch.int64, device=device),
+              torch.zeros(1, dtype=torch.float64, device=device))
+             for _ in range(2)]
 
     def step():
         out_rows = 0
         phase_ms = np.zeros(5)
-        for f in range(args.fields):
-            for i, st_ in enumerate(sets):
-                e = eng_of[i]
-                e.scan_async(st_, douts[i % 2][0], douts[i % 2][1],
-                             time_range=(lo, hi), agg=aggs[i % 2],
-                             field_col=f)
-            for i, st_ in enumerate(sets):
-                r = eng_of[i].scan_wait(st_)
-                out_rows += r.out_rows
-                phase_ms += [r.ms_decode_ts, r.ms_decode_val, r.ms_filter,
-                             r.ms_compact, r.ms_agg]
+        # one span/ts pass + all fields per set (gs_scan_fields): the
+        # round-1 per-field loop paid the span search + 8 B/row ts
+        # generation once per field
+        for i, st_ in enumerate(sets):
+            gs.scan_fields_async(eng_of[i], st_, fields_list,
+                                 dummy[i % 2][0], dummy[i % 2][1],
+                                 (lo, hi), douts[i % 2][0],
+                                 douts[i % 2][1], aggs[i % 2])
+        for i, st_ in enumerate(sets):
+            r = eng_of[i].scan_wait(st_)
+            out_rows += r.out_rows * nf
+            phase_ms += [r.ms_decode_ts, r.ms_decode_val, r.ms_filter,
+                         r.ms_compact, r.ms_agg]
         # combine the per-stream bucket partials (tiny)
         torch.add(aggs[0]["d_sum"], aggs[1]["d_sum"], out=d_sum)
         torch.add(aggs[0]["d_count"], aggs[1]["d_count"], out=d_cnt)

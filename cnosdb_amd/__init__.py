@@ -24,6 +24,8 @@ from .host import (
     page_of,
     prune_column_groups,
     groupby_tag,
+    scan_fields,
+    scan_fields_async,
     CT_TIME,
     CT_I64,
     CT_F64,
@@ -35,7 +37,7 @@ from .host import (
 __all__ = [
     "Engine", "GroupSet", "PageLib", "lib_path",
     "encode_ts", "encode_i64", "encode_f64", "encode_bool", "encode_str",
-    "build_page", "page_of", "str_page_of", "prune_column_groups", "groupby_tag",
+    "build_page", "page_of", "str_page_of", "prune_column_groups", "groupby_tag", "scan_fields", "scan_fields_async",
     "CT_TIME", "CT_I64", "CT_F64", "CT_BOOL", "CT_U64", "CT_STR",
 ]
 __version__ = "0.1"

@@ -3920,17 +3920,34 @@ GsStatus gs_apply_tombstone(GsCtx *ctx, GsGroupSet *set, const int64_t *d_ts,
     return GS_OK;
 }
 
-static bool fused_capable(GsGroupSet *set, const GsScanSpec *spec) {
-    uint32_t fc = 1 + uint32_t(spec->field_col);
-    return spec->d_out_ts && spec->d_out_val && spec->n_tombstones == 0 &&
-           spec->value_pred.op == GS_PRED_NONE &&
-           !set->any_nulls_field && fc < set->ncols &&
-           set->slots[0].n[PC_RLE_TS] == int(set->ngroups) &&
-           set->slots[fc].n[PC_GOR] == int(set->ngroups);
+static bool fused_capable_fields(GsGroupSet *set, const GsScanSpec *spec,
+                                 const int32_t *fields, int nf) {
+    if (!(spec->d_out_ts && spec->d_out_val && spec->n_tombstones == 0 &&
+          spec->value_pred.op == GS_PRED_NONE && !set->any_nulls_field &&
+          set->slots[0].n[PC_RLE_TS] == int(set->ngroups)))
+        return false;
+    for (int f = 0; f < nf; f++) {
+        uint32_t fc = 1 + uint32_t(fields[f]);
+        if (fc >= set->ncols ||
+            set->slots[fc].n[PC_GOR] != int(set->ngroups))
+            return false;
+    }
+    return true;
 }
 
+static bool fused_capable(GsGroupSet *set, const GsScanSpec *spec) {
+    int32_t f0 = spec->field_col;
+    return fused_capable_fields(set, spec, &f0, 1);
+}
+
+/* Fused scan over ONE span/ts pass and nf field columns (TSBS
+ * cpu-max-all-8, BASELINE config #3: the per-field loop of round 1 paid
+ * the span search and the 8 B/row ts generation once PER FIELD).
+ * Per-field outputs are strided: field i's compacted values go to
+ * d_out_val + i*total_rows; its aggregates to d_agg_* + i*n_buckets. */
 static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
-                                  const GsScanSpec *spec) {
+                                  const GsScanSpec *spec,
+                                  const int32_t *fields, int nf) {
     HIP_TRY(hipSetDevice(ctx->device));
     if (!set->sev_init) {
         for (int k = 0; k < 6; k++) HIP_TRY(hipEventCreate(&set->sev[k]));
@@ -3939,17 +3956,20 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
     hipEvent_t *ev = set->sev;
     int ng = int(set->ngroups);
     const DevPage *ts_pages = set->slots[0].dev[PC_RLE_TS];
-    SlotPages &fsp = set->slots[1 + spec->field_col];
-    int nch = fsp.n_gor_chunks;
+    int nch_max = 0;
+    for (int f = 0; f < nf; f++) {
+        int nc = set->slots[1 + fields[f]].n_gor_chunks;
+        nch_max = nc > nch_max ? nc : nch_max;
+    }
     if (!set->d_gor_nactive &&
         hipMalloc(&set->d_gor_nactive, sizeof(int)) != hipSuccess)
         return fail(GS_ERR, "hipMalloc active counter failed");
-    if (set->gor_active_cap < size_t(nch)) {
+    if (set->gor_active_cap < size_t(nch_max)) {
         if (set->d_gor_active) hipFree(set->d_gor_active);
-        if (hipMalloc(&set->d_gor_active, size_t(nch) * sizeof(int)) !=
+        if (hipMalloc(&set->d_gor_active, size_t(nch_max) * sizeof(int)) !=
             hipSuccess)
             return fail(GS_ERR, "hipMalloc active list failed");
-        set->gor_active_cap = size_t(nch);
+        set->gor_active_cap = size_t(nch_max);
     }
     int nblocks = (ng + SCAN_BLOCK * SCAN_ITEMS - 1) /
                   (SCAN_BLOCK * SCAN_ITEMS);
@@ -3975,22 +3995,10 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                        set->d_sp_start, set->d_sp_cnt, set->d_out_off,
                        spec->d_out_ts);
     HIP_TRY(hipEventRecord(ev[2], ctx->stream));
-    HIP_TRY(hipMemsetAsync(set->d_gor_nactive, 0, sizeof(int), ctx->stream));
-    hipLaunchKernelGGL(k_gor_active, dim3(grid_for(nch, 256)), dim3(256), 0,
-                       ctx->stream, fsp.d_gor_chunks, nch, set->d_sp_start,
-                       set->d_sp_cnt, set->d_gor_active, set->d_gor_nactive);
-    hipLaunchKernelGGL(k_gor_chunks_filtered,
-                       dim3(grid_for(nch, GS_GOR_BLOCK)), dim3(GS_GOR_BLOCK),
-                       0, ctx->stream, set->d_blob, fsp.d_gor_chunks,
-                       set->d_gor_active, set->d_gor_nactive, set->d_sp_start,
-                       set->d_sp_cnt, set->d_out_off, spec->d_out_val,
-                       ctx->d_err);
-    HIP_TRY(hipEventRecord(ev[3], ctx->stream));
-    HIP_TRY(hipEventRecord(ev[4], ctx->stream));
+    int nsg = set->nsgroups;
     if (spec->n_buckets > 0) {
         if (!spec->d_agg_max || !spec->d_agg_sum || !spec->d_agg_count)
             return fail(GS_ERR, "agg outputs missing");
-        int nsg = set->nsgroups;
         hipLaunchKernelGGL(k_build_sgroups_out, dim3(grid_for(nsg, 256)),
                            dim3(256), 0, ctx->stream, set->d_sgroup_first,
                            nsg, set->d_out_off, set->d_sgroups_out);
@@ -4005,38 +4013,65 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                 return fail(GS_ERR, "hipMalloc agg partials failed");
             set->partials_cap = cells;
         }
-        /* closed-form boundaries via the RLE group table when it fits in
-           LDS (it always does for realistic pages-per-series); otherwise
-           the out_ts binary-search kernel */
-        size_t shm_rle = size_t(set->max_span) * 16 +
-                         (size_t(set->max_span) + size_t(spec->n_buckets) + 2)
-                             * 4;
-        if (spec->n_buckets <= 8192 && shm_rle <= 64 * 1024) {
-            hipLaunchKernelGGL(k_agg_partial_rle,
-                               dim3(nsg > 65535 ? 65535 : nsg), dim3(256),
-                               shm_rle, ctx->stream, set->d_sgroups_out, nsg,
-                               set->d_sgroup_first, set->d_g_t0sel,
-                               set->d_g_delta, set->d_out_off,
-                               spec->d_out_val, spec->t0, spec->bucket_ns,
-                               spec->n_buckets, set->max_span, set->d_pmax,
-                               set->d_psum, set->d_pcnt);
-        } else {
-            size_t agg_shm = spec->n_buckets <= 8192
-                                 ? (size_t(spec->n_buckets) + 1) * 4 : 0;
-            hipLaunchKernelGGL(k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
-                               dim3(256), agg_shm, ctx->stream,
-                               set->d_sgroups_out, nsg,
-                               spec->d_out_ts, spec->d_out_val, nullptr,
-                               INT64_MIN, INT64_MAX, spec->t0, spec->bucket_ns,
-                               spec->n_buckets, set->d_pmax, set->d_psum,
-                               set->d_pcnt);
-        }
-        int mb = (spec->n_buckets + 3) / 4;
-        hipLaunchKernelGGL(k_agg_merge, dim3(mb > 2048 ? 2048 : mb), dim3(256),
-                           0, ctx->stream, nsg, spec->n_buckets, set->d_pmax,
-                           set->d_psum, set->d_pcnt, spec->d_agg_max,
-                           spec->d_agg_sum, spec->d_agg_count);
     }
+    for (int f = 0; f < nf; f++) {
+        SlotPages &fsp = set->slots[1 + fields[f]];
+        int nch = fsp.n_gor_chunks;
+        double *d_val_f = spec->d_out_val + size_t(f) * set->total_rows;
+        HIP_TRY(hipMemsetAsync(set->d_gor_nactive, 0, sizeof(int),
+                               ctx->stream));
+        hipLaunchKernelGGL(k_gor_active, dim3(grid_for(nch, 256)), dim3(256),
+                           0, ctx->stream, fsp.d_gor_chunks, nch,
+                           set->d_sp_start, set->d_sp_cnt, set->d_gor_active,
+                           set->d_gor_nactive);
+        hipLaunchKernelGGL(k_gor_chunks_filtered,
+                           dim3(grid_for(nch, GS_GOR_BLOCK)),
+                           dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
+                           fsp.d_gor_chunks, set->d_gor_active,
+                           set->d_gor_nactive, set->d_sp_start,
+                           set->d_sp_cnt, set->d_out_off, d_val_f,
+                           ctx->d_err);
+        if (f == 0) HIP_TRY(hipEventRecord(ev[3], ctx->stream));
+        if (spec->n_buckets > 0) {
+            /* closed-form boundaries via the RLE group table when it
+               fits in LDS (it always does for realistic pages-per-
+               series); otherwise the out_ts binary-search kernel */
+            size_t shm_rle =
+                size_t(set->max_span) * 16 +
+                (size_t(set->max_span) + size_t(spec->n_buckets) + 2) * 4;
+            if (spec->n_buckets <= 8192 && shm_rle <= 64 * 1024) {
+                hipLaunchKernelGGL(k_agg_partial_rle,
+                                   dim3(nsg > 65535 ? 65535 : nsg),
+                                   dim3(256), shm_rle, ctx->stream,
+                                   set->d_sgroups_out, nsg,
+                                   set->d_sgroup_first, set->d_g_t0sel,
+                                   set->d_g_delta, set->d_out_off, d_val_f,
+                                   spec->t0, spec->bucket_ns,
+                                   spec->n_buckets, set->max_span,
+                                   set->d_pmax, set->d_psum, set->d_pcnt);
+            } else {
+                size_t agg_shm = spec->n_buckets <= 8192
+                                     ? (size_t(spec->n_buckets) + 1) * 4
+                                     : 0;
+                hipLaunchKernelGGL(
+                    k_agg_partial, dim3(nsg > 65535 ? 65535 : nsg),
+                    dim3(256), agg_shm, ctx->stream, set->d_sgroups_out,
+                    nsg, spec->d_out_ts, d_val_f, nullptr, INT64_MIN,
+                    INT64_MAX, spec->t0, spec->bucket_ns, spec->n_buckets,
+                    set->d_pmax, set->d_psum, set->d_pcnt);
+            }
+            int mb = (spec->n_buckets + 3) / 4;
+            hipLaunchKernelGGL(k_agg_merge, dim3(mb > 2048 ? 2048 : mb),
+                               dim3(256), 0, ctx->stream, nsg,
+                               spec->n_buckets, set->d_pmax, set->d_psum,
+                               set->d_pcnt,
+                               spec->d_agg_max + size_t(f) * spec->n_buckets,
+                               spec->d_agg_sum + size_t(f) * spec->n_buckets,
+                               spec->d_agg_count +
+                                   size_t(f) * spec->n_buckets);
+        }
+    }
+    HIP_TRY(hipEventRecord(ev[4], ctx->stream));
     HIP_TRY(hipEventRecord(ev[5], ctx->stream));
     set->pending = true;
     return GS_OK;
@@ -4060,10 +4095,10 @@ static GsStatus scan_fused_wait(GsCtx *ctx, GsGroupSet *set,
     HIP_TRY(hipEventElapsedTime(&ms, ev[1], ev[2]));
     result->ms_decode_ts = ms;
     HIP_TRY(hipEventElapsedTime(&ms, ev[2], ev[3]));
-    result->ms_decode_val = ms;
+    result->ms_decode_val = ms; /* nf>1: field 0 only */
     result->ms_compact = 0.0;
-    HIP_TRY(hipEventElapsedTime(&ms, ev[4], ev[5]));
-    result->ms_agg = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, ev[3], ev[4]));
+    result->ms_agg = ms; /* nf>1: field 0's agg + remaining fields */
     result->out_rows = acc;
     result->decoded_rows = set->total_rows;
     return GS_OK;
@@ -4076,7 +4111,8 @@ GsStatus gs_scan_async(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec) {
         return fail(GS_ERR, "gs_scan_async requires the fused-capable shape "
                             "(RLE ts + all-valid Gorilla, no tombstones, "
                             "compacted outputs)");
-    return scan_fused_launch(ctx, set, spec);
+    int32_t f0 = spec->field_col;
+    return scan_fused_launch(ctx, set, spec, &f0, 1);
 }
 
 GsStatus gs_scan_wait(GsCtx *ctx, GsGroupSet *set, GsScanResult *result) {
@@ -4154,6 +4190,36 @@ GsGroupSet *gs_raw_set(GsCtx *ctx, const int64_t *counts, int64_t nseries) {
     return set;
 }
 
+/* Fused scan of nf field columns over ONE span/ts pass (TSBS
+ * cpu-max-all-8, BASELINE config #3; the reference decodes each field
+ * column of the column group in the same decode_pages pass,
+ * tsm/reader.rs:494-560).  Fused-capable shapes only.  Strides:
+ * d_out_val + f*total_rows per field; d_agg_* + f*n_buckets per field. */
+GsStatus gs_scan_fields(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
+                        const int32_t *field_cols, int nf,
+                        GsScanResult *result) {
+    if (!ctx || !set || !spec || !result || !field_cols || nf <= 0)
+        return fail(GS_ERR, "bad args to gs_scan_fields");
+    if (!fused_capable_fields(set, spec, field_cols, nf))
+        return fail(GS_ERR, "gs_scan_fields needs a fused-capable shape "
+                            "(all-RLE ts, all-valid Gorilla fields, no "
+                            "tombstones/predicates, compacted outputs)");
+    GsStatus st = scan_fused_launch(ctx, set, spec, field_cols, nf);
+    if (st != GS_OK) return st;
+    return scan_fused_wait(ctx, set, result);
+}
+
+/* async variant: pair with gs_scan_wait */
+GsStatus gs_scan_fields_async(GsCtx *ctx, GsGroupSet *set,
+                              const GsScanSpec *spec,
+                              const int32_t *field_cols, int nf) {
+    if (!ctx || !set || !spec || !field_cols || nf <= 0)
+        return fail(GS_ERR, "bad args to gs_scan_fields_async");
+    if (!fused_capable_fields(set, spec, field_cols, nf))
+        return fail(GS_ERR, "gs_scan_fields needs a fused-capable shape");
+    return scan_fused_launch(ctx, set, spec, field_cols, nf);
+}
+
 GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
                  GsScanResult *result) {
     if (!ctx || !set || !spec || !result || !spec->d_ts || !spec->d_val)
@@ -4183,7 +4249,8 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
      * every field page all-valid Gorilla, no tombstones, compacted
      * outputs requested.  Falls back to the general path otherwise. */
     if (fused_capable(set, spec)) {
-        GsStatus fst = scan_fused_launch(ctx, set, spec);
+        int32_t f0 = spec->field_col;
+        GsStatus fst = scan_fused_launch(ctx, set, spec, &f0, 1);
         if (fst != GS_OK) return fst;
         return scan_fused_wait(ctx, set, result);
     }

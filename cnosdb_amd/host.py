@@ -698,3 +698,80 @@ def groupby_tag(engine, gset, n_buckets, d_max, d_sum, d_count,
     if st != 0:
         raise RuntimeError(f"gs_groupby_tag failed ({st}): {engine._pl.err()}")
     return ng.value, rep[:ng.value]
+
+
+def _bind_scan_fields(lib):
+    import ctypes
+    if getattr(lib, "_sf_bound", False):
+        return
+    lib.gs_scan_fields.restype = ctypes.c_int
+    lib.gs_scan_fields.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                   ctypes.c_void_p, ctypes.c_void_p,
+                                   ctypes.c_int, ctypes.c_void_p]
+    lib._sf_bound = True
+
+
+def scan_fields(engine, gset, fields, d_ts, d_val, time_range,
+                d_out_ts, d_out_val, agg=None):
+    """Fused multi-field scan (gs_scan_fields): one span/ts pass, then
+    decode+aggregate per field.  d_out_val must hold len(fields)*rows
+    doubles (field i at offset i*rows); agg buffers len(fields)*n_buckets."""
+    import ctypes
+    _bind_scan_fields(engine.lib)
+    spec = GsScanSpec()
+    spec.field_col = fields[0]
+    lo, hi = time_range
+    spec.range = GsTimeRange(lo, hi)
+    spec.d_ts = d_ts.data_ptr()
+    spec.d_val = d_val.data_ptr()
+    spec.d_out_ts = d_out_ts.data_ptr()
+    spec.d_out_val = d_out_val.data_ptr()
+    if agg:
+        spec.bucket_ns = agg["bucket_ns"]
+        spec.t0 = agg["t0"]
+        spec.n_buckets = agg["n_buckets"]
+        spec.d_agg_max = agg["d_max"].data_ptr()
+        spec.d_agg_sum = agg["d_sum"].data_ptr()
+        spec.d_agg_count = agg["d_count"].data_ptr()
+    fc = (ctypes.c_int32 * len(fields))(*fields)
+    res = GsScanResult()
+    st = engine.lib.gs_scan_fields(engine._ctx, gset._h, ctypes.byref(spec),
+                                   fc, len(fields), ctypes.byref(res))
+    if st != 0:
+        raise RuntimeError(
+            f"gs_scan_fields failed ({st}): {engine._pl.err()}")
+    return res
+
+
+def scan_fields_async(engine, gset, fields, d_ts, d_val, time_range,
+                      d_out_ts, d_out_val, agg=None):
+    """Async gs_scan_fields; pair with engine.scan_wait(gset)."""
+    import ctypes
+    _bind_scan_fields(engine.lib)
+    if not getattr(engine.lib, "_sfa_bound", False):
+        engine.lib.gs_scan_fields_async.restype = ctypes.c_int
+        engine.lib.gs_scan_fields_async.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_void_p, ctypes.c_int]
+        engine.lib._sfa_bound = True
+    spec = GsScanSpec()
+    spec.field_col = fields[0]
+    lo, hi = time_range
+    spec.range = GsTimeRange(lo, hi)
+    spec.d_ts = d_ts.data_ptr()
+    spec.d_val = d_val.data_ptr()
+    spec.d_out_ts = d_out_ts.data_ptr()
+    spec.d_out_val = d_out_val.data_ptr()
+    if agg:
+        spec.bucket_ns = agg["bucket_ns"]
+        spec.t0 = agg["t0"]
+        spec.n_buckets = agg["n_buckets"]
+        spec.d_agg_max = agg["d_max"].data_ptr()
+        spec.d_agg_sum = agg["d_sum"].data_ptr()
+        spec.d_agg_count = agg["d_count"].data_ptr()
+    fc = (ctypes.c_int32 * len(fields))(*fields)
+    st = engine.lib.gs_scan_fields_async(engine._ctx, gset._h,
+                                         ctypes.byref(spec), fc, len(fields))
+    if st != 0:
+        raise RuntimeError(
+            f"gs_scan_fields_async failed ({st}): {engine._pl.err()}")

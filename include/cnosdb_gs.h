@@ -282,6 +282,19 @@ GsStatus gs_groupby_tag(GsCtx *ctx, GsGroupSet *set, int tag_col,
                         long long *d_out_count, int64_t *tag_rep_row,
                         int cap_gids, int *out_ngids);
 
+/* Fused scan of nf field columns over ONE span/ts pass (TSBS
+ * cpu-max-all-8, BASELINE config #3; the reference decodes every
+ * projected field column of the column group in one decode_pages pass,
+ * tsm/reader.rs:494-560).  Fused-capable shapes only.  Output strides:
+ * d_out_val + f*total_rows per field; d_agg_* + f*n_buckets per field. */
+GsStatus gs_scan_fields(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
+                        const int32_t *field_cols, int nf,
+                        GsScanResult *result);
+/* async variant: pair with gs_scan_wait */
+GsStatus gs_scan_fields_async(GsCtx *ctx, GsGroupSet *set,
+                              const GsScanSpec *spec,
+                              const int32_t *field_cols, int nf);
+
 GsStatus gs_scan_async(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec);
 GsStatus gs_scan_wait(GsCtx *ctx, GsGroupSet *set, GsScanResult *result);
 

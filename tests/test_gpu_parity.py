@@ -764,3 +764,66 @@ def test_truncated_string_payload_errors(engine):
     with pytest.raises(RuntimeError):
         engine.decode_str(gset, 1, d_off, d_bytes)
     gset.free()
+
+
+def test_scan_fields_equals_per_field(engine):
+    """gs_scan_fields (one span/ts pass, N fields) must equal N separate
+    fused scans — compacted values and aggregates, bit-exact."""
+    nseries, npts, nf = 10, 8192, 3
+    r = np.random.default_rng(21)
+    t0 = 1_700_000_000_000_000_000
+    groups = []
+    per_field_vals = [[] for _ in range(nf)]
+    for s in range(nseries):
+        ts = t0 + np.arange(npts, dtype=np.int64) * 1_000_000_000
+        cols = [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME)]
+        for f in range(nf):
+            v = np.round(np.clip(np.cumsum(r.normal(0, 0.5, npts)) + 50,
+                                 0, 100), 1)
+            cols.append((gs.page_of(v, gs.CT_F64), gs.CT_F64))
+            per_field_vals[f].append(v)
+        groups.append((s, cols))
+    gset = engine.upload(groups)
+    rows = gset.rows
+    lo = t0 + 1000 * 10**9
+    hi = t0 + 7000 * 10**9
+    bucket_ns = 300_000_000_000
+    nb = int(npts * 10**9 // bucket_ns) + 1
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    # multi-field call
+    d_oval_m = torch.zeros(nf * rows, dtype=torch.float64, device="cuda")
+    agg_m = dict(bucket_ns=bucket_ns, t0=t0, n_buckets=nb,
+                 d_max=torch.full((nf * nb,), -np.inf, dtype=torch.float64,
+                                  device="cuda"),
+                 d_sum=torch.zeros(nf * nb, dtype=torch.float64,
+                                   device="cuda"),
+                 d_count=torch.zeros(nf * nb, dtype=torch.int64,
+                                     device="cuda"))
+    res_m = gs.scan_fields(engine, gset, list(range(nf)), d_ts, d_val,
+                           (lo, hi), d_ots, d_oval_m, agg_m)
+    # per-field reference calls
+    for f in range(nf):
+        d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+        d_ots2 = torch.zeros(rows, dtype=torch.int64, device="cuda")
+        agg = dict(bucket_ns=bucket_ns, t0=t0, n_buckets=nb,
+                   d_max=torch.full((nb,), -np.inf, dtype=torch.float64,
+                                    device="cuda"),
+                   d_sum=torch.zeros(nb, dtype=torch.float64, device="cuda"),
+                   d_count=torch.zeros(nb, dtype=torch.int64, device="cuda"))
+        res = engine.scan(gset, d_ts, d_val, time_range=(lo, hi),
+                          d_out_ts=d_ots2, d_out_val=d_oval, agg=agg,
+                          field_col=f)
+        assert res.out_rows == res_m.out_rows
+        n = res.out_rows
+        assert bool((d_ots[:n] == d_ots2[:n]).all())
+        assert bool((d_oval_m[f * rows:f * rows + n] == d_oval[:n]).all())
+        assert bool((agg_m["d_count"][f * nb:(f + 1) * nb] ==
+                     agg["d_count"]).all())
+        nz = agg["d_count"] > 0
+        assert bool((agg_m["d_max"][f * nb:(f + 1) * nb][nz] ==
+                     agg["d_max"][nz]).all())
+        assert bool(torch.allclose(agg_m["d_sum"][f * nb:(f + 1) * nb],
+                                   agg["d_sum"], rtol=1e-12))
+    gset.free()
