@@ -961,10 +961,8 @@ def main():
     # moved).  The unfused number (8 B x all rows) is stated in DESIGN.md.
     raw_f64_bytes = raw_f64_bytes_sb0 - \
         (16 + (page_rows + 7) // 8) * (sb_rows // page_rows)
-    sel_rows_sb = int(out_rows) // args.sub_batches // args.fields
-    # the fused kernel now also writes the closed-form ts alongside each
-    # selected value (16 B/selected row out total)
-    alg_bytes_launch = raw_f64_bytes + 16 * sel_rows_sb
+    sel_rows_sb = int(out_rows) // args.sub_batches
+    alg_bytes_launch = raw_f64_bytes + 8 * sel_rows_sb
     ms_gorilla_launch = r_cal.ms_decode_val
     achieved = alg_bytes_launch / (ms_gorilla_launch / 1000) if ms_gorilla_launch > 0 else 0
     peak = 8.0e12

@@ -2401,20 +2401,13 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
                                       const int64_t *__restrict__ sp_start,
                                       const int64_t *__restrict__ sp_cnt,
                                       const int64_t *__restrict__ out_off,
-                                      const int64_t *__restrict__ g_t0sel,
-                                      const int64_t *__restrict__ g_delta,
-                                      int64_t ts_off_bytes, int write_ts,
                                       double *__restrict__ out,
                                       unsigned *__restrict__ err) {
     __shared__ double ring[GS_GOR_BLOCK / 64][GS_RING][64 + 1];
-    /* per-flush lane descriptors: {val dst pointer, staged count, ts of the
-       run's first row, ts delta} packed so the flush loop reads broadcast
-       ds_reads per source lane instead of cross-lane shuffles.  The
-       closed-form ts of the selected rows (t0sel + i*delta from
-       k_spans_rle's RLE header scan) is written HERE, fused into the value
-       flush: the gor kernel is execution-bound with spare store bandwidth,
-       so the separate 8 B/row ts kernel disappears from the fused path. */
-    __shared__ uint64_t fdesc[GS_GOR_BLOCK / 64][64][4];
+    /* per-flush lane descriptors: {dst pointer, staged count} packed so the
+       flush loop reads ONE broadcast ds_read per source lane instead of
+       three cross-lane shuffles */
+    __shared__ uint64_t fdesc[GS_GOR_BLOCK / 64][64][2];
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     auto rslot = ring[wv];
@@ -2434,10 +2427,6 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
         int32_t sel_lo = int32_t(sp_start[ch.grp]);
         int32_t sel_hi = int32_t(sel_lo + sp_cnt[ch.grp]);
         if (!have) { sel_lo = 0; sel_hi = 0; }
-        const int64_t ts_d = have ? g_delta[ch.grp] : 0;
-        /* ts of sel_lo minus sel_lo*delta: ts(r) = ts_base + r*delta */
-        const int64_t ts_base =
-            have ? g_t0sel[ch.grp] - int64_t(sel_lo) * ts_d : 0;
         double *o = out + out_off[ch.grp] - sel_lo; /* o[r] valid in span */
         int32_t r = int32_t(ch.row0);
         int32_t end = r + int32_t(ch.cnt);
@@ -2477,31 +2466,21 @@ __global__ void k_gor_chunks_filtered(const uint8_t *__restrict__ blob,
         auto flush = [&]() {
             fd[lane][0] = (uint64_t)(uintptr_t)(o + int64_t(run0));
             fd[lane][1] = uint64_t(rfill);
-            fd[lane][2] = uint64_t(ts_base + int64_t(run0) * ts_d);
-            fd[lane][3] = uint64_t(ts_d);
             __builtin_amdgcn_wave_barrier();
             constexpr int SRCP = 64 / GS_RING; /* sources per store */
             for (int src0 = 0; src0 < 64; src0 += SRCP * 4) {
                 double vbuf[4];
                 uint64_t ob[4];
                 int cnt[4];
-                uint64_t t0b[4], dlt[4];
                 for (int t = 0; t < 4; t++) {
                     int src = src0 + f_sq + t * SRCP;
                     vbuf[t] = rslot[f_idx][src];
                     ob[t] = fd[src][0];
                     cnt[t] = int(fd[src][1]);
-                    t0b[t] = fd[src][2];
-                    dlt[t] = fd[src][3];
                 }
                 for (int t = 0; t < 4; t++)
-                    if (f_idx < cnt[t]) {
+                    if (f_idx < cnt[t])
                         ((double *)(uintptr_t)ob[t])[f_idx] = vbuf[t];
-                        if (write_ts)
-                            ((int64_t *)(uintptr_t)(ob[t] +
-                                                    ts_off_bytes))[f_idx] =
-                                int64_t(t0b[t] + uint64_t(f_idx) * dlt[t]);
-                    }
             }
             rfill = 0;
             rcur = &rslot[0][lane];
@@ -4011,8 +3990,10 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
     hipLaunchKernelGGL(k_scan_add, dim3(grid_for(ng, 256)), dim3(256), 0,
                        ctx->stream, ng, set->d_out_off, set->d_blocksums);
     HIP_TRY(hipEventRecord(ev[1], ctx->stream));
-    /* ts generation is fused into field 0's value flush (closed-form from
-       the RLE group table) — no separate 8 B/row ts kernel */
+    hipLaunchKernelGGL(k_rle_ts_filtered, dim3(ng > 2048 ? 2048 : ng),
+                       dim3(256), 0, ctx->stream, set->d_blob, ts_pages, ng,
+                       set->d_sp_start, set->d_sp_cnt, set->d_out_off,
+                       spec->d_out_ts);
     HIP_TRY(hipEventRecord(ev[2], ctx->stream));
     int nsg = set->nsgroups;
     if (spec->n_buckets > 0) {
@@ -4043,16 +4024,13 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                            0, ctx->stream, fsp.d_gor_chunks, nch,
                            set->d_sp_start, set->d_sp_cnt, set->d_gor_active,
                            set->d_gor_nactive);
-        const int64_t ts_off_bytes =
-            (const char *)spec->d_out_ts - (const char *)d_val_f;
         hipLaunchKernelGGL(k_gor_chunks_filtered,
                            dim3(grid_for(nch, GS_GOR_BLOCK)),
                            dim3(GS_GOR_BLOCK), 0, ctx->stream, set->d_blob,
                            fsp.d_gor_chunks, set->d_gor_active,
                            set->d_gor_nactive, set->d_sp_start,
-                           set->d_sp_cnt, set->d_out_off, set->d_g_t0sel,
-                           set->d_g_delta, ts_off_bytes, f == 0 ? 1 : 0,
-                           d_val_f, ctx->d_err);
+                           set->d_sp_cnt, set->d_out_off, d_val_f,
+                           ctx->d_err);
         if (f == 0) HIP_TRY(hipEventRecord(ev[3], ctx->stream));
         if (spec->n_buckets > 0) {
             /* closed-form boundaries via the RLE group table when it
