@@ -1425,8 +1425,8 @@ __global__ void k_tombstone(const DevGroup *__restrict__ groups, int n,
 /* value-predicate mask over the selected span (DataFilter semantics,
  * reader/filter.rs:91-142): mask = validity AND pred(value); per-group
  * selected count by block reduction */
-__device__ __forceinline__ bool dev_pred(int op, double a, double b,
-                                         double x) {
+template <typename T>
+__device__ __forceinline__ bool dev_pred_t(int op, T a, T b, T x) {
     switch (op) {
     case GS_PRED_GT: return x > a;
     case GS_PRED_GE: return x >= a;
@@ -1439,12 +1439,26 @@ __device__ __forceinline__ bool dev_pred(int op, double a, double b,
     }
 }
 
+/* vt: 0 = f64, 1 = i64, 2 = u64 (bit-cast slot, unsigned.rs:20-45) — the
+ * reference's DataFilter evaluates the pushed expr in the column's own
+ * type (reader/filter.rs:91-142) */
+__device__ __forceinline__ bool dev_pred(int op, double a, double b,
+                                         double x, int vt) {
+    if (vt == 1)
+        return dev_pred_t<int64_t>(op, int64_t(a), int64_t(b),
+                                   __double_as_longlong(x));
+    if (vt == 2)
+        return dev_pred_t<uint64_t>(op, uint64_t(a), uint64_t(b),
+                                    uint64_t(__double_as_longlong(x)));
+    return dev_pred_t<double>(op, a, b, x);
+}
+
 __global__ void k_vmask(const DevGroup *__restrict__ groups, int n,
                         const double *__restrict__ val,
                         const uint8_t *__restrict__ valid,
                         const int64_t *__restrict__ sp_start,
                         const int64_t *__restrict__ sp_cnt, int op, double a,
-                        double b, uint8_t *__restrict__ mask,
+                        double b, int vt, uint8_t *__restrict__ mask,
                         int64_t *__restrict__ sel_cnt) {
     __shared__ long long sred[256];
     for (int g = blockIdx.x; g < n; g += gridDim.x) {
@@ -1453,7 +1467,7 @@ __global__ void k_vmask(const DevGroup *__restrict__ groups, int n,
         long long c = 0;
         for (int64_t r = threadIdx.x; r < cnt; r += blockDim.x) {
             bool ok = (!valid || valid[base + r]) &&
-                      dev_pred(op, a, b, val[base + r]);
+                      dev_pred(op, a, b, val[base + r], vt);
             mask[base + r] = ok;
             c += ok;
         }
@@ -4229,9 +4243,16 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
     if (!raw) {
         if (set->ncols < 2)
             return fail(GS_ERR, "gs_scan needs a time page + one f64 field page");
-        if (spec->field_col < 0 || 1 + uint32_t(spec->field_col) >= set->ncols ||
-            set->slots[1 + spec->field_col].ctype != GS_CT_F64)
-            return fail(GS_ERR, "gs_scan field column must be an f64 slot");
+        const uint32_t fcol = 1 + uint32_t(spec->field_col);
+        if (spec->field_col < 0 || fcol >= set->ncols)
+            return fail(GS_ERR, "gs_scan field column out of range");
+        const uint8_t fct2 = set->slots[fcol].ctype;
+        if (fct2 != GS_CT_F64 && fct2 != GS_CT_I64 && fct2 != GS_CT_U64)
+            return fail(GS_ERR,
+                        "gs_scan field column must be f64/i64/u64");
+        if (fct2 != GS_CT_F64 && spec->n_buckets > 0)
+            return fail(GS_ERR, "aggregates are f64-only (TSBS path); "
+                                "decode+filter+compact support i64/u64");
     }
     HIP_TRY(hipSetDevice(ctx->device));
 
@@ -4304,11 +4325,13 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
             HIP_TRY(hipMalloc(&set->d_sel_cnt,
                               set->ngroups * sizeof(int64_t)));
         d_mask = set->d_mask;
+        const uint8_t fct = set->slots[1 + spec->field_col].ctype;
+        const int vt = fct == GS_CT_I64 ? 1 : fct == GS_CT_U64 ? 2 : 0;
         hipLaunchKernelGGL(k_vmask, dim3(ng > 65535 ? 65535 : ng), dim3(256),
                            0, ctx->stream, set->d_groups, ng, spec->d_val,
                            d_valid, set->d_sp_start, set->d_sp_cnt,
                            spec->value_pred.op, spec->value_pred.a,
-                           spec->value_pred.b, d_mask, set->d_sel_cnt);
+                           spec->value_pred.b, vt, d_mask, set->d_sel_cnt);
         d_counts = set->d_sel_cnt;
     }
     HIP_TRY(hipEventRecord(ev[3], ctx->stream));

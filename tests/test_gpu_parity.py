@@ -827,3 +827,70 @@ def test_scan_fields_equals_per_field(engine):
         assert bool(torch.allclose(agg_m["d_sum"][f * nb:(f + 1) * nb],
                                    agg["d_sum"], rtol=1e-12))
     gset.free()
+
+
+def test_scan_i64_field_with_pred(engine):
+    """Integer fields flow through the general scan path: decode + time
+    filter + typed value predicate + compact (DataFilter evaluates the
+    pushed expr in the column's own type, reader/filter.rs:91-142).
+    Aggregates stay f64-only (TSBS path) and are not requested here."""
+    t0 = 1_700_000_000_000_000_000
+    r = np.random.default_rng(31)
+    groups, truth = [], []
+    for s in range(6):
+        n = 5000
+        ts = t0 + np.arange(n, dtype=np.int64) * 1_000_000_000
+        vals = r.integers(-1000, 1000, n).astype(np.int64)
+        groups.append((s, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                           (gs.page_of(vals, gs.CT_I64), gs.CT_I64)]))
+        truth.append((ts, vals))
+    gset = engine.upload(groups)
+    rows = gset.rows
+    lo = t0 + 500 * 10**9
+    hi = t0 + 4500 * 10**9
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    res = engine.scan(gset, d_ts, d_val, time_range=(lo, hi),
+                      d_out_ts=d_ots, d_out_val=d_oval,
+                      value_pred=("gt", 250.0))
+    exp_ts, exp_val = [], []
+    for ts, vals in truth:
+        sel = (ts >= lo) & (ts <= hi) & (vals > 250)
+        exp_ts.append(ts[sel])
+        exp_val.append(vals[sel])
+    exp_ts = np.concatenate(exp_ts)
+    exp_val = np.concatenate(exp_val)
+    assert res.out_rows == exp_ts.size
+    assert (d_ots[:res.out_rows].cpu().numpy() == exp_ts).all()
+    got = d_oval[:res.out_rows].cpu().numpy().view(np.int64)
+    assert (got == exp_val).all()
+    gset.free()
+
+
+def test_scan_u64_field_unsigned_compare(engine):
+    """u64 fields (bit-cast i64 slot, unsigned.rs:20-45): the predicate
+    must compare UNSIGNED — a value with the top bit set is larger than
+    any positive literal, not negative."""
+    t0 = 1_700_000_000_000_000_000
+    n = 1000
+    ts = t0 + np.arange(n, dtype=np.int64) * 1_000_000_000
+    vals = np.arange(n, dtype=np.uint64)
+    vals[::10] += np.uint64(2**63)  # huge unsigned values
+    gset = engine.upload([(0, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                               (gs.page_of(vals.view(np.int64), gs.CT_I64),
+                                gs.CT_U64)])])
+    rows = gset.rows
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    res = engine.scan(gset, d_ts, d_val, time_range=(int(ts[0]), int(ts[-1])),
+                      d_out_ts=d_ots, d_out_val=d_oval,
+                      value_pred=("gt", 500.0))
+    sel = vals > np.uint64(500)
+    assert res.out_rows == int(sel.sum())
+    got = d_oval[:res.out_rows].cpu().numpy().view(np.uint64)
+    assert (got == vals[sel]).all()
+    gset.free()
