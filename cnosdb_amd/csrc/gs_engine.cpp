@@ -4325,8 +4325,11 @@ GsStatus gs_scan(GsCtx *ctx, GsGroupSet *set, const GsScanSpec *spec,
             HIP_TRY(hipMalloc(&set->d_sel_cnt,
                               set->ngroups * sizeof(int64_t)));
         d_mask = set->d_mask;
-        const uint8_t fct = set->slots[1 + spec->field_col].ctype;
-        const int vt = fct == GS_CT_I64 ? 1 : fct == GS_CT_U64 ? 2 : 0;
+        int vt = 0; /* raw (memcache) sets have no column slots: f64 */
+        if (!raw) {
+            const uint8_t fct = set->slots[1 + spec->field_col].ctype;
+            vt = fct == GS_CT_I64 ? 1 : fct == GS_CT_U64 ? 2 : 0;
+        }
         hipLaunchKernelGGL(k_vmask, dim3(ng > 65535 ? 65535 : ng), dim3(256),
                            0, ctx->stream, set->d_groups, ng, spec->d_val,
                            d_valid, set->d_sp_start, set->d_sp_cnt,
