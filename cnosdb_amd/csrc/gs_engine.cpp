@@ -4054,9 +4054,14 @@ static GsStatus scan_fused_launch(GsCtx *ctx, GsGroupSet *set,
                 size_t(set->max_span) * 16 +
                 (size_t(set->max_span) + size_t(spec->n_buckets) + 2) * 4;
             if (spec->n_buckets <= 8192 && shm_rle <= 64 * 1024) {
+                static int agg_block = [] { /* sweepable (GS_AGG_BLOCK) */
+                    const char *e = getenv("GS_AGG_BLOCK");
+                    long v = e ? atol(e) : 0;
+                    return int(v == 128 || v == 512 ? v : 256);
+                }();
                 hipLaunchKernelGGL(k_agg_partial_rle,
                                    dim3(nsg > 65535 ? 65535 : nsg),
-                                   dim3(256), shm_rle, ctx->stream,
+                                   dim3(agg_block), shm_rle, ctx->stream,
                                    set->d_sgroups_out, nsg,
                                    set->d_sgroup_first, set->d_g_t0sel,
                                    set->d_g_delta, set->d_out_off, d_val_f,
