@@ -877,11 +877,15 @@ __global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
         const uint8_t *bs = blob + ch.bitset_off;
         double *o = out + ch.row_off;
         uint32_t r = ch.row0;
-        uint32_t bb = 0;
-        uint32_t bb_idx = 0xffffffffu; /* lazy per-8-row bitset byte (byte
-                                          loads are always aligned; the
-                                          8-B word variant misread flakily
-                                          at high blob offsets) */
+        /* per-8-row bitset byte, prefetched ONE AHEAD like the stream
+           refills: the unprefetched dependent byte load left the kernel
+           71% memory-parked (SQ_WAIT_ANY PMC).  Byte loads are always
+           aligned (the 8-B word variant misread flakily at high blob
+           offsets); the +1 read tops out one byte past the bitset into
+           the page's own data region. */
+        uint32_t bb_idx = ch.row0 >> 3;
+        uint32_t bb = bs[bb_idx];
+        uint32_t bb_next = bs[bb_idx + 1];
         uint32_t end = ch.row0 + ch.cnt;
         bool sent_seen = (ch.flags & GORF_SENT_SEEN) != 0;
         GorChunkState st;
@@ -984,7 +988,11 @@ __global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
             if (!done) {
                 if (r < end) {
                     const uint32_t bi = r >> 3;
-                    if (bi != bb_idx) { bb = bs[bi]; bb_idx = bi; }
+                    if (bi != bb_idx) {
+                        bb = bb_next;
+                        bb_idx = bi;
+                        bb_next = bs[bi + 1];
+                    }
                     const int bit = int((bb >> (r & 7)) & 1);
                     uint64_t sbits = 0;
                     bool hole = false; /* set row after the sentinel */
