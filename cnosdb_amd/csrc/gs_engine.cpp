@@ -877,12 +877,12 @@ __global__ void k_gor_chunks_null(const uint8_t *__restrict__ blob,
         const uint8_t *bs = blob + ch.bitset_off;
         double *o = out + ch.row_off;
         uint32_t r = ch.row0;
-        /* per-8-row bitset byte, prefetched ONE AHEAD like the stream
-           refills: the unprefetched dependent byte load left the kernel
-           71% memory-parked (SQ_WAIT_ANY PMC).  Byte loads are always
-           aligned (the 8-B word variant misread flakily at high blob
-           offsets); the +1 read tops out one byte past the bitset into
-           the page's own data region. */
+        /* per-8-row bitset byte, prefetched one ahead like the stream
+           refills (measured neutral: the kernel is ~72% memory-parked
+           with or without it — the stall is elsewhere, see DESIGN.md
+           round-3 notes).  Byte loads are always aligned (the 8-B word
+           variant misread flakily at high blob offsets); the +1 read
+           tops out one byte past the bitset into the page's own data. */
         uint32_t bb_idx = ch.row0 >> 3;
         uint32_t bb = bs[bb_idx];
         uint32_t bb_next = bs[bb_idx + 1];
