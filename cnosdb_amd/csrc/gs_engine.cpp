@@ -2067,28 +2067,7 @@ struct CompactArgs {
 
 __device__ __forceinline__ bool dev_ts_contains(const int64_t *t, int64_t n,
                                                 int64_t x, int64_t *pos) {
-    /* guarded interpolation search: time-series timestamps are near-
-       uniform, so the expected probe count is O(log log n) (~3-4 global
-       loads for 25k-row streams) vs ~15 for bisection — the merge
-       kernels' dominant cost.  Up to 4 interpolation steps with
-       guaranteed progress, then bisection on whatever bracket remains
-       (adversarial spacing degrades gracefully, never wrong). */
-    int64_t lo = 0, hi = n; /* lower bound in [lo, hi) */
-    if (n > 16) {
-        int64_t tl = t[0], th = t[n - 1];
-        if (x <= tl) { *pos = 0; return x == tl; }
-        if (x > th) { *pos = n; return false; }
-        int64_t a = 0, b = n - 1; /* t[a] < x <= t[b] invariant below */
-        for (int it = 0; it < 4 && b - a > 8; it++) {
-            double frac = double(x - t[a]) / double(t[b] - t[a]);
-            int64_t m = a + 1 + int64_t(frac * double(b - a - 1));
-            if (m <= a) m = a + 1;
-            if (m >= b) m = b - 1;
-            if (t[m] < x) a = m; else b = m;
-        }
-        lo = a + 1;
-        hi = b + 1; /* lower bound lies in (a, b] */
-    }
+    int64_t lo = 0, hi = n;
     while (lo < hi) { int64_t m = (lo + hi) >> 1; if (t[m] < x) lo = m + 1; else hi = m; }
     *pos = lo;
     return lo < n && t[lo] == x;
