@@ -2073,32 +2073,98 @@ __device__ __forceinline__ bool dev_ts_contains(const int64_t *t, int64_t n,
     return lo < n && t[lo] == x;
 }
 
-/* block per (set, series): owner flags + per-series owner count */
+/* block per (set, series): owner flags + per-series owner count.
+ *
+ * LDS-staged tiled merge (round-2 rework): per 1024-element tile of
+ * stream f, the matching window of each NEWER stream is block-loaded
+ * into LDS ONCE (coalesced), and every lane searches LDS instead of
+ * global memory.  Round-1 searched global per element (log2(n) loads,
+ * tolerable only because bisection's top tree levels are L1-shared);
+ * three per-lane-probe variants (tiles+gallop, strided+gallop,
+ * interpolation) all measured SLOWER than that — staging is the only
+ * scheme whose global rank traffic is "each t2 element loaded once".
+ * All block-wide loops are uniform (lanes idle within chunks, never
+ * diverge across a __syncthreads). */
+#define CM_TILE 1024 /* stream-f elements per tile (4 per lane) */
+#define CM_W 2048    /* LDS window entries (16 KiB) per chunk */
+
 __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
                            unsigned *__restrict__ err) {
+    __shared__ int64_t wts[CM_W];
+    __shared__ int64_t wb_sh[2];
+    __shared__ long long sred[256];
     int f = blockIdx.y;
     for (int s = blockIdx.x; s < nseries; s += gridDim.x) {
         DevGroup g = a.groups[f][s];
         const int64_t *t = a.ts[f] + g.row_off;
         uint8_t *fl = a.flags[f] + g.row_off;
         long long cnt = 0;
-        for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
-            int64_t x = t[j];
-            if (j > 0 && t[j - 1] >= x) atomicOr(err, DERR_FORMAT); /* "data in stream is not sorted" */
-            bool owner = true;
+        for (int64_t j0 = 0; j0 < g.nrows; j0 += CM_TILE) {
+            const int64_t je = j0 + CM_TILE < g.nrows ? j0 + CM_TILE
+                                                      : g.nrows;
+            /* this lane's elements: j0+tid, j0+tid+256, ... (sorted) */
+            int64_t x[CM_TILE / 256];
+            bool own[CM_TILE / 256];
+            int ne = 0;
+            for (int64_t j = j0 + threadIdx.x; j < je; j += blockDim.x) {
+                x[ne] = t[j];
+                own[ne] = true;
+                if (j > 0 && t[j - 1] >= x[ne])
+                    atomicOr(err, DERR_FORMAT); /* "not sorted" */
+                ne++;
+            }
+            const int64_t x_lo = t[j0], x_hi = t[je - 1];
             for (int f2 = f + 1; f2 < nsets; f2++) {
                 DevGroup g2 = a.groups[f2][s];
-                int64_t pos;
-                if (dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &pos)) {
-                    owner = false;
-                    break;
+                const int64_t *t2 = a.ts[f2] + g2.row_off;
+                const int64_t n2 = g2.nrows;
+                if (n2 == 0) continue;
+                /* window [wl, wh) = rows of t2 that can equal a tile x */
+                if (threadIdx.x == 0) {
+                    int64_t lo = 0, hi = n2;
+                    while (lo < hi) {
+                        int64_t m = (lo + hi) >> 1;
+                        if (t2[m] < x_lo) lo = m + 1; else hi = m;
+                    }
+                    wb_sh[0] = lo;
+                    int64_t lo2 = lo, hi2 = n2;
+                    while (lo2 < hi2) {
+                        int64_t m = (lo2 + hi2) >> 1;
+                        if (t2[m] <= x_hi) lo2 = m + 1; else hi2 = m;
+                    }
+                    wb_sh[1] = lo2;
+                }
+                __syncthreads();
+                const int64_t wl = wb_sh[0], wend = wb_sh[1];
+                __syncthreads();
+                int k = 0; /* this lane's next unresolved element */
+                for (int64_t wbase = wl; wbase < wend; wbase += CM_W) {
+                    const int wcnt = int(wend - wbase < CM_W ? wend - wbase
+                                                             : CM_W);
+                    for (int i = threadIdx.x; i < wcnt; i += blockDim.x)
+                        wts[i] = t2[wbase + i];
+                    __syncthreads();
+                    const int64_t w_last = wts[wcnt - 1];
+                    const bool last_chunk = wbase + wcnt == wend;
+                    while (k < ne && (x[k] <= w_last || last_chunk)) {
+                        int lo = 0, hi = wcnt; /* LDS lower bound */
+                        while (lo < hi) {
+                            int m = (lo + hi) >> 1;
+                            if (wts[m] < x[k]) lo = m + 1; else hi = m;
+                        }
+                        if (lo < wcnt && wts[lo] == x[k]) own[k] = false;
+                        k++;
+                    }
+                    __syncthreads();
                 }
             }
-            fl[j] = owner;
-            cnt += owner;
+            for (int i = 0; i < ne; i++) {
+                fl[j0 + threadIdx.x + int64_t(i) * blockDim.x] = own[i];
+                cnt += own[i];
+            }
+            __syncthreads();
         }
         /* block reduce cnt -> counts[f][s] */
-        __shared__ long long sred[256];
         sred[threadIdx.x] = cnt;
         __syncthreads();
         for (int w = blockDim.x >> 1; w > 0; w >>= 1) {
@@ -2107,6 +2173,7 @@ __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
         }
         if (threadIdx.x == 0)
             a.counts[size_t(f) * nseries + s] = sred[0];
+        __syncthreads();
     }
 }
 
@@ -2145,53 +2212,120 @@ __global__ void k_cm_prefix(CompactArgs a, int nsets, int nseries) {
 
 /* block per (set, series): scatter owners to merged positions with
  * newest-non-null value selection */
+/* LDS-staged tiled scatter (see k_cm_flags): the merged position (rank
+ * over all streams) and the dedup value come from the same LDS windows.
+ * The dedup walk is folded into a DESCENDING stream pass: the newest
+ * non-null among {f, older hits} wins (batch_builder.rs:106-155), so the
+ * first hit with a valid value while walking f-1, f-2, ... resolves it
+ * without a per-element hit array. */
 __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
                              int64_t *__restrict__ out_ts,
                              double *__restrict__ out_val,
                              uint8_t *__restrict__ out_valid) {
+    __shared__ int64_t wts[CM_W];
+    __shared__ int64_t wb_sh[2];
     int f = blockIdx.y;
     for (int s = blockIdx.x; s < nseries; s += gridDim.x) {
         DevGroup g = a.groups[f][s];
         const int64_t *t = a.ts[f] + g.row_off;
         const uint8_t *fl = a.flags[f] + g.row_off;
         const int32_t *pf = a.prefix[f] + g.row_off;
-        int64_t base = a.out_off[s];
-        for (int64_t j = threadIdx.x; j < g.nrows; j += blockDim.x) {
-            if (!fl[j]) continue;
-            int64_t x = t[j];
-            /* merged position (owners with smaller ts across all streams)
-               and, in the same pass, where x occurs in OLDER streams —
-               the dedup value walk reuses those hits instead of searching
-               again (batch_builder.rs:139-151 semantics: this element is
-               the owner, so no NEWER stream contains x; the newest
-               non-null among {f, older hits} wins, all-null -> null) */
-            int64_t pos = base + pf[j];
-            int64_t hit[GS_MAX_STREAMS];
-            for (int f2 = 0; f2 < nsets; f2++) {
-                if (f2 == f) { hit[f2] = g.row_off + j; continue; }
-                DevGroup g2 = a.groups[f2][s];
-                int64_t ins;
-                bool found =
-                    dev_ts_contains(a.ts[f2] + g2.row_off, g2.nrows, x, &ins);
-                hit[f2] = (found && f2 < f) ? g2.row_off + ins : -1;
-                pos += (ins > 0) ? a.prefix[f2][g2.row_off + ins - 1] +
-                                       a.flags[f2][g2.row_off + ins - 1]
-                                 : 0;
+        const uint8_t *vdf = a.valid[f];
+        const int64_t base = a.out_off[s];
+        for (int64_t j0 = 0; j0 < g.nrows; j0 += CM_TILE) {
+            const int64_t je = j0 + CM_TILE < g.nrows ? j0 + CM_TILE
+                                                      : g.nrows;
+            int64_t x[CM_TILE / 256];
+            int64_t pos[CM_TILE / 256];
+            double v[CM_TILE / 256];
+            bool own[CM_TILE / 256], okv[CM_TILE / 256];
+            int ne = 0;
+            for (int64_t j = j0 + threadIdx.x; j < je; j += blockDim.x) {
+                x[ne] = t[j];
+                own[ne] = fl[j] != 0;
+                pos[ne] = base + pf[j];
+                const int64_t row = g.row_off + j;
+                const bool o = !vdf || vdf[row];
+                v[ne] = o ? a.val[f][row] : 0.0;
+                okv[ne] = o;
+                ne++;
             }
-            double v = 0.0;
-            uint8_t ok = 0;
-            for (int f2 = f; f2 >= 0; f2--) {
-                if (hit[f2] < 0) continue;
-                const uint8_t *vd2 = a.valid[f2];
-                if (!vd2 || vd2[hit[f2]]) {
-                    v = a.val[f2][hit[f2]];
-                    ok = 1;
-                    break;
+            const int64_t x_lo = t[j0], x_hi = t[je - 1];
+            /* descending pass: newer streams contribute rank only; older
+               streams contribute rank + the first valid dedup value */
+            for (int f2 = nsets - 1; f2 >= 0; f2--) {
+                if (f2 == f) continue;
+                DevGroup g2 = a.groups[f2][s];
+                const int64_t *t2 = a.ts[f2] + g2.row_off;
+                const int32_t *pf2 = a.prefix[f2] + g2.row_off;
+                const uint8_t *fl2 = a.flags[f2] + g2.row_off;
+                const int64_t n2 = g2.nrows;
+                if (n2 == 0) continue;
+                if (threadIdx.x == 0) {
+                    int64_t lo = 0, hi = n2;
+                    while (lo < hi) {
+                        int64_t m = (lo + hi) >> 1;
+                        if (t2[m] < x_lo) lo = m + 1; else hi = m;
+                    }
+                    wb_sh[0] = lo;
+                    int64_t lo2 = lo, hi2 = n2;
+                    while (lo2 < hi2) {
+                        int64_t m = (lo2 + hi2) >> 1;
+                        if (t2[m] <= x_hi) lo2 = m + 1; else hi2 = m;
+                    }
+                    wb_sh[1] = lo2;
+                }
+                __syncthreads();
+                const int64_t wl = wb_sh[0], wend = wb_sh[1];
+                __syncthreads();
+                if (wl == wend) { /* tile range absent from t2: rank only */
+                    if (wl > 0)
+                        for (int i = 0; i < ne; i++)
+                            if (own[i])
+                                pos[i] += pf2[wl - 1] + fl2[wl - 1];
+                    continue;
+                }
+                int k = 0;
+                for (int64_t wbase = wl; wbase < wend; wbase += CM_W) {
+                    const int wcnt = int(wend - wbase < CM_W ? wend - wbase
+                                                             : CM_W);
+                    for (int i = threadIdx.x; i < wcnt; i += blockDim.x)
+                        wts[i] = t2[wbase + i];
+                    __syncthreads();
+                    const int64_t w_last = wts[wcnt - 1];
+                    const bool last_chunk = wbase + wcnt == wend;
+                    while (k < ne && (x[k] <= w_last || last_chunk)) {
+                        if (own[k]) {
+                            int lo = 0, hi = wcnt; /* LDS lower bound */
+                            while (lo < hi) {
+                                int m = (lo + hi) >> 1;
+                                if (wts[m] < x[k]) lo = m + 1; else hi = m;
+                            }
+                            const int64_t p2 = wbase + lo;
+                            if (p2 > 0)
+                                pos[k] += pf2[p2 - 1] + fl2[p2 - 1];
+                            if (f2 < f && !okv[k] && lo < wcnt &&
+                                wts[lo] == x[k]) {
+                                const int64_t row2 = g2.row_off + p2;
+                                const uint8_t *vd2 = a.valid[f2];
+                                if (!vd2 || vd2[row2]) {
+                                    v[k] = a.val[f2][row2];
+                                    okv[k] = true;
+                                }
+                            }
+                        }
+                        k++;
+                    }
+                    __syncthreads();
                 }
             }
-            out_ts[pos] = x;
-            out_val[pos] = v;
-            if (out_valid) out_valid[pos] = ok;
+            for (int i = 0; i < ne; i++) {
+                if (!own[i]) continue;
+                out_ts[pos[i]] = x[i];
+                out_val[pos[i]] = okv[i] ? v[i] : 0.0;
+                if (out_valid) out_valid[pos[i]] = okv[i];
+            }
+            __syncthreads();
         }
     }
 }
