@@ -2223,6 +2223,10 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
                              double *__restrict__ out_val,
                              uint8_t *__restrict__ out_valid) {
     __shared__ int64_t wts[CM_W];
+    __shared__ int32_t wpf[CM_W]; /* prefix window: the per-owner rank adds
+                                     were ~645M scattered 4/1-B global
+                                     loads — staged once, coalesced */
+    __shared__ uint8_t wfl[CM_W];
     __shared__ int64_t wb_sh[2];
     int f = blockIdx.y;
     for (int s = blockIdx.x; s < nseries; s += gridDim.x) {
@@ -2289,8 +2293,11 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
                 for (int64_t wbase = wl; wbase < wend; wbase += CM_W) {
                     const int wcnt = int(wend - wbase < CM_W ? wend - wbase
                                                              : CM_W);
-                    for (int i = threadIdx.x; i < wcnt; i += blockDim.x)
+                    for (int i = threadIdx.x; i < wcnt; i += blockDim.x) {
                         wts[i] = t2[wbase + i];
+                        wpf[i] = pf2[wbase + i];
+                        wfl[i] = fl2[wbase + i];
+                    }
                     __syncthreads();
                     const int64_t w_last = wts[wcnt - 1];
                     const bool last_chunk = wbase + wcnt == wend;
@@ -2302,7 +2309,9 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
                                 if (wts[m] < x[k]) lo = m + 1; else hi = m;
                             }
                             const int64_t p2 = wbase + lo;
-                            if (p2 > 0)
+                            if (lo > 0)
+                                pos[k] += wpf[lo - 1] + wfl[lo - 1];
+                            else if (p2 > 0)
                                 pos[k] += pf2[p2 - 1] + fl2[p2 - 1];
                             if (f2 < f && !okv[k] && lo < wcnt &&
                                 wts[lo] == x[k]) {
