@@ -2227,8 +2227,6 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
                                      were ~645M scattered 4/1-B global
                                      loads — staged once, coalesced */
     __shared__ uint8_t wfl[CM_W];
-    __shared__ double wval[CM_W]; /* dedup-value window (older streams) */
-    __shared__ uint8_t wvd[CM_W];
     __shared__ int64_t wb_sh[2];
     int f = blockIdx.y;
     for (int s = blockIdx.x; s < nseries; s += gridDim.x) {
@@ -2295,17 +2293,10 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
                 for (int64_t wbase = wl; wbase < wend; wbase += CM_W) {
                     const int wcnt = int(wend - wbase < CM_W ? wend - wbase
                                                              : CM_W);
-                    const uint8_t *vd2 = a.valid[f2];
-                    const bool want_val = f2 < f;
                     for (int i = threadIdx.x; i < wcnt; i += blockDim.x) {
                         wts[i] = t2[wbase + i];
                         wpf[i] = pf2[wbase + i];
                         wfl[i] = fl2[wbase + i];
-                        if (want_val) {
-                            const int64_t row2 = g2.row_off + wbase + i;
-                            wval[i] = a.val[f2][row2];
-                            wvd[i] = vd2 ? vd2[row2] : 1;
-                        }
                     }
                     __syncthreads();
                     const int64_t w_last = wts[wcnt - 1];
@@ -2322,10 +2313,14 @@ __global__ void k_cm_scatter(CompactArgs a, int nsets, int nseries,
                                 pos[k] += wpf[lo - 1] + wfl[lo - 1];
                             else if (p2 > 0)
                                 pos[k] += pf2[p2 - 1] + fl2[p2 - 1];
-                            if (want_val && !okv[k] && lo < wcnt &&
-                                wts[lo] == x[k] && wvd[lo]) {
-                                v[k] = wval[lo];
-                                okv[k] = true;
+                            if (f2 < f && !okv[k] && lo < wcnt &&
+                                wts[lo] == x[k]) {
+                                const int64_t row2 = g2.row_off + p2;
+                                const uint8_t *vd2 = a.valid[f2];
+                                if (!vd2 || vd2[row2]) {
+                                    v[k] = a.val[f2][row2];
+                                    okv[k] = true;
+                                }
                             }
                         }
                         k++;
