@@ -2085,7 +2085,7 @@ __device__ __forceinline__ bool dev_ts_contains(const int64_t *t, int64_t n,
  * scheme whose global rank traffic is "each t2 element loaded once".
  * All block-wide loops are uniform (lanes idle within chunks, never
  * diverge across a __syncthreads). */
-#define CM_TILE 1024 /* stream-f elements per tile (4 per lane) */
+#define CM_TILE 2048 /* stream-f elements per tile (8 per lane; 1024 and 4096 measured slower) */
 #define CM_W 2048    /* LDS window entries (16 KiB) per chunk */
 
 __global__ void k_cm_flags(CompactArgs a, int nsets, int nseries,
