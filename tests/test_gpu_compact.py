@@ -130,3 +130,66 @@ def test_compact_merge_disjoint_streams(engine):
         assert gv.view(np.uint64).tolist() == ev.view(np.uint64).tolist()
     for g in gsets:
         g.free()
+
+
+def test_compact_merge_fuzz(engine):
+    """Randomized merge configurations stressing the LDS-window kernels:
+    k up to 16 (GS_MAX_STREAMS), wildly different stream densities (empty
+    windows / windows much larger than a tile), tiny and tile-boundary
+    series sizes, heavy collisions, nulls."""
+    r = np.random.default_rng(13)
+    for trial in range(10):
+        k = int(r.choice([2, 3, 5, 8, 16]))
+        nseries = int(r.integers(1, 4))
+        sizes = [int(r.choice([1, 7, 100, 2047, 2048, 2049, 5000]))
+                 for _ in range(k)]
+        nulls = float(r.choice([0.0, 0.3]))
+        grid = T0 + np.arange(20000, dtype=np.int64) * NS
+        streams = []
+        for f in range(k):
+            per = []
+            for s in range(nseries):
+                take = np.sort(r.choice(grid.size, size=min(sizes[f],
+                                                            grid.size),
+                                        replace=False))
+                ts = grid[take]
+                vals = np.round(r.normal(50, 10, ts.size), 1)
+                valid = (r.random(ts.size) > nulls) if nulls else None
+                per.append((ts, vals, valid))
+            streams.append(per)
+        gsets, tss, vls, vds = [], [], [], []
+        total = 0
+        for f in range(k):
+            gset = _upload_stream(engine, streams[f])
+            d_ts = torch.zeros(gset.rows, dtype=torch.int64, device="cuda")
+            d_val = torch.zeros(gset.rows, dtype=torch.float64,
+                                device="cuda")
+            d_vd = torch.zeros(gset.rows, dtype=torch.uint8, device="cuda")
+            engine.decode(gset, 0, d_ts)
+            engine.decode(gset, 1, d_val, d_vd)
+            gsets.append(gset)
+            tss.append(d_ts)
+            vls.append(d_val)
+            vds.append(d_vd)
+            total += gset.rows
+        d_ots = torch.zeros(total, dtype=torch.int64, device="cuda")
+        d_oval = torch.zeros(total, dtype=torch.float64, device="cuda")
+        d_ovd = torch.zeros(total, dtype=torch.uint8, device="cuda")
+        out_rows, offs = engine.compact_merge(gsets, tss, vls, vds,
+                                              d_ots, d_oval, d_ovd)
+        got_ts = d_ots[:out_rows].cpu().numpy()
+        got_val = d_oval[:out_rows].cpu().numpy()
+        got_vd = d_ovd[:out_rows].cpu().numpy()
+        for s in range(nseries):
+            per = [streams[f][s] for f in range(k)]
+            ets, ev, evd = orc.merge_dedup(per)
+            lo, hi = offs[s], offs[s + 1]
+            assert hi - lo == ets.size, f"trial {trial} series {s}"
+            assert (got_ts[lo:hi] == ets).all(), f"trial {trial} ts"
+            assert (got_vd[lo:hi].astype(bool) == evd).all(), \
+                f"trial {trial} vd"
+            gv = got_val[lo:hi]
+            assert gv[evd].view(np.uint64).tolist() == \
+                ev[evd].view(np.uint64).tolist(), f"trial {trial} val"
+        for g in gsets:
+            g.free()
