@@ -140,3 +140,54 @@ def test_groupby_tag_high_cardinality(engine):
         assert gc[s].sum() == npts
         assert (gm[s][nz] == float(s)).all()
     gset.free()
+
+
+def test_groupby_error_paths(engine):
+    """gs_groupby_tag preconditions fail loudly: no prior aggregate scan,
+    no decoded string column, cap too small."""
+    rng = np.random.default_rng(3)
+    n = 1024
+    ts = T0 + np.arange(n, dtype=np.int64) * NS
+    vals = rng.normal(50, 5, n)
+    gset = engine.upload([(0, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                               (gs.page_of(vals, gs.CT_F64), gs.CT_F64),
+                               (gs.str_page_of([b"h"] * n), gs.CT_STR)])])
+    rows = gset.rows
+    nb = 4
+    g_max = torch.zeros(nb, dtype=torch.float64, device="cuda")
+    g_sum = torch.zeros(nb, dtype=torch.float64, device="cuda")
+    g_cnt = torch.zeros(nb, dtype=torch.int64, device="cuda")
+    with pytest.raises(RuntimeError, match="prior aggregate scan"):
+        gs.groupby_tag(engine, gset, nb, g_max, g_sum, g_cnt, 4)
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    agg = dict(bucket_ns=BUCKET, t0=T0, n_buckets=nb,
+               d_max=torch.full((nb,), -np.inf, dtype=torch.float64,
+                                device="cuda"),
+               d_sum=torch.zeros(nb, dtype=torch.float64, device="cuda"),
+               d_count=torch.zeros(nb, dtype=torch.int64, device="cuda"))
+    engine.scan(gset, d_ts, d_val, agg=agg)
+    with pytest.raises(RuntimeError, match="decoded string column"):
+        gs.groupby_tag(engine, gset, nb, g_max, g_sum, g_cnt, 4)
+    gset.free()
+
+
+def test_scan_fields_rejects_unfusable(engine):
+    """gs_scan_fields is fused-only: null-carrying fields must error."""
+    rng = np.random.default_rng(4)
+    n = 1024
+    ts = T0 + np.arange(n, dtype=np.int64) * NS
+    vals = rng.normal(50, 5, n)
+    valid = rng.random(n) > 0.5
+    gset = engine.upload([(0, [(gs.page_of(ts, gs.CT_TIME), gs.CT_TIME),
+                               (gs.page_of(vals, gs.CT_F64, valid),
+                                gs.CT_F64)])])
+    rows = gset.rows
+    d_ts = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_val = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    d_ots = torch.zeros(rows, dtype=torch.int64, device="cuda")
+    d_oval = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    with pytest.raises(RuntimeError, match="fused-capable"):
+        gs.scan_fields(engine, gset, [0], d_ts, d_val,
+                       (int(ts[0]), int(ts[-1])), d_ots, d_oval)
+    gset.free()
