@@ -87,3 +87,17 @@ def test_mixed_everything_scan():
     assert np.allclose(d_sum.cpu().numpy(), esm, rtol=1e-12)
     gset.free()
     eng.close()
+
+
+def test_cpp_shim_host_end_to_end():
+    """The C++ host-side mirror of ColumnGroupReader (examples/
+    shim_host.cpp) exercises the C ABI with no Python in the loop:
+    build pages -> upload -> scan -> verify -> error behaviour."""
+    import subprocess
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exe = os.path.join(repo, "examples", "shim_host")
+    assert os.path.exists(exe), "built by __graft_entry__.build()"
+    out = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    assert "shim_host OK" in out.stdout
