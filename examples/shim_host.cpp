@@ -39,7 +39,9 @@ static std::vector<uint8_t> build_page(const uint8_t *data, size_t len,
     const uint32_t nb = (nrows + 7) / 8;
     std::vector<uint8_t> p(16 + nb + len);
     const uint32_t crc = gs_crc32(data, len);
-    const uint64_t dl = len;
+    /* the u64 header field is the ROW COUNT (the reference names it
+       data_len but stores num_values — page.rs:32-38) */
+    const uint64_t dl = nrows;
     for (int i = 0; i < 4; i++) p[i] = uint8_t(nb >> (24 - 8 * i));
     for (int i = 0; i < 8; i++) p[4 + i] = uint8_t(dl >> (56 - 8 * i));
     for (int i = 0; i < 4; i++) p[12 + i] = uint8_t(crc >> (24 - 8 * i));
